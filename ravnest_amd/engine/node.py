@@ -1,0 +1,731 @@
+"""Per-GPU Node runtime: one process per MI355X, one pipeline stage per
+process, RCCL channels between stages.
+
+Capability parity with the reference's Node (ravnest/node.py:23-783):
+role dispatch (root/stem/leaf), bounded in-flight async pipeline, action
+handlers (root_forward/forward/backward/find_loss/no_grad variants/
+save_submodel), periodic DP parameter averaging, latest-weights pull,
+submodel save cascade. Differences are deliberate MI355X-first design:
+
+* transport is RCCL P2P over xGMI (comm/p2p.py), not gRPC + pickle; the
+  gRPC server subprocess + mp.Manager buffers (node.py:259-328,
+  endpoints.py) are replaced by listener threads + an event-driven
+  dispatch queue (no busy-poll);
+* producers send outputs DIRECTLY to each consumer stage (skip
+  connections included), so the reference's hop-by-hop payload forwarding
+  and pass-through grad accumulation (node.py:533-540) become direct
+  grad messages that the producer sums per output;
+* DP parameter averaging is a bucketed RCCL all_reduce on the per-stage
+  communicator (comm/collectives.py), not a hand-rolled chunked ring.
+"""
+from __future__ import annotations
+
+import itertools
+import json
+import queue
+import threading
+import time
+from pathlib import Path
+
+import torch
+
+from ..comm import CommBackend, Message
+from ..comm.collectives import average_parameters, average_optimizer_state
+from ..strings import ActionTypes, NodeTypes
+from ..utils import load_node_json_configs
+from .compute import ComputeEngine
+
+_MAX_OUTS = 1024
+_MODEL_INPUT_BASE = 1 << 20
+
+# output_type codes carried in Message.extra for NO_GRAD_FORWARD
+_OUT_TYPES = {"val_accuracy": 0, "accuracy": 1, "prediction": 2}
+_OUT_TYPES_REV = {v: k for k, v in _OUT_TYPES.items()}
+
+
+def _dtype_from_str(s: str) -> torch.dtype:
+    return getattr(torch, s.replace("torch.", ""))
+
+
+class _CyclingIterator:
+    """Auto-resetting DataLoader iterator (the leaf's label feed must
+    survive epoch boundaries; parity: reference labels iterator,
+    node.py:144-163)."""
+
+    def __init__(self, loader):
+        self.loader = loader
+        self.it = iter(loader) if loader is not None else None
+
+    def __next__(self):
+        if self.it is None:
+            return None
+        try:
+            return next(self.it)
+        except StopIteration:
+            self.it = iter(self.loader)
+            return next(self.it)
+
+
+class Node:
+    def __init__(self, name: str | None = None,
+                 base_dir: str = "node_data",
+                 config: dict | None = None,
+                 model: torch.nn.Module | None = None,
+                 input_template: list | None = None,
+                 output_template: dict | None = None,
+                 optimizer=None,
+                 optimizer_params: dict | None = None,
+                 device: torch.device | None = None,
+                 criterion=None,
+                 labels=None,
+                 test_labels=None,
+                 update_frequency: int = 1,
+                 reduce_factor: int | None = None,
+                 average_optim: bool = False,
+                 loss_filename: str = "losses.txt",
+                 backend: str | None = None,
+                 master_addr: str = "127.0.0.1",
+                 master_port: int = 29500,
+                 comm: CommBackend | None = None,
+                 compression: bool = False,
+                 wire_dtype: torch.dtype | None = None):
+        self.base_dir = base_dir
+        if config is None:
+            config = load_node_json_configs(name, base_dir)
+        self.config = config
+        self.rank = config["rank"]
+        self.world_size = config["world_size"]
+        self.cluster_id = config.get("cluster_id", 0)
+        self.stage = config.get("stage", 0)
+        self.n_stages = config.get("n_stages", 1)
+        self.cluster_length = config.get("cluster_length", self.n_stages)
+        self.stage_ranks = config.get("stage_ranks", [self.rank])
+        self.dp_ranks = config.get("dp_ranks", [self.rank])
+        self.node_type = NodeTypes(config.get("node_type", "root"))
+        self.template_path = config.get("template_path")
+        self.name = config.get("name", f"node_{self.rank}")
+
+        if device is not None:
+            self.device = torch.device(device)
+        elif torch.cuda.is_available():
+            local = int(__import__("os").environ.get(
+                "LOCAL_RANK", self.rank % max(1, torch.cuda.device_count())))
+            self.device = torch.device("cuda", local)
+        else:
+            self.device = torch.device("cpu")
+
+        self.fused = self.n_stages == 1  # root==leaf: whole model on 1 GPU
+
+        # ---- model & templates --------------------------------------
+        if model is None:
+            model = torch.load(Path(self.template_path) / "submod.pt",
+                               map_location="cpu", weights_only=False)
+        self.model = model.to(self.device)
+        if input_template is None and self.template_path:
+            with open(Path(self.template_path) / "inputs.json") as f:
+                input_template = json.load(f)
+        if output_template is None and self.template_path:
+            with open(Path(self.template_path) / "outputs.json") as f:
+                output_template = {int(k): v for k, v in json.load(f).items()}
+        self.input_template = input_template or []
+        self.output_template = output_template or {}
+        mi_path = Path(base_dir) / "model_inputs.json"
+        if mi_path.exists():
+            with open(mi_path) as f:
+                self.model_input_names = json.load(f)["input_names"]
+        else:
+            self.model_input_names = []
+
+        # ---- optimizer / engine -------------------------------------
+        optimizer_params = optimizer_params or {}
+        opt = None
+        if optimizer is not None:
+            opt = optimizer(self.model.parameters(), **optimizer_params)
+        self.optimizer = opt
+        self.criterion = criterion
+        self.labels = _CyclingIterator(labels)
+        self.test_labels = _CyclingIterator(test_labels)
+        self.update_frequency = update_frequency
+        self.reduce_factor = reduce_factor
+        self.reduce_threshold = (update_frequency * reduce_factor
+                                 if reduce_factor else None)
+        self.average_optim = average_optim
+        self.loss_filename = loss_filename
+        self.engine = ComputeEngine(self.model, opt, self.device,
+                                    update_frequency=update_frequency,
+                                    criterion=criterion,
+                                    loss_filename=loss_filename)
+
+        # ---- routing precomputation ---------------------------------
+        self._build_routing()
+
+        # ---- pipeline state -----------------------------------------
+        self.forward_pass_id = 0
+        self.eval_pass_id = 0
+        self.latest_backward_id = -1
+        self._backward_done = threading.Condition()
+        self._pending_fwd: dict[tuple, dict] = {}
+        self._pending_grads: dict[int, dict] = {}
+        self._pending_lock = threading.Lock()
+        self._actions: queue.PriorityQueue = queue.PriorityQueue()
+        self._action_seq = itertools.count()
+        self._stop = threading.Event()
+        self._weights_reply_q: queue.Queue = queue.Queue()
+        self.losses: list[float] = []
+        self.val_accuracies: list[float] = []
+
+        # ---- comm ----------------------------------------------------
+        if comm is not None:
+            self.comm = comm
+        elif self.world_size > 1:
+            self.comm = CommBackend(
+                rank=self.rank, world_size=self.world_size,
+                base_dir=base_dir, device=self.device, backend=backend,
+                master_addr=master_addr, master_port=master_port)
+        else:
+            self.comm = None
+        self.comm_session = _CommSessionFacade(self)
+
+        self._dispatch_thread: threading.Thread | None = None
+        self._started = False
+
+    # ==================================================================
+    # routing tables
+    # ==================================================================
+    def _build_routing(self):
+        """Precompute, from the templates: per-input source rank/gid,
+        per-output consumer ranks, expected grad contributions."""
+        self.root_rank = self.stage_ranks[0]
+        # one gid may feed several input positions (an output consumed
+        # twice by this stage); the producer sends the tensor once
+        self._gid_to_pos: dict[int, list[int]] = {}
+        self._pos_src_rank: dict[int, int] = {}
+        self._needs_grad: list[bool] = []
+        self._pos_to_producer: dict[int, tuple[int, int, int]] = {}
+        self._n_remote_inputs = 0
+        self._local_positions: list[int] = []
+
+        for pos, src in enumerate(self.input_template):
+            kind = src.get("kind")
+            dtype_s = src.get("dtype", "torch.float32")
+            floaty = _dtype_from_str(dtype_s).is_floating_point \
+                if hasattr(_dtype_from_str(dtype_s), "is_floating_point") \
+                else True
+            if kind == "stage":
+                j = src["stage"]
+                k = src.get("out_idx", 0)
+                gid = j * _MAX_OUTS + k
+                src_rank = self.stage_ranks[j]
+                self._needs_grad.append(bool(floaty))
+                self._pos_to_producer[pos] = (src_rank, j, k)
+            elif kind == "model_input":
+                idx = (self.model_input_names.index(src["name"])
+                       if src["name"] in self.model_input_names else 0)
+                gid = _MODEL_INPUT_BASE + idx
+                src_rank = self.root_rank
+                self._needs_grad.append(False)
+            else:  # const
+                self._needs_grad.append(False)
+                self._local_positions.append(pos)
+                continue
+            first_use = gid not in self._gid_to_pos
+            self._gid_to_pos.setdefault(gid, []).append(pos)
+            self._pos_src_rank[pos] = src_rank
+            if src_rank != self.rank:
+                if first_use:
+                    self._n_remote_inputs += 1  # tensor arrives once per gid
+            else:
+                self._local_positions.append(pos)
+
+        # outputs: consumers grouped by rank; expected grad contributions
+        self._out_consumers: dict[int, list[int]] = {}  # out_idx -> ranks
+        self._expected_grad_contribs: dict[int, int] = {}
+        for k, entry in self.output_template.items():
+            ranks = []
+            for cons in entry.get("consumers", []):
+                j = cons["stage"]
+                if j != self.stage:
+                    ranks.append(self.stage_ranks[j])
+            self._out_consumers[k] = ranks
+            dt = entry.get("dtype", "torch.float32")
+            floaty = _dtype_from_str(dt).is_floating_point
+            self._expected_grad_contribs[k] = len(ranks) if floaty else 0
+        self._total_expected_grads = sum(self._expected_grad_contribs.values())
+
+        # model inputs this root must forward to later stages:
+        # recomputed lazily from other stages' templates at plan time is
+        # encoded in the comm edge list; the root resolves targets from
+        # all stage input templates on disk.
+        self._model_input_targets: dict[str, list[tuple[int, int]]] = {}
+        if self.node_type == NodeTypes.ROOT and self.template_path:
+            base = Path(self.base_dir)
+            for s in range(1, self.n_stages):
+                p = base / f"cluster_{self.cluster_id}" / f"stage_{s}" / "inputs.json"
+                if not p.exists():
+                    continue
+                with open(p) as f:
+                    tmpl = json.load(f)
+                for src in tmpl:
+                    if src.get("kind") == "model_input":
+                        self._model_input_targets.setdefault(
+                            src["name"], []).append(
+                                (self.stage_ranks[s],
+                                 _MODEL_INPUT_BASE +
+                                 (self.model_input_names.index(src["name"])
+                                  if src["name"] in self.model_input_names
+                                  else 0)))
+
+    # ==================================================================
+    # lifecycle
+    # ==================================================================
+    def start(self):
+        if self._started:
+            return
+        self._started = True
+        if self.comm is not None:
+            self.comm.start(self._deliver)
+        self._dispatch_thread = threading.Thread(
+            target=self._dispatch_loop, daemon=True,
+            name=f"dispatch:rank{self.rank}")
+        self._dispatch_thread.start()
+
+    def stop(self):
+        self._stop.set()
+        self._enqueue(2, ("_stop", None, None))
+        if self.comm is not None:
+            self.comm.stop()
+
+    # ==================================================================
+    # message delivery (listener threads) -> assembly -> action queue
+    # ==================================================================
+    def _enqueue(self, prio: int, item):
+        self._actions.put((prio, next(self._action_seq), item))
+
+    def _deliver(self, channel, msg: Message):
+        if msg.action in (ActionTypes.FORWARD, ActionTypes.NO_GRAD_FORWARD):
+            self._assemble_forward(msg)
+        elif msg.action == ActionTypes.BACKWARD:
+            self._assemble_backward(msg)
+        elif msg.action == ActionTypes.SAVE_SUBMODEL:
+            self._enqueue(1, ("save_submodel", None, None))
+        elif msg.action == ActionTypes.STOP:
+            self._enqueue(3, ("stop_cascade", None, None))
+        elif msg.action == ActionTypes.PREDICTION and channel.kind == "ctrl":
+            # WEIGHTS protocol over ctrl: extra==1 request, extra==2 reply
+            if msg.extra == 1:
+                self._enqueue(1, ("weights_request", channel.src, None))
+            else:
+                self._weights_reply_q.put(msg.tensors)
+
+    def _assemble_forward(self, msg: Message):
+        is_eval = msg.action == ActionTypes.NO_GRAD_FORWARD
+        key = (is_eval, msg.fpid)
+        with self._pending_lock:
+            ent = self._pending_fwd.setdefault(
+                key, {"tensors": {}, "gids": set(), "extra": msg.extra})
+            for gid, t in msg.tensors:
+                for pos in self._gid_to_pos.get(gid, ()):
+                    ent["tensors"][pos] = t
+                ent["gids"].add(gid)
+            if msg.extra:
+                ent["extra"] = msg.extra
+            done = len(ent["gids"]) >= self._n_remote_inputs
+            if done:
+                del self._pending_fwd[key]
+        if done:
+            action = (ActionTypes.NO_GRAD_FORWARD if is_eval
+                      else ActionTypes.FORWARD)
+            self._enqueue(1, (str(action), msg.fpid,
+                              {"tensors": ent["tensors"],
+                               "extra": ent["extra"]}))
+
+    def _assemble_backward(self, msg: Message):
+        with self._pending_lock:
+            ent = self._pending_grads.setdefault(
+                msg.fpid, {"grads": {}, "count": 0})
+            for gid, g in msg.tensors:
+                k = gid % _MAX_OUTS
+                if k in ent["grads"]:
+                    ent["grads"][k] = ent["grads"][k] + g
+                else:
+                    ent["grads"][k] = g
+                ent["count"] += 1
+            done = ent["count"] >= self._total_expected_grads
+            if done:
+                del self._pending_grads[msg.fpid]
+        if done:
+            self._enqueue(0, (str(ActionTypes.BACKWARD), msg.fpid,
+                              ent["grads"]))
+
+    # ==================================================================
+    # dispatch loop
+    # ==================================================================
+    def _dispatch_loop(self):
+        while not self._stop.is_set():
+            try:
+                _, _, item = self._actions.get(timeout=1.0)
+            except queue.Empty:
+                continue
+            kind, fpid, payload = item
+            if kind == "_stop":
+                return
+            try:
+                self._dispatch(kind, fpid, payload)
+            except Exception:
+                import traceback
+                traceback.print_exc()
+                self._stop.set()
+                raise
+
+    def _dispatch(self, kind: str, fpid, payload):
+        if kind == str(ActionTypes.FORWARD):
+            if self.node_type == NodeTypes.LEAF and not self.fused:
+                self._handle_find_loss(fpid, payload)
+            else:
+                self._handle_forward(fpid, payload)
+        elif kind == str(ActionTypes.BACKWARD):
+            self._handle_backward(fpid, payload)
+        elif kind == str(ActionTypes.NO_GRAD_FORWARD):
+            self._handle_no_grad_forward(fpid, payload)
+        elif kind == str(ActionTypes.ROOT_FORWARD):
+            self._handle_root_forward(fpid, payload)
+        elif kind == "root_no_grad":
+            self._handle_root_no_grad(fpid, payload)
+        elif kind == "save_submodel":
+            self._handle_save()
+        elif kind == "weights_request":
+            self._handle_weights_request(payload_src=fpid)
+        elif kind == "stop_cascade":
+            self._forward_stop()
+            self._stop.set()
+
+    # ==================================================================
+    # argument assembly helpers
+    # ==================================================================
+    def _build_args(self, tensors_by_pos: dict[int, torch.Tensor],
+                    model_inputs: dict | None = None) -> list:
+        args = []
+        for pos, src in enumerate(self.input_template):
+            kind = src.get("kind")
+            if kind == "const":
+                args.append(src.get("value"))
+            elif pos in tensors_by_pos:
+                t = tensors_by_pos[pos]
+                args.append(t.to(self.device) if torch.is_tensor(t) else t)
+            elif kind == "model_input" and model_inputs is not None:
+                args.append(model_inputs[src["name"]])
+            else:
+                raise RuntimeError(
+                    f"rank {self.rank}: missing input pos {pos} ({src})")
+        return args
+
+    def _send_outputs(self, action: ActionTypes, fpid: int, outputs: tuple,
+                      extra: int = 0):
+        """Group outputs per consumer rank into one message each."""
+        by_rank: dict[int, list] = {}
+        for k, ranks in self._out_consumers.items():
+            if k >= len(outputs):
+                continue
+            for r in set(ranks):
+                by_rank.setdefault(r, []).append(
+                    (self.stage * _MAX_OUTS + k, outputs[k].detach()))
+        for r, tensors in by_rank.items():
+            self.comm.send(r, "fwd", Message(action=action, fpid=fpid,
+                                             tensors=tensors, extra=extra))
+
+    def _send_input_grads(self, fpid: int, input_grads: list):
+        by_rank: dict[int, list] = {}
+        for pos, g in enumerate(input_grads):
+            if g is None or pos not in self._pos_to_producer:
+                continue
+            src_rank, j, k = self._pos_to_producer[pos]
+            if src_rank == self.rank:
+                continue
+            by_rank.setdefault(src_rank, []).append((j * _MAX_OUTS + k, g))
+        for r, tensors in by_rank.items():
+            self.comm.send(r, "bwd", Message(action=ActionTypes.BACKWARD,
+                                             fpid=fpid, tensors=tensors))
+
+    # ==================================================================
+    # user-facing API (parity with reference node.py:370-428,702-746)
+    # ==================================================================
+    def forward_compute(self, tensors=None, **kwargs):
+        """Root entry: inject one microbatch into the async pipeline,
+        bounded by the in-flight window (parity node.py:370-397)."""
+        assert self.node_type == NodeTypes.ROOT or self.fused
+        # throttle: in-flight window <= cluster_length
+        with self._backward_done:
+            while (self.forward_pass_id - self.latest_backward_id
+                   > self.cluster_length) and not self._stop.is_set():
+                self._backward_done.wait(timeout=1.0)
+        if self.reduce_threshold and self.forward_pass_id > 0 and \
+                self.forward_pass_id % self.reduce_threshold == 0:
+            self.wait_for_backwards()
+        fpid = self.forward_pass_id
+        self.forward_pass_id += 1
+        self._enqueue(1, (str(ActionTypes.ROOT_FORWARD), fpid,
+                          {"tensors": tensors, "kwargs": kwargs}))
+
+    def no_grad_forward_compute(self, tensors=None, output_type="val_accuracy",
+                                **kwargs):
+        assert self.node_type == NodeTypes.ROOT or self.fused
+        self.wait_for_backwards()
+        fpid = self.eval_pass_id
+        self.eval_pass_id += 1
+        self._enqueue(1, ("root_no_grad", fpid,
+                          {"tensors": tensors, "kwargs": kwargs,
+                           "output_type": output_type}))
+        if self.fused:
+            return None
+
+    def wait_for_backwards(self, timeout: float = 600.0):
+        """Block until every injected fpid's backward completed at this
+        root (parity node.py:702-710)."""
+        if self.node_type != NodeTypes.ROOT and not self.fused:
+            return
+        deadline = time.monotonic() + timeout
+        with self._backward_done:
+            while self.latest_backward_id < self.forward_pass_id - 1:
+                if self._stop.is_set() or time.monotonic() > deadline:
+                    raise TimeoutError(
+                        f"rank {self.rank}: pipeline did not drain "
+                        f"(fwd={self.forward_pass_id}, "
+                        f"bwd={self.latest_backward_id})")
+                self._backward_done.wait(timeout=0.5)
+
+    def trigger_save_submodel(self):
+        self._handle_save(cascade=True)
+
+    def _forward_stop(self):
+        if self.fused or self.comm is None:
+            return
+        if self.stage + 1 < self.n_stages:
+            nxt = self.stage_ranks[self.stage + 1]
+            try:
+                ch = self.comm.channel(self.rank, nxt, "fwd")
+                ch.send_sync(Message(action=ActionTypes.STOP, fpid=0,
+                                     tensors=[]))
+            except KeyError:
+                pass
+
+    def stop_cluster(self):
+        """Root: drain, cascade STOP down the pipeline, stop this node."""
+        if self.node_type == NodeTypes.ROOT and not self.fused:
+            self.wait_for_backwards()
+        self._forward_stop()
+        self._stop.set()
+
+    def update_with_latest_weights(self, src_rank: int | None = None,
+                                   timeout: float = 120.0):
+        """Pull the latest parameter snapshot from a DP peer over the ctrl
+        channel (parity: reference get_latest_weights,
+        communication.py:279-330 / node.py:726-730)."""
+        if src_rank is None:
+            peers = [r for r in self.dp_ranks if r != self.rank]
+            if not peers:
+                return
+            src_rank = peers[0]
+        self.comm.send(src_rank, "ctrl",
+                       Message(action=ActionTypes.PREDICTION, fpid=0,
+                               tensors=[], extra=1))
+        tensors = self._weights_reply_q.get(timeout=timeout)
+        self.engine.load_param_list([t for _, t in tensors])
+
+    # ==================================================================
+    # action handlers
+    # ==================================================================
+    def _mark_backward_complete(self, fpid: int):
+        with self._backward_done:
+            self.latest_backward_id = max(self.latest_backward_id, fpid)
+            self._backward_done.notify_all()
+
+    def _maybe_reduce(self):
+        if not self.reduce_threshold or self.comm is None:
+            return
+        if self.engine.n_backwards % self.reduce_threshold == 0 and \
+                len(self.dp_ranks) > 1:
+            average_parameters(self.model, self.comm.my_dp_group)
+            if self.average_optim and self.optimizer is not None:
+                average_optimizer_state(self.optimizer, self.comm.my_dp_group)
+            self.engine.bump_version()
+
+    def _handle_root_forward(self, fpid: int, payload):
+        tensors = payload["tensors"]
+        kwargs = payload["kwargs"] or {}
+        model_inputs = self._map_model_inputs(tensors, kwargs)
+        if self.fused:
+            targets = next(self.labels)
+            args = self._root_args(model_inputs)
+            _, stepped, _ = self.engine.find_loss(
+                fpid, args, [False] * len(args), targets)
+            self._mark_backward_complete(fpid)
+            self._maybe_reduce()
+            return
+        # forward model inputs consumed by later stages
+        for name, targets_list in self._model_input_targets.items():
+            val = model_inputs.get(name)
+            if val is None:
+                continue
+            by_rank: dict[int, list] = {}
+            for (r, gid) in targets_list:
+                by_rank.setdefault(r, []).append((gid, val))
+            for r, ts in by_rank.items():
+                self.comm.send(r, "fwd", Message(
+                    action=ActionTypes.FORWARD, fpid=fpid, tensors=ts))
+        args = self._root_args(model_inputs)
+        outputs = self.engine.forward(fpid, args, self._needs_grad)
+        self._send_outputs(ActionTypes.FORWARD, fpid, outputs)
+
+    def _root_args(self, model_inputs: dict) -> list:
+        args = []
+        for src in self.input_template:
+            if src.get("kind") == "model_input":
+                v = model_inputs.get(src["name"])
+                if torch.is_tensor(v):
+                    v = v.to(self.device)
+                args.append(v)
+            elif src.get("kind") == "const":
+                args.append(src.get("value"))
+            else:
+                raise RuntimeError("root stage with non-model inputs")
+        return args
+
+    def _map_model_inputs(self, tensors, kwargs) -> dict:
+        mi = {}
+        names = self.model_input_names or \
+            [s["name"] for s in self.input_template
+             if s.get("kind") == "model_input"]
+        if tensors is not None and names:
+            mi[names[0]] = tensors.to(self.device) \
+                if torch.is_tensor(tensors) else tensors
+        for k, v in kwargs.items():
+            if torch.is_tensor(v):
+                v = v.to(self.device)
+            mi[k] = v
+        return mi
+
+    def _handle_forward(self, fpid: int, payload):
+        args = self._build_args(payload["tensors"])
+        outputs = self.engine.forward(fpid, args, self._needs_grad)
+        self._send_outputs(ActionTypes.FORWARD, fpid, outputs)
+
+    def _handle_find_loss(self, fpid: int, payload):
+        args = self._build_args(payload["tensors"])
+        targets = next(self.labels)
+        input_grads, stepped, loss = self.engine.find_loss(
+            fpid, args, self._needs_grad, targets)
+        self.losses.append(loss)
+        self._send_input_grads(fpid, input_grads)
+        self._maybe_reduce()
+
+    def _handle_backward(self, fpid: int, grads: dict):
+        input_grads, stepped = self.engine.backward(fpid, grads)
+        if self.node_type != NodeTypes.ROOT:
+            self._send_input_grads(fpid, input_grads)
+        else:
+            self._mark_backward_complete(fpid)
+        self._maybe_reduce()
+
+    def _handle_root_no_grad(self, fpid: int, payload):
+        tensors = payload["tensors"]
+        kwargs = payload["kwargs"] or {}
+        output_type = payload.get("output_type", "val_accuracy")
+        model_inputs = self._map_model_inputs(tensors, kwargs)
+        if self.fused:
+            args = self._root_args(model_inputs)
+            outputs = self.engine.no_grad_forward(args)
+            self._leaf_eval_output(outputs, output_type)
+            return
+        for name, targets_list in self._model_input_targets.items():
+            val = model_inputs.get(name)
+            if val is None:
+                continue
+            by_rank: dict[int, list] = {}
+            for (r, gid) in targets_list:
+                by_rank.setdefault(r, []).append((gid, val))
+            for r, ts in by_rank.items():
+                self.comm.send(r, "fwd", Message(
+                    action=ActionTypes.NO_GRAD_FORWARD, fpid=fpid, tensors=ts,
+                    extra=_OUT_TYPES[output_type]))
+        args = self._root_args(model_inputs)
+        outputs = self.engine.no_grad_forward(args)
+        self._send_outputs(ActionTypes.NO_GRAD_FORWARD, fpid, outputs,
+                           extra=_OUT_TYPES[output_type])
+
+    def _handle_no_grad_forward(self, fpid: int, payload):
+        args = self._build_args(payload["tensors"])
+        outputs = self.engine.no_grad_forward(args)
+        if self.node_type == NodeTypes.LEAF:
+            self._leaf_eval_output(outputs,
+                                   _OUT_TYPES_REV[payload.get("extra", 0)])
+        else:
+            self._send_outputs(ActionTypes.NO_GRAD_FORWARD, fpid, outputs,
+                               extra=payload.get("extra", 0))
+
+    def _leaf_eval_output(self, outputs: tuple, output_type: str):
+        out = outputs[0]
+        if output_type == "val_accuracy":
+            batch = next(self.test_labels)
+            if batch is None:
+                return
+            y = batch[1] if isinstance(batch, (tuple, list)) else batch
+            y = y.to(out.device) if torch.is_tensor(y) else y
+            pred = out.argmax(dim=-1)
+            truth = y.argmax(dim=-1) if torch.is_tensor(y) and y.dim() > 1 \
+                else y
+            acc = float((pred == truth).float().mean())
+            self.val_accuracies.append(acc)
+            with open("val_accuracies.txt", "a") as f:
+                f.write(f"{round(acc, 4)}\n")
+        elif output_type == "accuracy":
+            batch = next(self.test_labels)
+            if batch is None:
+                return
+            y = batch[1] if isinstance(batch, (tuple, list)) else batch
+            y = y.to(out.device) if torch.is_tensor(y) else y
+            pred = out.argmax(dim=-1)
+            truth = y.argmax(dim=-1) if torch.is_tensor(y) and y.dim() > 1 \
+                else y
+            acc = float((pred == truth).float().mean())
+            print("Accuracy:", acc)
+        else:  # prediction
+            torch.save(out.detach().cpu(), "prediction.pt")
+
+    def _handle_save(self, cascade: bool = False):
+        if self.template_path:
+            p = Path(self.template_path)
+            torch.save(self.model.state_dict(), p / "trained_state_dict.pt")
+        if (cascade or self.node_type != NodeTypes.LEAF) and not self.fused \
+                and self.comm is not None and self.stage + 1 < self.n_stages:
+            nxt = self.stage_ranks[self.stage + 1]
+            try:
+                self.comm.send(nxt, "fwd", Message(
+                    action=ActionTypes.SAVE_SUBMODEL, fpid=0, tensors=[]))
+            except KeyError:
+                pass
+
+    def _handle_weights_request(self, payload_src: int):
+        snap = self.engine.latest_state_snapshot()
+        tensors = [(i, t) for i, t in enumerate(snap["params"])]
+        self.comm.send(payload_src, "ctrl",
+                       Message(action=ActionTypes.PREDICTION, fpid=0,
+                               tensors=tensors, extra=2))
+
+
+class _CommSessionFacade:
+    """API-parity shim: the reference exposes
+    `node.comm_session.parallel_ring_reduce()` (trainer.py:99,
+    communication.py:125); here that is a bucketed RCCL all_reduce on the
+    per-stage DP group."""
+
+    def __init__(self, node: Node):
+        self.node = node
+
+    def parallel_ring_reduce(self):
+        n = self.node
+        if n.comm is None or len(n.dp_ranks) <= 1:
+            return
+        average_parameters(n.model, n.comm.my_dp_group)
+        if n.average_optim and n.optimizer is not None:
+            average_optimizer_state(n.optimizer, n.comm.my_dp_group)
+        n.engine.bump_version()

@@ -1,0 +1,417 @@
+"""Proportional, parameter-size-based pipeline-stage splitting on torch.fx.
+
+Capability parity with the reference's offline splitter
+(ravnest/operations/pippy_utils.py:43-155 `_split_on_size_thresholds_with_max_stages`
+/ `split_on_proportions`, and ravnest/operations/utils.py:265-349
+`split_model_equal` which derives the input/output routing templates), built
+instead on stock torch.fx + `torch.fx.passes.split_module` — no PiPPy.
+
+The output of :func:`split_model_by_proportions` is a list of per-stage
+``torch.fx.GraphModule`` submodels plus explicit routing templates:
+
+* ``stage_inputs[i]``  — ordered source descriptors for stage *i*'s
+  positional inputs: ``{"kind": "model_input", "name": ...}`` or
+  ``{"kind": "stage", "stage": j, "out_idx": k}``.
+* ``stage_outputs[i]`` — for each output index of stage *i*, the list of
+  consumers ``[(stage_j, input_pos)]`` and whether it is (part of) the
+  final model output.
+
+These templates drive the runtime's direct producer->consumer RCCL sends
+(multi-consumer outputs, tuple `getitem` indices and root model-input
+forwarding all become explicit edges — the reference instead forwarded
+payload dicts hop-by-hop over gRPC, communication.py:98-123).
+"""
+from __future__ import annotations
+
+import math
+from dataclasses import dataclass, field
+from typing import Any, Callable
+
+import torch
+import torch.fx as fx
+from torch.fx.passes.split_module import split_module
+
+
+class LeafHonoringTracer(fx.Tracer):
+    """fx tracer that honors a module's `_is_leaf_module = True` attribute
+    (parity: reference CustomTracer, operations/utils.py:200-232 and the
+    sorter example's fx-friendly modules)."""
+
+    def is_leaf_module(self, m: torch.nn.Module, qualname: str) -> bool:
+        if getattr(m, "_is_leaf_module", False):
+            return True
+        return super().is_leaf_module(m, qualname)
+
+
+def trace_model(model: torch.nn.Module,
+                concrete_args: dict | None = None) -> fx.GraphModule:
+    tracer = LeafHonoringTracer()
+    graph = tracer.trace(model, concrete_args=concrete_args)
+    return fx.GraphModule(tracer.root, graph)
+
+
+def _node_param_bytes(gm: fx.GraphModule, node: fx.Node,
+                      counted: set[str]) -> int:
+    """Parameter+buffer bytes introduced by `node` (counted once per
+    qualified name so shared modules are charged to their first user)."""
+    size = 0
+    if node.op == "call_module":
+        target = str(node.target)
+        if target not in counted:
+            counted.add(target)
+            try:
+                sub = gm.get_submodule(target)
+            except AttributeError:
+                return 0
+            for p in sub.parameters():
+                size += p.numel() * p.element_size()
+            for b in sub.buffers():
+                size += b.numel() * b.element_size()
+    elif node.op == "get_attr":
+        target = str(node.target)
+        if target not in counted:
+            counted.add(target)
+            obj: Any = gm
+            try:
+                for atom in target.split("."):
+                    obj = getattr(obj, atom)
+            except AttributeError:
+                return 0
+            if isinstance(obj, torch.Tensor):
+                size += obj.numel() * obj.element_size()
+    return size
+
+
+def hare_niemeyer_counts(total: int, proportions: list[float]) -> list[int]:
+    """Largest-remainder apportionment of `total` units over `proportions`
+    (parity: reference operations/utils.py:69-80)."""
+    shares = [total * p / sum(proportions) for p in proportions]
+    counts = [math.floor(s) for s in shares]
+    rem = total - sum(counts)
+    order = sorted(range(len(shares)), key=lambda i: shares[i] - counts[i],
+                   reverse=True)
+    for i in order[:rem]:
+        counts[i] += 1
+    return counts
+
+
+@dataclass
+class SplitResult:
+    stages: list[fx.GraphModule]
+    stage_inputs: list[list[dict]]
+    stage_outputs: list[dict[int, dict]]
+    model_input_names: list[str]
+    final_outputs: list[dict]
+    split_gm: fx.GraphModule = field(repr=False, default=None)
+    param_name_maps: list[dict] = field(default_factory=list)
+
+
+def _assign_partitions(gm: fx.GraphModule, n_stages: int,
+                       proportions: list[float]) -> dict[fx.Node, int]:
+    """Walk nodes in topological order, accumulating parameter bytes and
+    bumping the stage index when the cumulative size crosses each
+    proportional threshold."""
+    counted: set[str] = set()
+    weights = {}
+    total = 0
+    for node in gm.graph.nodes:
+        w = _node_param_bytes(gm, node, counted)
+        weights[node] = w
+        total += w
+    if total == 0:
+        # Parameter-free model: fall back to counting compute nodes evenly.
+        weights = {n: (1 if n.op not in ("placeholder", "output") else 0)
+                   for n in gm.graph.nodes}
+        total = sum(weights.values())
+
+    # cumulative thresholds
+    cum_props = []
+    acc = 0.0
+    for p in proportions[:-1]:
+        acc += p / sum(proportions)
+        cum_props.append(acc * total)
+
+    assignment: dict[fx.Node, int] = {}
+    cum = 0
+    stage = 0
+    stage_has_weight = False
+    for node in gm.graph.nodes:
+        if node.op in ("placeholder", "output"):
+            continue
+        w = weights[node]
+        # bump stage BEFORE assigning when the node lands mostly past the
+        # threshold (midpoint rule keeps a single huge layer from bloating
+        # the current stage) and the current stage already has weighted work
+        while (stage < n_stages - 1 and stage_has_weight
+               and cum + 0.5 * w >= cum_props[stage]):
+            stage += 1
+            stage_has_weight = False
+        assignment[node] = stage
+        cum += w
+        if w > 0:
+            stage_has_weight = True
+    return assignment
+
+
+def _force_chain_order(assignment: dict[fx.Node, int]) -> None:
+    """split_module emits partitions in dependency order; a monotone
+    assignment along the topological node order already forms a chain, but
+    make sure no node is assigned to an earlier stage than a producer."""
+    for node, stage in list(assignment.items()):
+        for inp in node.all_input_nodes:
+            if inp in assignment and assignment[inp] > stage:
+                assignment[node] = assignment[inp]
+
+
+def split_model_by_proportions(
+    model: torch.nn.Module,
+    proportions: list[float],
+    example_args: tuple = (),
+    example_kwargs: dict | None = None,
+    concrete_args: dict | None = None,
+    max_retries: int = 3,
+) -> SplitResult:
+    """Trace `model` and split it into ``len(proportions)`` pipeline stages
+    whose parameter bytes approximate `proportions`.
+
+    Includes the reference's trial-forward-with-repair loop
+    (operations/utils.py:256-278): if the split produces fewer stages than
+    requested (a stage got no nodes), proportions are re-balanced and the
+    split retried.
+    """
+    example_kwargs = example_kwargs or {}
+    n_stages = len(proportions)
+    gm = trace_model(model, concrete_args=concrete_args)
+
+    props = list(proportions)
+    last_err = None
+    for _ in range(max_retries):
+        assignment = _assign_partitions(gm, n_stages, props)
+        _force_chain_order(assignment)
+        used = sorted(set(assignment.values()))
+        if len(used) != n_stages:
+            props = [1.0 / n_stages] * n_stages  # repair: equalize
+            last_err = RuntimeError(
+                f"split produced {len(used)} stages, wanted {n_stages}")
+            continue
+        remap = {s: i for i, s in enumerate(used)}
+
+        def cb(node: fx.Node) -> int:
+            return remap.get(assignment.get(node, 0), 0)
+
+        split_gm = split_module(gm, model, cb)
+        _shape_propagate(split_gm, example_args, example_kwargs)
+        result = _extract_routing(split_gm, n_stages)
+        if result is None:
+            props = [1.0 / n_stages] * n_stages
+            last_err = RuntimeError("routing extraction failed")
+            continue
+        # trial forward: split must match the original numerically
+        # (eval mode: dropout/BN would make the two runs diverge)
+        if example_args or example_kwargs:
+            was_training = gm.training
+            gm.eval()
+            split_gm.eval()
+            with torch.no_grad():
+                ref = gm(*example_args, **example_kwargs)
+                out = split_gm(*example_args, **example_kwargs)
+            _assert_close_struct(ref, out)
+            if was_training:
+                gm.train()
+                split_gm.train()
+        result.split_gm = split_gm
+        result.param_name_maps = [
+            {k: k for k, _ in stage.state_dict().items()}
+            for stage in result.stages
+        ]
+        return result
+    raise last_err or RuntimeError("split failed")
+
+
+def _assert_close_struct(a, b):
+    if isinstance(a, torch.Tensor):
+        assert torch.allclose(a, b, atol=1e-5, rtol=1e-4), \
+            "split model diverges from original"
+    elif isinstance(a, (tuple, list)):
+        for x, y in zip(a, b):
+            _assert_close_struct(x, y)
+    elif isinstance(a, dict):
+        for k in a:
+            _assert_close_struct(a[k], b[k])
+
+
+def _shape_propagate(split_gm: fx.GraphModule, example_args: tuple,
+                     example_kwargs: dict | None) -> None:
+    """Run fx ShapeProp so every top-graph value carries tensor_meta
+    (dtype/shape). The runtime's grad-return contract (only floating,
+    stage-sourced tensors get grads) and static recv-buffer sizing depend
+    on these annotations."""
+    example_kwargs = example_kwargs or {}
+    if not example_args and not example_kwargs:
+        return
+    from torch.fx.passes.shape_prop import ShapeProp
+    placeholders = [n for n in split_gm.graph.nodes if n.op == "placeholder"]
+    args = list(example_args)
+    for p in placeholders[len(args):]:
+        name = p.target if isinstance(p.target, str) else p.name
+        if name in example_kwargs:
+            args.append(example_kwargs[name])
+        elif p.name in example_kwargs:
+            args.append(example_kwargs[p.name])
+    try:
+        ShapeProp(split_gm).propagate(*args)
+    except Exception:
+        pass  # meta stays absent; runtime falls back to float32 defaults
+
+
+def _meta_of(node: fx.Node, out_idx: int | None = None, whole: bool = False):
+    tm = node.meta.get("tensor_meta")
+    if tm is None:
+        return None
+    if whole or out_idx is None:
+        return tm if hasattr(tm, "dtype") else None
+    if hasattr(tm, "dtype"):
+        return tm if out_idx == 0 else None
+    try:
+        sub = tm[out_idx]
+        return sub if hasattr(sub, "dtype") else None
+    except (IndexError, TypeError, KeyError):
+        return None
+
+
+def _annot(desc: dict, meta) -> dict:
+    if meta is not None:
+        desc["dtype"] = str(meta.dtype)
+        desc["shape"] = list(meta.shape)
+    return desc
+
+
+def _extract_routing(split_gm: fx.GraphModule, n_stages: int) -> SplitResult | None:
+    """Walk the top-level split graph and derive explicit routing templates.
+
+    Handles multi-consumer outputs, tuple outputs (`getitem`), and model
+    inputs forwarded to arbitrary later stages (parity with the template
+    semantics of reference operations/utils.py:280-325).
+    """
+    submod_nodes: dict[str, fx.Node] = {}
+    placeholders: list[fx.Node] = []
+    output_node = None
+    for node in split_gm.graph.nodes:
+        if node.op == "placeholder":
+            placeholders.append(node)
+        elif node.op == "call_module" and str(node.target).startswith("submod_"):
+            submod_nodes[str(node.target)] = node
+        elif node.op == "output":
+            output_node = node
+
+    if len(submod_nodes) != n_stages:
+        return None
+
+    stage_order = [f"submod_{i}" for i in range(n_stages)]
+    if any(name not in submod_nodes for name in stage_order):
+        return None
+
+    stage_of_node: dict[fx.Node, int] = {
+        submod_nodes[name]: i for i, name in enumerate(stage_order)}
+    model_input_names = [p.name for p in placeholders]
+
+    # source descriptor of a value-producing top-graph node
+    # -> ("model_input", name) | (stage, out_idx)
+    def source_of(n: fx.Node):
+        import operator
+        if n.op == "placeholder":
+            return _annot({"kind": "model_input", "name": n.name},
+                          _meta_of(n, whole=True))
+        if n in stage_of_node:
+            return _annot({"kind": "stage", "stage": stage_of_node[n],
+                           "out_idx": 0, "whole": True},
+                          _meta_of(n, whole=True))
+        if n.op == "call_function" and n.target is operator.getitem:
+            base, idx = n.args
+            if base in stage_of_node and isinstance(idx, int):
+                return _annot({"kind": "stage", "stage": stage_of_node[base],
+                               "out_idx": idx}, _meta_of(n, whole=True))
+        return None
+
+    stage_inputs: list[list[dict]] = [[] for _ in range(n_stages)]
+    stage_outputs: list[dict[int, dict]] = [dict() for _ in range(n_stages)]
+    n_stage_outputs = [0] * n_stages
+
+    # how many outputs does each stage produce? look at users
+    import operator
+    for name, node in submod_nodes.items():
+        i = stage_of_node[node]
+        getitem_idxs = [u.args[1] for u in node.users
+                        if u.op == "call_function" and u.target is operator.getitem
+                        and isinstance(u.args[1], int)]
+        whole_users = [u for u in node.users
+                       if not (u.op == "call_function" and u.target is operator.getitem)]
+        if getitem_idxs:
+            n_stage_outputs[i] = max(getitem_idxs) + 1
+            if whole_users:
+                # both whole-tuple and item use: unsupported mix
+                return None
+        else:
+            n_stage_outputs[i] = 1
+
+    for i, name in enumerate(stage_order):
+        node = submod_nodes[name]
+        for pos, arg in enumerate(node.args):
+            if not isinstance(arg, fx.Node):
+                stage_inputs[i].append({"kind": "const", "value": arg})
+                continue
+            src = source_of(arg)
+            if src is None:
+                return None
+            stage_inputs[i].append(src)
+            if src["kind"] == "stage":
+                j, k = src["stage"], src["out_idx"]
+                entry = stage_outputs[j].setdefault(
+                    k, {"consumers": [], "final": False})
+                entry["consumers"].append({"stage": i, "input_pos": pos})
+        if node.kwargs:
+            return None  # split_module emits positional-only calls
+
+    # final outputs
+    final_outputs: list[dict] = []
+
+    def walk_output(val):
+        if isinstance(val, fx.Node):
+            src = source_of(val)
+            if src is None:
+                return False
+            final_outputs.append(src)
+            if src["kind"] == "stage":
+                j, k = src["stage"], src["out_idx"]
+                entry = stage_outputs[j].setdefault(
+                    k, {"consumers": [], "final": False})
+                entry["final"] = True
+            return True
+        if isinstance(val, (tuple, list)):
+            return all(walk_output(v) for v in val)
+        if val is None:
+            return True
+        return False
+
+    if output_node is not None:
+        for val in output_node.args:
+            if not walk_output(val):
+                return None
+
+    # annotate output entries with dtype/shape from the producing submod
+    for j, name in enumerate(stage_order):
+        for k, entry in stage_outputs[j].items():
+            _annot(entry, _meta_of(submod_nodes[name],
+                                   out_idx=None if n_stage_outputs[j] == 1
+                                   else k,
+                                   whole=n_stage_outputs[j] == 1))
+
+    stages = [split_gm.get_submodule(name) for name in stage_order]
+    return SplitResult(
+        stages=stages,
+        stage_inputs=stage_inputs,
+        stage_outputs=stage_outputs,
+        model_input_names=model_input_names,
+        final_outputs=final_outputs,
+        split_gm=split_gm,
+    )

@@ -1,0 +1,95 @@
+"""End-to-end async pipeline on CPU: 3 stages, 3 processes, gloo backend.
+
+This is the reference's canonical walkthrough workload (CNN on synthetic
+8x8 digits, 3-node pipeline — SURVEY.md section 7 minimum slice) run over
+the MI355X framework's comm stack with the gloo backend standing in for
+RCCL. Asserts: training runs, loss decreases, shutdown is clean.
+"""
+import json
+import os
+
+import numpy as np
+import pytest
+import torch
+import torch.multiprocessing as mp
+
+from ravnest_amd import clusterize, set_seed
+from ravnest_amd.models.cnn import CNN
+from ravnest_amd.planner import NodeSpec
+
+
+def _make_loaders(seed=42, n=256, batch=32):
+    g = torch.Generator()
+    g.manual_seed(seed)
+    rng = np.random.RandomState(0)
+    X = rng.rand(n, 1, 8, 8).astype("float32")
+    y = rng.randint(0, 10, size=n)
+    Y = np.zeros((n, 10), dtype="float32")
+    Y[np.arange(n), y] = 1.0
+    ds = list(zip(torch.tensor(X), torch.tensor(Y)))
+    from torch.utils.data import DataLoader
+    return DataLoader(ds, batch_size=batch, shuffle=True, generator=g)
+
+
+def _loss_fn(preds, targets):
+    return torch.nn.functional.mse_loss(preds, targets[1])
+
+
+def _worker(rank, world, base_dir, port, out_dir):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    os.environ["GLOO_SOCKET_IFNAME"] = os.environ.get("GLOO_SOCKET_IFNAME", "lo")
+    os.chdir(out_dir)
+    set_seed(42)
+    from ravnest_amd import Node, Trainer
+
+    train_loader = _make_loaders()
+    node = Node(name=f"node_{rank}", base_dir=base_dir,
+                optimizer=torch.optim.Adam,
+                device=torch.device("cpu"),
+                criterion=_loss_fn,
+                labels=_make_loaders(),  # same seed => same order as root
+                update_frequency=1)
+    node.start()
+    trainer = Trainer(node=node, train_loader=train_loader,
+                      epochs=3, batch_size=32, inputs_dtype=torch.float32)
+    if rank == 0:
+        trainer.train()
+        node.stop_cluster()
+        losses = np.loadtxt(os.path.join(out_dir, "leaf_losses.txt")) \
+            if os.path.exists(os.path.join(out_dir, "leaf_losses.txt")) else None
+    else:
+        trainer.prelim_checks()
+    node.stop()
+
+
+def test_three_stage_pipeline_cpu(tmp_path):
+    set_seed(42)
+    model = CNN()
+    x = torch.randn(2, 1, 8, 8)
+    base = str(tmp_path / "node_data")
+    pool = [NodeSpec(name=f"n{i}", ram=10 * 2**20) for i in range(3)]
+    clusterize(model, (x,), node_pool=pool, max_clusters=1, base_dir=base)
+
+    port = 29510 + (os.getpid() % 500)
+    ctx = mp.get_context("spawn")
+    procs = [ctx.Process(target=_worker,
+                         args=(r, 3, base, port, str(tmp_path)))
+             for r in range(3)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(timeout=180)
+    for p in procs:
+        assert p.exitcode == 0, f"worker exited {p.exitcode}"
+
+    # leaf (rank 2) wrote losses.txt in tmp_path (cwd of workers)
+    loss_file = tmp_path / "losses.txt"
+    assert loss_file.exists(), "leaf produced no losses.txt"
+    losses = [float(l) for l in loss_file.read_text().split()]
+    n_batches = 256 // 32
+    assert len(losses) == 3 * n_batches
+    first_epoch = sum(losses[:n_batches])
+    last_epoch = sum(losses[-n_batches:])
+    assert last_epoch < first_epoch, \
+        f"loss did not decrease: {first_epoch} -> {last_epoch}"

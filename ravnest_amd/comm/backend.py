@@ -175,9 +175,14 @@ class CommBackend:
                                 else "cpu")
                 dist.all_reduce(t, group=g)
 
-    def stop(self):
+    def stop(self, join_timeout: float = 10.0):
+        """Graceful shutdown: close every outgoing channel (flush + close
+        sentinel), then wait for local channel threads to exit."""
         for ch in self.channels.values():
-            ch.stop()
+            if ch.my_rank == ch.src:
+                ch.close()
+        for ch in self.channels.values():
+            ch.join(timeout=join_timeout)
 
     # -- sending helpers ----------------------------------------------
     def channel(self, src: int, dst: int, kind: str) -> Channel:

@@ -33,6 +33,8 @@ CODE_ACTIONS = {i: a for a, i in ACTION_CODES.items()}
 _MAX_DIMS = 8
 _HEADER_LEN = 4  # [action, fpid, n_tensors, extra]
 _META_PER_TENSOR = 3 + _MAX_DIMS  # [gid, dtype_code, ndim, d0..d7]
+CLOSE_EXTRA = (1 << 31) - 1  # channel-close sentinel (clean shutdown)
+_CLOSE = object()
 
 
 @dataclass
@@ -86,6 +88,17 @@ class Channel:
             except queue.Full:
                 pass
 
+    def close(self):
+        """Graceful close: drain queued sends, then ship a close sentinel
+        so the peer's recv loop exits cleanly (no thread left blocked in a
+        collective at interpreter teardown)."""
+        if self.my_rank == self.src:
+            self._send_q.put(_CLOSE)
+
+    def join(self, timeout: float = 5.0):
+        for t in self._threads:
+            t.join(timeout=timeout)
+
     # -- sending -------------------------------------------------------
     def send(self, msg: Message):
         assert self.my_rank == self.src
@@ -95,9 +108,15 @@ class Channel:
         self._do_send(msg)
 
     def _send_loop(self):
-        while not self._stop.is_set():
+        while True:
             msg = self._send_q.get()
             if msg is None:
+                return
+            if msg is _CLOSE:
+                self._do_send(Message(action=ActionTypes.STOP, fpid=-1,
+                                      tensors=[], extra=CLOSE_EXTRA))
+                return
+            if self._stop.is_set():
                 return
             self._do_send(msg)
 
@@ -149,6 +168,8 @@ class Channel:
                     return
                 raise
             if msg is None:
+                return
+            if msg.action == ActionTypes.STOP and msg.extra == CLOSE_EXTRA:
                 return
             if self.deliver is not None:
                 self.deliver(self, msg)

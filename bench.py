@@ -1,0 +1,213 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: BERT-base MLM pretraining step on 1..8 MI355X.
+
+Driver contract (BASELINE.json): metric = samples/sec (whole node) for the
+reference's BERT-base pretraining workload (examples/bert/provider.py
+config family), async DP via the ravnest_amd Node (one fused
+root+leaf replica per GPU, periodic RCCL parameter averaging — the
+reference's cross-cluster ring-averaging semantics). Weak scaling:
+per-GPU micro-batch fixed as N grows.
+
+Launch:
+  python bench.py --gpus 1 --steps 20 --warmup 5
+  python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
+      --master-addr 127.0.0.1 bench.py --gpus N --steps K --warmup W
+
+Rank 0 prints ONE JSON line with the whole-job aggregate.
+"""
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import sys
+import time
+
+import torch
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+from ravnest_amd import Node, set_seed  # noqa: E402
+from ravnest_amd.comm import CommBackend  # noqa: E402
+from ravnest_amd.ops import CrossEntropyLoss, FusedAdam  # noqa: E402
+from ravnest_amd.models import BertConfig, BertForMLM, GPTConfig, GPT  # noqa: E402
+
+
+def build_model(name: str, seq: int):
+    if name == "bert-base":
+        return BertForMLM(BertConfig.base(max_seq=seq)), "bert"
+    if name == "bert-tiny":
+        return BertForMLM(BertConfig.tiny(max_seq=seq)), "bert"
+    if name == "gpt2-small":
+        return GPT(GPTConfig.gpt2_small(block_size=seq)), "gpt"
+    raise ValueError(name)
+
+
+class _ListLoader:
+    def __init__(self, items):
+        self.items = items
+
+    def __iter__(self):
+        return iter(self.items)
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=20)
+    ap.add_argument("--warmup", type=int, default=5)
+    ap.add_argument("--model", default="bert-base")
+    ap.add_argument("--micro-batch", type=int, default=16)
+    ap.add_argument("--seq", type=int, default=512)
+    ap.add_argument("--reduce-factor", type=int, default=4,
+                    help="average DP replicas every N steps")
+    ap.add_argument("--dtype", default="bf16", choices=["bf16", "fp32"])
+    ap.add_argument("--cpu", action="store_true",
+                    help="CPU sanity mode (tiny model)")
+    args = ap.parse_args()
+
+    rank = int(os.environ.get("RANK", 0))
+    world = int(os.environ.get("WORLD_SIZE", args.gpus))
+    local_rank = int(os.environ.get("LOCAL_RANK", rank))
+    if world > 1:
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        os.environ.setdefault("MASTER_PORT", "29531")
+
+    on_gpu = torch.cuda.is_available() and not args.cpu
+    if not on_gpu and args.model == "bert-base" and args.cpu:
+        args.model = "bert-tiny"
+    if args.cpu:
+        args.micro_batch = min(args.micro_batch, 4)
+        args.seq = min(args.seq, 64)
+
+    device = torch.device("cuda", local_rank) if on_gpu else torch.device("cpu")
+    if on_gpu:
+        torch.cuda.set_device(device)
+    set_seed(1234 + 0)  # same init on every replica (DP semantics)
+
+    model, family = build_model(args.model, args.seq)
+    model = model.to(device)
+    amp = torch.bfloat16 if (args.dtype == "bf16" and on_gpu) else None
+
+    # ---- synthetic data of the workload's shape (no network: random ids,
+    # random MLM labels with 85% ignored), pre-staged on device ----------
+    set_seed(1234 + rank)  # different data per replica
+    vocab = model.cfg.vocab_size
+    nb = 4
+    batches, labels = [], []
+    for _ in range(nb):
+        ids = torch.randint(0, vocab, (args.micro_batch, args.seq),
+                            device=device)
+        if family == "bert":
+            mask = torch.ones(args.micro_batch, args.seq, dtype=torch.int64,
+                              device=device)
+            y = ids.clone()
+            drop = torch.rand(ids.shape, device=device) > 0.15
+            y[drop] = -100
+            batches.append({"input_ids": ids, "attention_mask": mask})
+            labels.append(y)
+        else:
+            batches.append({"idx": ids})
+            y = torch.roll(ids, -1, dims=1)
+            labels.append(y)
+
+    comm = None
+    if world > 1:
+        comm = CommBackend(rank=rank, world_size=world, edges=[],
+                           dp_groups=[list(range(world))], device=device)
+
+    mi_names = (["input_ids", "attention_mask"] if family == "bert"
+                else ["idx"])
+    cfg = {
+        "rank": rank, "world_size": world, "cluster_id": rank, "stage": 0,
+        "n_stages": 1, "cluster_length": 1, "stage_ranks": [rank],
+        "dp_ranks": list(range(world)), "node_type": "root",
+        "model_input_names": mi_names,
+    }
+    input_template = [{"kind": "model_input", "name": n,
+                       "dtype": "torch.int64"} for n in mi_names]
+    output_template = {0: {"consumers": [], "final": True,
+                           "dtype": "torch.float32"}}
+
+    node = Node(config=cfg, model=model, input_template=input_template,
+                output_template=output_template,
+                optimizer=FusedAdam, optimizer_params={"lr": 1e-4},
+                device=device, criterion=CrossEntropyLoss(-100),
+                labels=_ListLoader([labels[i % nb]
+                                    for i in range(args.warmup + args.steps + nb)]),
+                update_frequency=1, reduce_factor=args.reduce_factor,
+                comm=comm, amp_dtype=amp,
+                loss_filename="bench_losses.txt")
+    node.start()
+
+    def sync():
+        if on_gpu:
+            torch.cuda.synchronize(device)
+
+    def barrier():
+        if comm is not None:
+            comm.barrier()
+
+    # ---- warmup ------------------------------------------------------
+    for i in range(args.warmup):
+        node.forward_compute(**batches[i % nb])
+    node.wait_for_backwards()
+    sync()
+    barrier()
+    sync()
+
+    # ---- timed region: EXACTLY K steps -------------------------------
+    t0 = time.perf_counter()
+    for i in range(args.steps):
+        node.forward_compute(**batches[i % nb])
+    node.wait_for_backwards()
+    sync()
+    elapsed = time.perf_counter() - t0
+    barrier()
+
+    # max over ranks
+    if comm is not None:
+        t = torch.tensor([elapsed], dtype=torch.float64,
+                         device=device if comm.backend == "nccl" else "cpu")
+        torch.distributed.all_reduce(t, op=torch.distributed.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    global_batch = args.micro_batch * world
+    samples_per_sec = global_batch * args.steps / elapsed
+    ms_per_step = elapsed / args.steps * 1000.0
+
+    if rank == 0:
+        print(json.dumps({
+            "metric": "samples/sec (whole node), BERT-base MLM pretraining"
+                      if args.model.startswith("bert")
+                      else f"samples/sec (whole node), {args.model}",
+            "value": round(samples_per_sec, 2),
+            "unit": "samples/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(ms_per_step, 2),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": args.dtype if on_gpu else "fp32",
+            "data": "synthetic",
+            "config": {
+                "model": args.model,
+                "global_batch": global_batch,
+                "seq_len": args.seq,
+                "parallelism": f"dp{world}",
+                "reduce_factor": args.reduce_factor,
+                "optimizer": "fused_adam",
+            },
+        }), flush=True)
+
+    node.stop()
+    if comm is not None:
+        comm.stop()
+        torch.distributed.barrier()
+    os._exit(0)  # listener threads may be parked in recv; exit hard
+
+
+if __name__ == "__main__":
+    main()

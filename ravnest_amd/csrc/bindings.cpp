@@ -1,0 +1,48 @@
+// Python bindings for the ravnest_amd CDNA4 kernel library (_C).
+#include <torch/extension.h>
+
+#include <vector>
+
+std::vector<at::Tensor> layernorm_fwd(at::Tensor x, at::Tensor w,
+                                      at::Tensor b, double eps);
+std::vector<at::Tensor> layernorm_bwd(at::Tensor dy, at::Tensor x,
+                                      at::Tensor w, at::Tensor mean,
+                                      at::Tensor rstd);
+at::Tensor bias_gelu_fwd(at::Tensor x, at::Tensor bias);
+at::Tensor bias_gelu_bwd(at::Tensor dy, at::Tensor x, at::Tensor bias);
+std::vector<at::Tensor> dropout_fwd(at::Tensor x, double p, int64_t seed);
+at::Tensor dropout_bwd(at::Tensor dy, at::Tensor mask, double p);
+std::vector<at::Tensor> ce_fwd(at::Tensor logits, at::Tensor targets,
+                               int64_t ignore_index);
+at::Tensor ce_bwd(at::Tensor logits, at::Tensor targets, at::Tensor lse,
+                  at::Tensor gscale, int64_t ignore_index);
+void fused_adam(std::vector<at::Tensor> ps, std::vector<at::Tensor> gs,
+                std::vector<at::Tensor> ms, std::vector<at::Tensor> vs,
+                double lr, double b1, double b2, double eps, double wd,
+                double bc1, double bc2);
+void fused_sgd(std::vector<at::Tensor> ps, std::vector<at::Tensor> gs,
+               std::vector<at::Tensor> bufs, double lr, double momentum,
+               double wd, bool nesterov);
+void fused_lamb(std::vector<at::Tensor> ps, std::vector<at::Tensor> gs,
+                std::vector<at::Tensor> ms, std::vector<at::Tensor> vs,
+                double lr, double b1, double b2, double eps, double wd,
+                double bc1, double bc2, double clamp_trust);
+std::vector<at::Tensor> attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v,
+                                 at::Tensor mask, bool causal, double scale);
+at::Tensor mfma_probe(at::Tensor a, at::Tensor b);
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("layernorm_fwd", &layernorm_fwd, "fused LayerNorm fwd (CDNA4)");
+  m.def("layernorm_bwd", &layernorm_bwd, "fused LayerNorm bwd (CDNA4)");
+  m.def("bias_gelu_fwd", &bias_gelu_fwd, "fused bias+GELU fwd");
+  m.def("bias_gelu_bwd", &bias_gelu_bwd, "fused bias+GELU bwd");
+  m.def("dropout_fwd", &dropout_fwd, "philox dropout fwd (replayable)");
+  m.def("dropout_bwd", &dropout_bwd, "dropout bwd");
+  m.def("ce_fwd", &ce_fwd, "fused softmax cross-entropy fwd");
+  m.def("ce_bwd", &ce_bwd, "fused softmax cross-entropy bwd");
+  m.def("fused_adam", &fused_adam, "multi-tensor Adam");
+  m.def("fused_sgd", &fused_sgd, "multi-tensor SGD+momentum");
+  m.def("fused_lamb", &fused_lamb, "multi-tensor LAMB");
+  m.def("attn_fwd", &attn_fwd, "flash attention fwd (bf16, MFMA)");
+  m.def("mfma_probe", &mfma_probe, "32x32x16 bf16 MFMA layout probe");
+}

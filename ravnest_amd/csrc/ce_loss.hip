@@ -1,0 +1,146 @@
+// Fused softmax cross-entropy over large vocabularies (BERT V=30522).
+//
+// fwd: one 256-thread block per row chunk computes max and log-sum-exp in
+// one pass each (block reduction), accumulates the summed loss with one
+// atomic per row. bwd: dlogits = (softmax - onehot) * gscale, elementwise
+// from (logits, lse). SURVEY.md section 2.3 loss row (split-K style row
+// passes; vectorized bf16 loads).
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+#include "common.h"
+
+namespace {
+
+template <typename T>
+DEVINL float ld(const T* p);
+template <>
+DEVINL float ld<bf16_t>(const bf16_t* p) { return bf2f(*p); }
+template <>
+DEVINL float ld<float>(const float* p) { return *p; }
+
+// one block per row: two passes over V (max, then sum-exp) with
+// vectorized loads; writes lse and the row's loss contribution.
+template <typename T>
+__global__ void ce_fwd_kernel(const T* __restrict__ logits,
+                              const long* __restrict__ targets,
+                              float* __restrict__ lse_out,
+                              float* __restrict__ loss_accum,
+                              int* __restrict__ valid_accum, int V, long N,
+                              long ignore_index) {
+  __shared__ float scratch[16];
+  const long row = blockIdx.x;
+  if (row >= N) return;
+  const T* lr = logits + row * V;
+  const long tgt = targets[row];
+
+  float m = -INFINITY;
+  for (int i = threadIdx.x; i < V; i += blockDim.x)
+    m = fmaxf(m, ld(lr + i));
+  m = block_max(m, scratch);
+
+  float s = 0.f;
+  for (int i = threadIdx.x; i < V; i += blockDim.x)
+    s += __expf(ld(lr + i) - m);
+  s = block_sum(s, scratch);
+  const float lse = m + __logf(s);
+
+  if (threadIdx.x == 0) {
+    lse_out[row] = lse;
+    if (tgt != ignore_index) {
+      atomicAdd(loss_accum, lse - ld(lr + tgt));
+      atomicAdd(valid_accum, 1);
+    }
+  }
+}
+
+template <typename T>
+__global__ void ce_bwd_kernel(const T* __restrict__ logits,
+                              const long* __restrict__ targets,
+                              const float* __restrict__ lse,
+                              const float* __restrict__ gscale,
+                              T* __restrict__ dlogits, int V, long N,
+                              long ignore_index) {
+  const long row = blockIdx.x;
+  if (row >= N) return;
+  const T* lr = logits + row * V;
+  T* dr = dlogits + row * V;
+  const long tgt = targets[row];
+  const float g = gscale[0];
+  const float l = lse[row];
+  if (tgt == ignore_index) {
+    for (int i = threadIdx.x; i < V; i += blockDim.x) {
+      if constexpr (sizeof(T) == 2)
+        dr[i] = f2bf(0.f);
+      else
+        dr[i] = 0.f;
+    }
+    return;
+  }
+  for (int i = threadIdx.x; i < V; i += blockDim.x) {
+    float p = __expf(ld(lr + i) - l);
+    float d = (p - (i == (int)tgt ? 1.f : 0.f)) * g;
+    if constexpr (sizeof(T) == 2)
+      dr[i] = f2bf(d);
+    else
+      dr[i] = d;
+  }
+}
+
+}  // namespace
+
+std::vector<at::Tensor> ce_fwd(at::Tensor logits, at::Tensor targets,
+                               int64_t ignore_index) {
+  TORCH_CHECK(logits.is_cuda() && logits.is_contiguous());
+  TORCH_CHECK(logits.dim() == 2);
+  TORCH_CHECK(targets.scalar_type() == at::kLong);
+  const long N = logits.size(0);
+  const int V = logits.size(1);
+  auto lse = at::empty({N}, logits.options().dtype(at::kFloat));
+  auto loss = at::zeros({1}, logits.options().dtype(at::kFloat));
+  auto nvalid = at::zeros({1}, logits.options().dtype(at::kInt));
+  auto stream = c10::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
+  dim3 block(256);
+  dim3 grid(N);
+  if (logits.scalar_type() == at::kBFloat16) {
+    hipLaunchKernelGGL((ce_fwd_kernel<bf16_t>), grid, block, 0, stream,
+                       reinterpret_cast<const bf16_t*>(logits.data_ptr()),
+                       targets.data_ptr<long>(), lse.data_ptr<float>(),
+                       loss.data_ptr<float>(), nvalid.data_ptr<int>(), V, N,
+                       ignore_index);
+  } else {
+    hipLaunchKernelGGL((ce_fwd_kernel<float>), grid, block, 0, stream,
+                       logits.data_ptr<float>(), targets.data_ptr<long>(),
+                       lse.data_ptr<float>(), loss.data_ptr<float>(),
+                       nvalid.data_ptr<int>(), V, N, ignore_index);
+  }
+  HIP_CHECK_LAST();
+  return {loss.squeeze(0), lse, nvalid.squeeze(0)};
+}
+
+at::Tensor ce_bwd(at::Tensor logits, at::Tensor targets, at::Tensor lse,
+                  at::Tensor gscale, int64_t ignore_index) {
+  TORCH_CHECK(logits.is_cuda() && logits.is_contiguous());
+  const long N = logits.size(0);
+  const int V = logits.size(1);
+  auto dlogits = at::empty_like(logits);
+  auto gs = gscale.to(at::kFloat).reshape({1}).contiguous();
+  auto stream = c10::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
+  dim3 block(256);
+  dim3 grid(N);
+  if (logits.scalar_type() == at::kBFloat16) {
+    hipLaunchKernelGGL((ce_bwd_kernel<bf16_t>), grid, block, 0, stream,
+                       reinterpret_cast<const bf16_t*>(logits.data_ptr()),
+                       targets.data_ptr<long>(), lse.data_ptr<float>(),
+                       gs.data_ptr<float>(),
+                       reinterpret_cast<bf16_t*>(dlogits.data_ptr()), V, N,
+                       ignore_index);
+  } else {
+    hipLaunchKernelGGL((ce_bwd_kernel<float>), grid, block, 0, stream,
+                       logits.data_ptr<float>(), targets.data_ptr<long>(),
+                       lse.data_ptr<float>(), gs.data_ptr<float>(),
+                       dlogits.data_ptr<float>(), V, N, ignore_index);
+  }
+  HIP_CHECK_LAST();
+  return dlogits;
+}

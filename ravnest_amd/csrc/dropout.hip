@@ -1,0 +1,123 @@
+// Dropout with counter-based Philox RNG (replayable).
+//
+// The (seed, offset) pair comes from the HOST (drawn from the torch CPU
+// generator by the python wrapper), so the versioned-recompute engine's
+// RNG-state capture/restore replays the identical mask — SURVEY.md
+// section 2.3 "Dropout (with replayable RNG)". Each thread generates 4
+// uniforms per philox call; mask stored as uint8.
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+#include "common.h"
+
+namespace {
+
+template <typename T>
+__global__ void dropout_fwd_kernel(const T* __restrict__ x,
+                                   T* __restrict__ y,
+                                   unsigned char* __restrict__ mask, long n,
+                                   float p, float scale,
+                                   unsigned long long seed) {
+  const Philox4 ph(seed);
+  const long i0 = (long)(blockIdx.x * blockDim.x + threadIdx.x);
+  const long stride = (long)gridDim.x * blockDim.x;
+  const float thresh = p;
+  const long nq = (n + 3) / 4;  // philox quads
+  for (long q = i0; q < nq; q += stride) {
+    unsigned int r[4];
+    ph.gen((unsigned long long)q, 0ull, r);
+    const long base = q * 4;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      const long i = base + j;
+      if (i >= n) break;
+      const float u = (r[j] >> 8) * (1.0f / 16777216.0f);  // [0,1)
+      const bool keep = u >= thresh;
+      mask[i] = keep ? 1 : 0;
+      float v = 0.f;
+      if constexpr (sizeof(T) == 2)
+        v = bf2f(x[i]);
+      else
+        v = x[i];
+      v = keep ? v * scale : 0.f;
+      if constexpr (sizeof(T) == 2)
+        y[i] = f2bf(v);
+      else
+        y[i] = v;
+    }
+  }
+}
+
+template <typename T>
+__global__ void dropout_bwd_kernel(const T* __restrict__ dy,
+                                   const unsigned char* __restrict__ mask,
+                                   T* __restrict__ dx, long n, float scale) {
+  const long i0 = (long)(blockIdx.x * blockDim.x + threadIdx.x);
+  const long stride = (long)gridDim.x * blockDim.x;
+  for (long i = i0; i < n; i += stride) {
+    float v;
+    if constexpr (sizeof(T) == 2)
+      v = bf2f(dy[i]);
+    else
+      v = dy[i];
+    v = mask[i] ? v * scale : 0.f;
+    if constexpr (sizeof(T) == 2)
+      dx[i] = f2bf(v);
+    else
+      dx[i] = v;
+  }
+}
+
+}  // namespace
+
+static int grid_cap(long work, int block) {
+  long g = (work + block - 1) / block;
+  return (int)std::min<long>(g, 2048);
+}
+
+std::vector<at::Tensor> dropout_fwd(at::Tensor x, double p, int64_t seed) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous());
+  const long n = x.numel();
+  auto y = at::empty_like(x);
+  auto mask = at::empty({n}, x.options().dtype(at::kByte));
+  const float scale = 1.0f / (1.0f - (float)p);
+  auto stream = c10::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
+  dim3 block(256);
+  dim3 grid(grid_cap((n + 3) / 4, 256));
+  if (x.scalar_type() == at::kBFloat16) {
+    hipLaunchKernelGGL((dropout_fwd_kernel<bf16_t>), grid, block, 0, stream,
+                       reinterpret_cast<const bf16_t*>(x.data_ptr()),
+                       reinterpret_cast<bf16_t*>(y.data_ptr()),
+                       mask.data_ptr<unsigned char>(), n, (float)p, scale,
+                       (unsigned long long)seed);
+  } else {
+    hipLaunchKernelGGL((dropout_fwd_kernel<float>), grid, block, 0, stream,
+                       x.data_ptr<float>(), y.data_ptr<float>(),
+                       mask.data_ptr<unsigned char>(), n, (float)p, scale,
+                       (unsigned long long)seed);
+  }
+  HIP_CHECK_LAST();
+  return {y, mask};
+}
+
+at::Tensor dropout_bwd(at::Tensor dy, at::Tensor mask, double p) {
+  TORCH_CHECK(dy.is_cuda() && dy.is_contiguous());
+  const long n = dy.numel();
+  auto dx = at::empty_like(dy);
+  const float scale = 1.0f / (1.0f - (float)p);
+  auto stream = c10::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
+  dim3 block(256);
+  dim3 grid(grid_cap(n, 256));
+  if (dy.scalar_type() == at::kBFloat16) {
+    hipLaunchKernelGGL((dropout_bwd_kernel<bf16_t>), grid, block, 0, stream,
+                       reinterpret_cast<const bf16_t*>(dy.data_ptr()),
+                       mask.data_ptr<unsigned char>(),
+                       reinterpret_cast<bf16_t*>(dx.data_ptr()), n, scale);
+  } else {
+    hipLaunchKernelGGL((dropout_bwd_kernel<float>), grid, block, 0, stream,
+                       dy.data_ptr<float>(), mask.data_ptr<unsigned char>(),
+                       dx.data_ptr<float>(), n, scale);
+  }
+  HIP_CHECK_LAST();
+  return dx;
+}

@@ -1,0 +1,154 @@
+// Fused bias + GELU (tanh approximation) forward/backward for CDNA4.
+// Elementwise, HBM-bound: vectorized 16-byte bf16x8 accesses, grid-stride
+// with the grid capped (guide Guideline 11/13).
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+#include "common.h"
+
+namespace {
+
+template <typename PT>
+DEVINL float load_pt(const PT* p);
+template <>
+DEVINL float load_pt<float>(const float* p) { return *p; }
+template <>
+DEVINL float load_pt<bf16_t>(const bf16_t* p) { return bf2f(*p); }
+
+DEVINL float gelu_f(float x) {
+  const float k = 0.7978845608028654f;  // sqrt(2/pi)
+  float t = tanhf(k * (x + 0.044715f * x * x * x));
+  return 0.5f * x * (1.f + t);
+}
+
+DEVINL float gelu_grad_f(float x) {
+  const float k = 0.7978845608028654f;
+  float x2 = x * x;
+  float th = tanhf(k * (x + 0.044715f * x * x2));
+  float sech2 = 1.f - th * th;
+  return 0.5f * (1.f + th) + 0.5f * x * sech2 * k * (1.f + 3.f * 0.044715f * x2);
+}
+
+template <typename T, typename PT>
+__global__ void bias_gelu_fwd_kernel(const T* __restrict__ x,
+                                     const PT* __restrict__ bias,
+                                     T* __restrict__ y, long n, int H) {
+  const long i0 = (long)(blockIdx.x * blockDim.x + threadIdx.x);
+  const long stride = (long)gridDim.x * blockDim.x;
+  if constexpr (sizeof(T) == 2) {
+    const long nv = n >> 3;
+    for (long i = i0; i < nv; i += stride) {
+      bf16x8 v = *reinterpret_cast<const bf16x8*>(x + i * 8);
+      const int hb = (int)((i * 8) % H);
+      bf16x8 o;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float b = load_pt(bias + (hb + j) % H);
+        o[j] = (short)f2us(gelu_f(us2f((unsigned short)v[j]) + b));
+      }
+      *reinterpret_cast<bf16x8*>(y + i * 8) = o;
+    }
+    // tail
+    for (long i = nv * 8 + i0; i < n; i += stride)
+      y[i] = f2bf(gelu_f(bf2f(x[i]) + load_pt(bias + i % H)));
+  } else {
+    for (long i = i0; i < n; i += stride) {
+      float b = load_pt(bias + i % H);
+      y[i] = gelu_f(x[i] + b);
+    }
+  }
+}
+
+template <typename T, typename PT>
+__global__ void bias_gelu_bwd_kernel(const T* __restrict__ dy,
+                                     const T* __restrict__ x,
+                                     const PT* __restrict__ bias,
+                                     T* __restrict__ dx, long n, int H) {
+  const long i0 = (long)(blockIdx.x * blockDim.x + threadIdx.x);
+  const long stride = (long)gridDim.x * blockDim.x;
+  if constexpr (sizeof(T) == 2) {
+    const long nv = n >> 3;
+    for (long i = i0; i < nv; i += stride) {
+      bf16x8 v = *reinterpret_cast<const bf16x8*>(x + i * 8);
+      bf16x8 g = *reinterpret_cast<const bf16x8*>(dy + i * 8);
+      const int hb = (int)((i * 8) % H);
+      bf16x8 o;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float b = load_pt(bias + (hb + j) % H);
+        float xf = us2f((unsigned short)v[j]) + b;
+        o[j] = (short)f2us(us2f((unsigned short)g[j]) * gelu_grad_f(xf));
+      }
+      *reinterpret_cast<bf16x8*>(dx + i * 8) = o;
+    }
+    for (long i = nv * 8 + i0; i < n; i += stride)
+      dx[i] = f2bf(bf2f(dy[i]) * gelu_grad_f(bf2f(x[i]) + load_pt(bias + i % H)));
+  } else {
+    for (long i = i0; i < n; i += stride)
+      dx[i] = dy[i] * gelu_grad_f(x[i] + load_pt(bias + i % H));
+  }
+}
+
+}  // namespace
+
+static int grid_for(long work, int block) {
+  long g = (work + block - 1) / block;
+  return (int)std::min<long>(g, 2048);
+}
+
+at::Tensor bias_gelu_fwd(at::Tensor x, at::Tensor bias) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous());
+  const int H = bias.numel();
+  TORCH_CHECK(x.size(-1) == H, "bias size mismatch");
+  const long n = x.numel();
+  auto y = at::empty_like(x);
+  auto stream = c10::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
+  dim3 block(256);
+  dim3 grid(grid_for(n / 8 + 1, 256));
+  if (x.scalar_type() == at::kBFloat16 && bias.scalar_type() == at::kFloat) {
+    hipLaunchKernelGGL((bias_gelu_fwd_kernel<bf16_t, float>), grid, block, 0,
+                       stream, reinterpret_cast<const bf16_t*>(x.data_ptr()),
+                       bias.data_ptr<float>(),
+                       reinterpret_cast<bf16_t*>(y.data_ptr()), n, H);
+  } else if (x.scalar_type() == at::kBFloat16) {
+    hipLaunchKernelGGL((bias_gelu_fwd_kernel<bf16_t, bf16_t>), grid, block, 0,
+                       stream, reinterpret_cast<const bf16_t*>(x.data_ptr()),
+                       reinterpret_cast<const bf16_t*>(bias.data_ptr()),
+                       reinterpret_cast<bf16_t*>(y.data_ptr()), n, H);
+  } else {
+    hipLaunchKernelGGL((bias_gelu_fwd_kernel<float, float>), grid, block, 0,
+                       stream, x.data_ptr<float>(), bias.data_ptr<float>(),
+                       y.data_ptr<float>(), n, H);
+  }
+  HIP_CHECK_LAST();
+  return y;
+}
+
+at::Tensor bias_gelu_bwd(at::Tensor dy, at::Tensor x, at::Tensor bias) {
+  TORCH_CHECK(dy.is_cuda() && dy.is_contiguous() && x.is_contiguous());
+  const int H = bias.numel();
+  const long n = x.numel();
+  auto dx = at::empty_like(x);
+  auto stream = c10::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
+  dim3 block(256);
+  dim3 grid(grid_for(n / 8 + 1, 256));
+  if (x.scalar_type() == at::kBFloat16 && bias.scalar_type() == at::kFloat) {
+    hipLaunchKernelGGL((bias_gelu_bwd_kernel<bf16_t, float>), grid, block, 0,
+                       stream, reinterpret_cast<const bf16_t*>(dy.data_ptr()),
+                       reinterpret_cast<const bf16_t*>(x.data_ptr()),
+                       bias.data_ptr<float>(),
+                       reinterpret_cast<bf16_t*>(dx.data_ptr()), n, H);
+  } else if (x.scalar_type() == at::kBFloat16) {
+    hipLaunchKernelGGL((bias_gelu_bwd_kernel<bf16_t, bf16_t>), grid, block, 0,
+                       stream, reinterpret_cast<const bf16_t*>(dy.data_ptr()),
+                       reinterpret_cast<const bf16_t*>(x.data_ptr()),
+                       reinterpret_cast<const bf16_t*>(bias.data_ptr()),
+                       reinterpret_cast<bf16_t*>(dx.data_ptr()), n, H);
+  } else {
+    hipLaunchKernelGGL((bias_gelu_bwd_kernel<float, float>), grid, block, 0,
+                       stream, dy.data_ptr<float>(), x.data_ptr<float>(),
+                       bias.data_ptr<float>(), dx.data_ptr<float>(), n, H);
+  }
+  HIP_CHECK_LAST();
+  return dx;
+}

@@ -47,13 +47,15 @@ class ComputeEngine:
                  device: torch.device,
                  update_frequency: int = 1,
                  criterion=None,
-                 loss_filename: str = "losses.txt"):
+                 loss_filename: str = "losses.txt",
+                 amp_dtype: torch.dtype | None = None):
         self.model = model
         self.optimizer = optimizer
         self.device = device
         self.update_frequency = max(1, update_frequency)
         self.criterion = criterion
         self.loss_filename = loss_filename
+        self.amp_dtype = amp_dtype
 
         self.current_version = 0
         self.version_to_param: dict[int, list[torch.Tensor]] = {}
@@ -66,6 +68,12 @@ class ComputeEngine:
 
         self._params = list(self.model.parameters())
         self._snapshot_current()
+
+    def _autocast(self):
+        if self.amp_dtype is not None and self.device.type == "cuda":
+            return torch.autocast("cuda", dtype=self.amp_dtype)
+        import contextlib
+        return contextlib.nullcontext()
 
     # ------------------------------------------------------------------
     # version bookkeeping
@@ -135,7 +143,7 @@ class ComputeEngine:
                           args=[a.detach().clone() if torch.is_tensor(a) else a
                                 for a in args],
                           needs_grad=list(needs_grad))
-        with torch.no_grad():
+        with torch.no_grad(), self._autocast():
             out = self.model(*args)
         outputs = out if isinstance(out, tuple) else (out,)
         with self._lock:
@@ -163,7 +171,8 @@ class ComputeEngine:
             devices = [self.device] if self.device.type == "cuda" else []
             with torch.random.fork_rng(devices=devices):
                 restore_rng_states(rec.rng, self.device)
-                out = self.model(*args)
+                with self._autocast():
+                    out = self.model(*args)
         finally:
             self._restore_params()
         rec.recomputed_outputs = out if isinstance(out, tuple) else (out,)
@@ -240,8 +249,9 @@ class ComputeEngine:
             if ng and torch.is_tensor(a) and a.is_floating_point():
                 a = a.detach().clone().requires_grad_(True)
             live_args.append(a)
-        out = self.model(*live_args)
-        loss = self.criterion(out, targets)
+        with self._autocast():
+            out = self.model(*live_args)
+            loss = self.criterion(out, targets)
         loss.backward()
         loss_val = float(loss.detach())
         self.file_loss += loss_val
@@ -275,6 +285,6 @@ class ComputeEngine:
     def no_grad_forward(self, args: list) -> tuple:
         self.join_recompute()
         self.model.eval()
-        with torch.no_grad():
+        with torch.no_grad(), self._autocast():
             out = self.model(*args)
         return out if isinstance(out, tuple) else (out,)

@@ -88,7 +88,8 @@ class Node:
                  master_port: int = 29500,
                  comm: CommBackend | None = None,
                  compression: bool = False,
-                 wire_dtype: torch.dtype | None = None):
+                 wire_dtype: torch.dtype | None = None,
+                 amp_dtype: torch.dtype | None = None):
         self.base_dir = base_dir
         if config is None:
             config = load_node_json_configs(name, base_dir)
@@ -129,12 +130,15 @@ class Node:
                 output_template = {int(k): v for k, v in json.load(f).items()}
         self.input_template = input_template or []
         self.output_template = output_template or {}
-        mi_path = Path(base_dir) / "model_inputs.json"
-        if mi_path.exists():
-            with open(mi_path) as f:
-                self.model_input_names = json.load(f)["input_names"]
+        if config.get("model_input_names"):
+            self.model_input_names = config["model_input_names"]
         else:
-            self.model_input_names = []
+            mi_path = Path(base_dir) / "model_inputs.json"
+            if mi_path.exists():
+                with open(mi_path) as f:
+                    self.model_input_names = json.load(f)["input_names"]
+            else:
+                self.model_input_names = []
 
         # ---- optimizer / engine -------------------------------------
         optimizer_params = optimizer_params or {}
@@ -154,7 +158,8 @@ class Node:
         self.engine = ComputeEngine(self.model, opt, self.device,
                                     update_frequency=update_frequency,
                                     criterion=criterion,
-                                    loss_filename=loss_filename)
+                                    loss_filename=loss_filename,
+                                    amp_dtype=amp_dtype)
 
         # ---- routing precomputation ---------------------------------
         self._build_routing()
@@ -207,10 +212,11 @@ class Node:
 
         for pos, src in enumerate(self.input_template):
             kind = src.get("kind")
-            dtype_s = src.get("dtype", "torch.float32")
-            floaty = _dtype_from_str(dtype_s).is_floating_point \
-                if hasattr(_dtype_from_str(dtype_s), "is_floating_point") \
-                else True
+            if src.get("pyscalar"):
+                floaty = False
+            else:
+                dtype_s = src.get("dtype", "torch.float32")
+                floaty = _dtype_from_str(dtype_s).is_floating_point
             if kind == "stage":
                 j = src["stage"]
                 k = src.get("out_idx", 0)
@@ -247,8 +253,11 @@ class Node:
                 if j != self.stage:
                     ranks.append(self.stage_ranks[j])
             self._out_consumers[k] = ranks
-            dt = entry.get("dtype", "torch.float32")
-            floaty = _dtype_from_str(dt).is_floating_point
+            if entry.get("pyscalar"):
+                floaty = False
+            else:
+                floaty = _dtype_from_str(
+                    entry.get("dtype", "torch.float32")).is_floating_point
             self._expected_grad_contribs[k] = len(ranks) if floaty else 0
         self._total_expected_grads = sum(self._expected_grad_contribs.values())
 
@@ -411,7 +420,12 @@ class Node:
                 args.append(src.get("value"))
             elif pos in tensors_by_pos:
                 t = tensors_by_pos[pos]
-                args.append(t.to(self.device) if torch.is_tensor(t) else t)
+                if src.get("pyscalar") and torch.is_tensor(t):
+                    args.append(int(t.item()))
+                elif torch.is_tensor(t):
+                    args.append(t.to(self.device))
+                else:
+                    args.append(t)
             elif kind == "model_input" and model_inputs is not None:
                 args.append(model_inputs[src["name"]])
             else:
@@ -426,9 +440,15 @@ class Node:
         for k, ranks in self._out_consumers.items():
             if k >= len(outputs):
                 continue
+            out = outputs[k]
+            if torch.is_tensor(out):
+                out = out.detach()
+            else:  # routed python scalar (a .size() value): ship as int64
+                out = torch.tensor([int(out)], dtype=torch.int64,
+                                   device=self.device)
             for r in set(ranks):
                 by_rank.setdefault(r, []).append(
-                    (self.stage * _MAX_OUTS + k, outputs[k].detach()))
+                    (self.stage * _MAX_OUTS + k, out))
         for r, tensors in by_rank.items():
             self.comm.send(r, "fwd", Message(action=action, fpid=fpid,
                                              tensors=tensors, extra=extra))

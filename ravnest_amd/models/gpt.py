@@ -1,0 +1,119 @@
+"""GPT decoder family (gpt-nano ... gpt2-small) on the MI355X op library.
+
+Workload parity: the reference's minGPT sorter example
+(examples/sorter/mingpt/model_without_padding_mask.py) including the
+fx-friendly structure (custom-op modules are fx leaves — the reference
+patched arange/masked_fill into leaf modules for the same reason,
+model_without_padding_mask.py:34-48). Causal attention runs on the fused
+CDNA4 attention kernel.
+"""
+from __future__ import annotations
+
+import torch
+import torch.nn as nn
+
+from ..ops import AttentionCore, Dropout, FusedLayerNorm, LinearGelu
+
+
+class GPTConfig:
+    def __init__(self, vocab_size, block_size, n_layer, n_head, n_embd,
+                 dropout=0.1):
+        self.vocab_size = vocab_size
+        self.block_size = block_size
+        self.n_layer = n_layer
+        self.n_head = n_head
+        self.n_embd = n_embd
+        self.dropout = dropout
+
+    @classmethod
+    def nano(cls, vocab_size=16, block_size=16):
+        return cls(vocab_size, block_size, n_layer=3, n_head=3, n_embd=48)
+
+    @classmethod
+    def gpt2_small(cls, vocab_size=50257, block_size=1024):
+        return cls(vocab_size, block_size, n_layer=12, n_head=12, n_embd=768)
+
+
+class GPTEmbeddings(nn.Module):
+    _is_leaf_module = True  # fx: data-dependent position slice
+
+    def __init__(self, cfg: GPTConfig):
+        super().__init__()
+        self.tok_emb = nn.Embedding(cfg.vocab_size, cfg.n_embd)
+        self.pos_emb = nn.Embedding(cfg.block_size, cfg.n_embd)
+        self.drop = Dropout(cfg.dropout)
+        self.register_buffer("pos_ids",
+                             torch.arange(cfg.block_size).unsqueeze(0),
+                             persistent=False)
+
+    def forward(self, idx):
+        S = idx.size(1)
+        x = self.tok_emb(idx) + self.pos_emb(self.pos_ids[:, :S])
+        return self.drop(x)
+
+
+class GPTBlock(nn.Module):
+    def __init__(self, cfg: GPTConfig):
+        super().__init__()
+        self.n_head = cfg.n_head
+        self.head_dim = cfg.n_embd // cfg.n_head
+        self.ln1 = FusedLayerNorm(cfg.n_embd)
+        self.qkv = nn.Linear(cfg.n_embd, 3 * cfg.n_embd)
+        self.core = AttentionCore(causal=True)
+        self.proj = nn.Linear(cfg.n_embd, cfg.n_embd)
+        self.attn_drop = Dropout(cfg.dropout)
+        self.ln2 = FusedLayerNorm(cfg.n_embd)
+        self.mlp_in = LinearGelu(cfg.n_embd, 4 * cfg.n_embd)
+        self.mlp_out = nn.Linear(4 * cfg.n_embd, cfg.n_embd)
+        self.mlp_drop = Dropout(cfg.dropout)
+
+    def forward(self, x):
+        h = self.ln1(x)
+        qkv = self.qkv(h).unflatten(-1, (3, self.n_head, self.head_dim))
+        qkv = qkv.permute(2, 0, 3, 1, 4)
+        o = self.core(qkv[0], qkv[1], qkv[2])
+        o = o.transpose(1, 2).flatten(2)
+        x = x + self.attn_drop(self.proj(o))
+        x = x + self.mlp_drop(self.mlp_out(self.mlp_in(self.ln2(x))))
+        return x
+
+
+class GPT(nn.Module):
+    """idx (B,S) int64 -> logits (B,S,V)."""
+
+    def __init__(self, cfg: GPTConfig):
+        super().__init__()
+        self.cfg = cfg
+        self.embeddings = GPTEmbeddings(cfg)
+        self.blocks = nn.ModuleList(
+            [GPTBlock(cfg) for _ in range(cfg.n_layer)])
+        self.ln_f = FusedLayerNorm(cfg.n_embd)
+        self.lm_head = nn.Linear(cfg.n_embd, cfg.vocab_size, bias=False)
+        self.apply(self._init)
+
+    @staticmethod
+    def _init(m):
+        if isinstance(m, nn.Linear):
+            nn.init.normal_(m.weight, std=0.02)
+            if m.bias is not None:
+                nn.init.zeros_(m.bias)
+        elif isinstance(m, nn.Embedding):
+            nn.init.normal_(m.weight, std=0.02)
+
+    def forward(self, idx):
+        x = self.embeddings(idx)
+        for b in self.blocks:
+            x = b(x)
+        return self.lm_head(self.ln_f(x))
+
+    @torch.no_grad()
+    def generate(self, idx, max_new_tokens):
+        """Greedy decode (parity: sorter inference,
+        examples/sorter/mingpt/utils.py)."""
+        self.eval()
+        for _ in range(max_new_tokens):
+            idx_cond = idx[:, -self.cfg.block_size:]
+            logits = self(idx_cond)
+            nxt = logits[:, -1, :].argmax(dim=-1, keepdim=True)
+            idx = torch.cat([idx, nxt], dim=1)
+        return idx

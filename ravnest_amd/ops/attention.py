@@ -1,0 +1,94 @@
+"""Fused multi-head attention core (flash-style forward on CDNA4).
+
+Reference workload parity: minGPT causal self-attention
+(model_without_padding_mask.py:50-113) and BERT's padded-mask attention —
+SURVEY.md section 2.3. Forward is the hand-written HIP kernel
+(csrc/attention.hip): per Q-tile online-softmax over K/V tiles, bf16 MFMA
+(32x32x16), XOR-swizzled K LDS tiles, no S x S score matrix in HBM; it
+returns O and the log-sum-exp rows. Backward recomputes P from (Q,K,lse)
+with plain hipBLASLt GEMMs + elementwise torch (library GEMMs are the
+sanctioned path for non-fused matmuls; the fused bwd kernel is tracked
+for a later round).
+
+Shapes: q,k,v (B, H, S, D); additive mask broadcastable to (B, 1, S, S)
+or None; causal flag for GPT-style models.
+"""
+from __future__ import annotations
+
+import math
+
+import torch
+import torch.nn as nn
+
+from ._ext import get_ext
+
+
+def _math_attention(q, k, v, mask, causal, scale):
+    s = (q @ k.transpose(-2, -1)) * scale
+    if causal:
+        S = q.shape[-2]
+        cm = torch.full((S, S), float("-inf"), device=q.device)
+        cm = torch.triu(cm, diagonal=1)
+        s = s + cm
+    if mask is not None:
+        s = s + mask
+    p = torch.softmax(s.float(), dim=-1).to(q.dtype)
+    return p @ v
+
+
+class _AttnFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, q, k, v, mask, causal, scale):
+        ext = get_ext(required=True)
+        o, lse = ext.attn_fwd(q, k, v,
+                              mask if mask is not None else torch.Tensor(),
+                              causal, scale)
+        ctx.save_for_backward(q, k, v, mask if mask is not None else None,
+                              o, lse)
+        ctx.causal = causal
+        ctx.scale = scale
+        return o
+
+    @staticmethod
+    def backward(ctx, do):
+        q, k, v, mask, o, lse = ctx.saved_tensors
+        causal, scale = ctx.causal, ctx.scale
+        do = do.contiguous()
+        # recompute P from lse: P = exp(S*scale + mask - lse)
+        s = (q @ k.transpose(-2, -1)).float() * scale
+        if causal:
+            S = q.shape[-2]
+            cm = torch.triu(torch.full((S, S), float("-inf"),
+                                       device=q.device), diagonal=1)
+            s = s + cm
+        if mask is not None:
+            s = s + mask.float()
+        p = torch.exp(s - lse.unsqueeze(-1)).to(q.dtype)
+        dv = p.transpose(-2, -1) @ do
+        dp = (do @ v.transpose(-2, -1)).float()
+        delta = (do.float() * o.float()).sum(-1, keepdim=True)
+        ds = (p.float() * (dp - delta)) * scale
+        ds = ds.to(q.dtype)
+        dq = ds @ k
+        dk = ds.transpose(-2, -1) @ q
+        return dq, dk, dv, None, None, None
+
+
+def attention(q, k, v, mask=None, causal=False, scale=None):
+    if scale is None:
+        scale = 1.0 / math.sqrt(q.shape[-1])
+    if q.is_cuda:
+        return _AttnFn.apply(q.contiguous(), k.contiguous(), v.contiguous(),
+                             mask, causal, scale)
+    return _math_attention(q, k, v, mask, causal, scale)
+
+
+class AttentionCore(nn.Module):
+    _is_leaf_module = True
+
+    def __init__(self, causal: bool = False):
+        super().__init__()
+        self.causal = causal
+
+    def forward(self, q, k, v, mask=None):
+        return attention(q, k, v, mask=mask, causal=self.causal)

@@ -399,12 +399,22 @@ def _extract_routing(split_gm: fx.GraphModule, n_stages: int) -> SplitResult | N
                 return None
 
     # annotate output entries with dtype/shape from the producing submod
+    sp_ok = any(n.meta.get("tensor_meta") is not None
+                for n in split_gm.graph.nodes)
     for j, name in enumerate(stage_order):
         for k, entry in stage_outputs[j].items():
             _annot(entry, _meta_of(submod_nodes[name],
                                    out_idx=None if n_stage_outputs[j] == 1
                                    else k,
                                    whole=n_stage_outputs[j] == 1))
+            if sp_ok and "dtype" not in entry:
+                entry["pyscalar"] = True  # a routed .size()/python scalar
+    if sp_ok:
+        for ins in stage_inputs:
+            for src in ins:
+                if src.get("kind") in ("stage", "model_input") and \
+                        "dtype" not in src:
+                    src["pyscalar"] = True
 
     stages = [split_gm.get_submodule(name) for name in stage_order]
     return SplitResult(

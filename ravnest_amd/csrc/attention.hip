@@ -111,18 +111,22 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
     tile_max = fmaxf(tile_max, __shfl_xor(tile_max, 32, WAVE));
 
     const float m_new = fmaxf(m_run, tile_max);
+    // alpha is per Q-ROW. This lane's softmax state tracks qrow = j32,
+    // but the PV accumulator's rows follow the MFMA C layout
+    // (row = (r&3)+8*(r>>2)+4*hi), so the O rescale needs each row's
+    // alpha broadcast from the lane that owns it.
     float alpha = 1.f;
     if (m_new != m_run) {
       alpha = (m_run == -INFINITY) ? 0.f : __expf(m_run - m_new);
       m_run = m_new;
     }
-    // rescale O and l
-    if (alpha != 1.f) {
+    l_run *= alpha;
 #pragma unroll
-      for (int h = 0; h < DHALF; ++h)
+    for (int r = 0; r < 16; ++r) {
+      const int row_local = (r & 3) + 8 * (r >> 2) + 4 * hi;
+      const float a_r = __shfl(alpha, row_local, WAVE);
 #pragma unroll
-        for (int r = 0; r < 16; ++r) oacc[h][r] *= alpha;
-      l_run *= alpha;
+      for (int h = 0; h < DHALF; ++h) oacc[h][r] *= a_r;
     }
 
     // ---- exponentiate + row-sum ----

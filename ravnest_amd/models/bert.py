@@ -37,8 +37,9 @@ class BertConfig:
 
     @classmethod
     def tiny(cls, **kw):
-        d = dict(vocab_size=1024, hidden=64, layers=2, heads=2,
-                 intermediate=128, max_seq=64)
+        # head_dim 64 (the attention kernel's native size)
+        d = dict(vocab_size=1024, hidden=128, layers=2, heads=2,
+                 intermediate=256, max_seq=64)
         d.update(kw)
         return cls(**d)
 

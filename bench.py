@@ -206,7 +206,7 @@ def main():
     if comm is not None:
         comm.stop()
         torch.distributed.barrier()
-    os._exit(0)  # listener threads may be parked in recv; exit hard
+        torch.distributed.destroy_process_group()
 
 
 if __name__ == "__main__":

@@ -18,15 +18,16 @@ at::Tensor ce_bwd(at::Tensor logits, at::Tensor targets, at::Tensor lse,
                   at::Tensor gscale, int64_t ignore_index);
 void fused_adam(std::vector<at::Tensor> ps, std::vector<at::Tensor> gs,
                 std::vector<at::Tensor> ms, std::vector<at::Tensor> vs,
-                double lr, double b1, double b2, double eps, double wd,
-                double bc1, double bc2);
+                std::vector<at::Tensor> masters, double lr, double b1,
+                double b2, double eps, double wd, double bc1, double bc2);
 void fused_sgd(std::vector<at::Tensor> ps, std::vector<at::Tensor> gs,
-               std::vector<at::Tensor> bufs, double lr, double momentum,
-               double wd, bool nesterov);
+               std::vector<at::Tensor> bufs, std::vector<at::Tensor> masters,
+               double lr, double momentum, double wd, bool nesterov);
 void fused_lamb(std::vector<at::Tensor> ps, std::vector<at::Tensor> gs,
                 std::vector<at::Tensor> ms, std::vector<at::Tensor> vs,
-                double lr, double b1, double b2, double eps, double wd,
-                double bc1, double bc2, double clamp_trust);
+                std::vector<at::Tensor> masters, double lr, double b1,
+                double b2, double eps, double wd, double bc1, double bc2,
+                double clamp_trust);
 std::vector<at::Tensor> attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v,
                                  at::Tensor mask, bool causal, double scale);
 at::Tensor mfma_probe(at::Tensor a, at::Tensor b);

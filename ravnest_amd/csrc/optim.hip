@@ -3,8 +3,13 @@
 // ONE kernel launch applies the update across every parameter tensor
 // (apex-style chunking: host builds a chunk table, each 256-thread block
 // owns one chunk) — SURVEY.md section 2.3 "fused multi-tensor apply
-// kernels (one launch per bucket), incl. LAMB trust-ratio". fp32 master
-// params/grads/states.
+// kernels (one launch per bucket), incl. LAMB trust-ratio".
+//
+// Two parameter modes:
+//  * fp32 params + fp32 grads (CPU-parity training)
+//  * bf16 params + bf16 grads with an fp32 MASTER copy in optimizer
+//    state (the MI355X-native mode: bf16 weights halve HBM traffic for
+//    every GEMM; the master preserves convergence).
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
 
@@ -17,45 +22,69 @@ namespace {
 constexpr int CHUNK = 1 << 16;
 
 struct Chunk {
-  float* p;
-  float* g;
+  void* p;        // param (fp32 or bf16)
+  void* g;        // grad  (same dtype as p)
   float* m;
   float* v;
+  float* master;  // fp32 master (bf16 mode) or nullptr
   int n;
   int tensor_idx;
 };
 
+template <typename T>
+DEVINL float ldv(const void* p, int i) {
+  if constexpr (sizeof(T) == 2)
+    return bf2f(reinterpret_cast<const bf16_t*>(p)[i]);
+  else
+    return reinterpret_cast<const float*>(p)[i];
+}
+
+template <typename T>
+DEVINL void stv(void* p, int i, float v) {
+  if constexpr (sizeof(T) == 2)
+    reinterpret_cast<bf16_t*>(p)[i] = f2bf(v);
+  else
+    reinterpret_cast<float*>(p)[i] = v;
+}
+
+template <typename T>
 __global__ void adam_mt_kernel(const Chunk* __restrict__ chunks, float lr,
                                float b1, float b2, float eps, float wd,
                                float bc1, float bc2) {
   const Chunk c = chunks[blockIdx.x];
   for (int i = threadIdx.x; i < c.n; i += blockDim.x) {
-    float g = c.g[i];
-    float p = c.p[i];
+    float g = ldv<T>(c.g, i);
+    float p = c.master ? c.master[i] : ldv<T>(c.p, i);
     if (wd != 0.f) g += wd * p;
     float m = c.m[i] = b1 * c.m[i] + (1.f - b1) * g;
     float v = c.v[i] = b2 * c.v[i] + (1.f - b2) * g * g;
     const float denom = sqrtf(v / bc2) + eps;
-    c.p[i] = p - lr * (m / bc1) / denom;
+    p -= lr * (m / bc1) / denom;
+    if (c.master) c.master[i] = p;
+    stv<T>(c.p, i, p);
   }
 }
 
+template <typename T>
 __global__ void sgd_mt_kernel(const Chunk* __restrict__ chunks, float lr,
                               float momentum, float wd, int nesterov) {
   const Chunk c = chunks[blockIdx.x];
   for (int i = threadIdx.x; i < c.n; i += blockDim.x) {
-    float g = c.g[i];
-    float p = c.p[i];
+    float g = ldv<T>(c.g, i);
+    float p = c.master ? c.master[i] : ldv<T>(c.p, i);
     if (wd != 0.f) g += wd * p;
     if (momentum != 0.f) {
       float b = c.m[i] = momentum * c.m[i] + g;
       g = nesterov ? g + momentum * b : b;
     }
-    c.p[i] = p - lr * g;
+    p -= lr * g;
+    if (c.master) c.master[i] = p;
+    stv<T>(c.p, i, p);
   }
 }
 
 // LAMB phase 1: update m/v, accumulate ||w||^2 and ||update||^2 per tensor
+template <typename T>
 __global__ void lamb_phase1_kernel(const Chunk* __restrict__ chunks,
                                    float* __restrict__ norms,  // [T][2]
                                    float b1, float b2, float eps, float wd,
@@ -64,8 +93,8 @@ __global__ void lamb_phase1_kernel(const Chunk* __restrict__ chunks,
   const Chunk c = chunks[blockIdx.x];
   float wsq = 0.f, usq = 0.f;
   for (int i = threadIdx.x; i < c.n; i += blockDim.x) {
-    float g = c.g[i];
-    float p = c.p[i];
+    float g = ldv<T>(c.g, i);
+    float p = c.master ? c.master[i] : ldv<T>(c.p, i);
     float m = c.m[i] = b1 * c.m[i] + (1.f - b1) * g;
     float v = c.v[i] = b2 * c.v[i] + (1.f - b2) * g * g;
     float upd = (m / bc1) / (sqrtf(v / bc2) + eps);
@@ -81,6 +110,7 @@ __global__ void lamb_phase1_kernel(const Chunk* __restrict__ chunks,
   }
 }
 
+template <typename T>
 __global__ void lamb_phase2_kernel(const Chunk* __restrict__ chunks,
                                    const float* __restrict__ norms, float lr,
                                    float b1, float b2, float eps, float wd,
@@ -93,33 +123,48 @@ __global__ void lamb_phase2_kernel(const Chunk* __restrict__ chunks,
   const float step = lr * trust;
   for (int i = threadIdx.x; i < c.n; i += blockDim.x) {
     float m = c.m[i], v = c.v[i];
+    float p = c.master ? c.master[i] : ldv<T>(c.p, i);
     float upd = (m / bc1) / (sqrtf(v / bc2) + eps);
-    if (wd != 0.f) upd += wd * c.p[i];
-    c.p[i] -= step * upd;
+    if (wd != 0.f) upd += wd * p;
+    p -= step * upd;
+    if (c.master) c.master[i] = p;
+    stv<T>(c.p, i, p);
   }
 }
 
-// build the device-side chunk table
+// build the device-side chunk table; returns element dtype size (2|4)
 at::Tensor build_chunks(const std::vector<at::Tensor>& ps,
                         const std::vector<at::Tensor>& gs,
                         const std::vector<at::Tensor>& ms,
-                        const std::vector<at::Tensor>& vs, int& nchunks) {
+                        const std::vector<at::Tensor>& vs,
+                        const std::vector<at::Tensor>& masters, int& nchunks,
+                        int& esize) {
   std::vector<Chunk> chunks;
+  esize = ps[0].scalar_type() == at::kBFloat16 ? 2 : 4;
   for (size_t t = 0; t < ps.size(); ++t) {
-    TORCH_CHECK(ps[t].scalar_type() == at::kFloat,
-                "fused optimizers expect fp32 master params");
     TORCH_CHECK(ps[t].is_contiguous() && gs[t].is_contiguous());
+    TORCH_CHECK(gs[t].scalar_type() == ps[t].scalar_type(),
+                "grad dtype must match param dtype");
     long n = ps[t].numel();
-    float* p = ps[t].data_ptr<float>();
-    float* g = gs[t].data_ptr<float>();
-    float* m = ms.empty() ? nullptr : ms[t].data_ptr<float>();
-    float* v = vs.empty() ? nullptr : vs[t].data_ptr<float>();
+    void* p = ps[t].data_ptr();
+    void* g = gs[t].data_ptr();
+    float* m = (ms.empty() || ms[t].numel() == 0)
+                   ? nullptr : ms[t].data_ptr<float>();
+    float* v = (vs.empty() || vs[t].numel() == 0)
+                   ? nullptr : vs[t].data_ptr<float>();
+    float* master = masters.empty() ? nullptr
+                                    : masters[t].data_ptr<float>();
+    const int es = ps[t].scalar_type() == at::kBFloat16 ? 2 : 4;
+    TORCH_CHECK(es == esize, "mixed param dtypes in one group");
+    TORCH_CHECK(es == 4 || master != nullptr,
+                "bf16 params require fp32 masters");
     for (long off = 0; off < n; off += CHUNK) {
       Chunk c;
-      c.p = p + off;
-      c.g = g + off;
+      c.p = (char*)p + off * es;
+      c.g = (char*)g + off * es;
       c.m = m ? m + off : nullptr;
       c.v = v ? v + off : nullptr;
+      c.master = master ? master + off : nullptr;
       c.n = (int)std::min<long>(CHUNK, n - off);
       c.tensor_idx = (int)t;
       chunks.push_back(c);
@@ -136,51 +181,75 @@ at::Tensor build_chunks(const std::vector<at::Tensor>& ps,
 
 void fused_adam(std::vector<at::Tensor> ps, std::vector<at::Tensor> gs,
                 std::vector<at::Tensor> ms, std::vector<at::Tensor> vs,
-                double lr, double b1, double b2, double eps, double wd,
-                double bc1, double bc2) {
+                std::vector<at::Tensor> masters, double lr, double b1,
+                double b2, double eps, double wd, double bc1, double bc2) {
   if (ps.empty()) return;
-  int nchunks = 0;
-  auto dev_chunks = build_chunks(ps, gs, ms, vs, nchunks);
+  int nchunks = 0, esize = 4;
+  auto dev_chunks = build_chunks(ps, gs, ms, vs, masters, nchunks, esize);
   auto stream = c10::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
-  hipLaunchKernelGGL(adam_mt_kernel, dim3(nchunks), dim3(256), 0, stream,
-                     reinterpret_cast<const Chunk*>(dev_chunks.data_ptr()),
-                     (float)lr, (float)b1, (float)b2, (float)eps, (float)wd,
-                     (float)bc1, (float)bc2);
+  if (esize == 2)
+    hipLaunchKernelGGL((adam_mt_kernel<bf16_t>), dim3(nchunks), dim3(256), 0,
+                       stream,
+                       reinterpret_cast<const Chunk*>(dev_chunks.data_ptr()),
+                       (float)lr, (float)b1, (float)b2, (float)eps, (float)wd,
+                       (float)bc1, (float)bc2);
+  else
+    hipLaunchKernelGGL((adam_mt_kernel<float>), dim3(nchunks), dim3(256), 0,
+                       stream,
+                       reinterpret_cast<const Chunk*>(dev_chunks.data_ptr()),
+                       (float)lr, (float)b1, (float)b2, (float)eps, (float)wd,
+                       (float)bc1, (float)bc2);
   HIP_CHECK_LAST();
 }
 
 void fused_sgd(std::vector<at::Tensor> ps, std::vector<at::Tensor> gs,
-               std::vector<at::Tensor> bufs, double lr, double momentum,
-               double wd, bool nesterov) {
+               std::vector<at::Tensor> bufs, std::vector<at::Tensor> masters,
+               double lr, double momentum, double wd, bool nesterov) {
   if (ps.empty()) return;
-  int nchunks = 0;
-  auto dev_chunks = build_chunks(ps, gs, bufs, {}, nchunks);
+  int nchunks = 0, esize = 4;
+  auto dev_chunks = build_chunks(ps, gs, bufs, {}, masters, nchunks, esize);
   auto stream = c10::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
-  hipLaunchKernelGGL(sgd_mt_kernel, dim3(nchunks), dim3(256), 0, stream,
-                     reinterpret_cast<const Chunk*>(dev_chunks.data_ptr()),
-                     (float)lr, (float)momentum, (float)wd,
-                     nesterov ? 1 : 0);
+  if (esize == 2)
+    hipLaunchKernelGGL((sgd_mt_kernel<bf16_t>), dim3(nchunks), dim3(256), 0,
+                       stream,
+                       reinterpret_cast<const Chunk*>(dev_chunks.data_ptr()),
+                       (float)lr, (float)momentum, (float)wd, nesterov);
+  else
+    hipLaunchKernelGGL((sgd_mt_kernel<float>), dim3(nchunks), dim3(256), 0,
+                       stream,
+                       reinterpret_cast<const Chunk*>(dev_chunks.data_ptr()),
+                       (float)lr, (float)momentum, (float)wd, nesterov);
   HIP_CHECK_LAST();
 }
 
 void fused_lamb(std::vector<at::Tensor> ps, std::vector<at::Tensor> gs,
                 std::vector<at::Tensor> ms, std::vector<at::Tensor> vs,
-                double lr, double b1, double b2, double eps, double wd,
-                double bc1, double bc2, double clamp_trust) {
+                std::vector<at::Tensor> masters, double lr, double b1,
+                double b2, double eps, double wd, double bc1, double bc2,
+                double clamp_trust) {
   if (ps.empty()) return;
-  int nchunks = 0;
-  auto dev_chunks = build_chunks(ps, gs, ms, vs, nchunks);
+  int nchunks = 0, esize = 4;
+  auto dev_chunks = build_chunks(ps, gs, ms, vs, masters, nchunks, esize);
   auto norms = at::zeros({(long)ps.size(), 2},
                          ps[0].options().dtype(at::kFloat));
   auto stream = c10::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
-  hipLaunchKernelGGL(lamb_phase1_kernel, dim3(nchunks), dim3(256), 0, stream,
-                     reinterpret_cast<const Chunk*>(dev_chunks.data_ptr()),
-                     norms.data_ptr<float>(), (float)b1, (float)b2,
-                     (float)eps, (float)wd, (float)bc1, (float)bc2);
-  hipLaunchKernelGGL(lamb_phase2_kernel, dim3(nchunks), dim3(256), 0, stream,
-                     reinterpret_cast<const Chunk*>(dev_chunks.data_ptr()),
-                     norms.data_ptr<float>(), (float)lr, (float)b1, (float)b2,
-                     (float)eps, (float)wd, (float)bc1, (float)bc2,
-                     (float)clamp_trust);
+#define LAMB_LAUNCH(T)                                                       \
+  do {                                                                       \
+    hipLaunchKernelGGL((lamb_phase1_kernel<T>), dim3(nchunks), dim3(256), 0, \
+                       stream,                                               \
+                       reinterpret_cast<const Chunk*>(dev_chunks.data_ptr()),\
+                       norms.data_ptr<float>(), (float)b1, (float)b2,        \
+                       (float)eps, (float)wd, (float)bc1, (float)bc2);       \
+    hipLaunchKernelGGL((lamb_phase2_kernel<T>), dim3(nchunks), dim3(256), 0, \
+                       stream,                                               \
+                       reinterpret_cast<const Chunk*>(dev_chunks.data_ptr()),\
+                       norms.data_ptr<float>(), (float)lr, (float)b1,        \
+                       (float)b2, (float)eps, (float)wd, (float)bc1,         \
+                       (float)bc2, (float)clamp_trust);                      \
+  } while (0)
+  if (esize == 2)
+    LAMB_LAUNCH(bf16_t);
+  else
+    LAMB_LAUNCH(float);
   HIP_CHECK_LAST();
 }

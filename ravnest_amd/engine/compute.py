@@ -48,7 +48,8 @@ class ComputeEngine:
                  update_frequency: int = 1,
                  criterion=None,
                  loss_filename: str = "losses.txt",
-                 amp_dtype: torch.dtype | None = None):
+                 amp_dtype: torch.dtype | None = None,
+                 versioning: bool = True):
         self.model = model
         self.optimizer = optimizer
         self.device = device
@@ -56,6 +57,9 @@ class ComputeEngine:
         self.criterion = criterion
         self.loss_filename = loss_filename
         self.amp_dtype = amp_dtype
+        # versioning=False (single-fused-stage training) skips the
+        # per-step parameter snapshot clones: no recompute ever happens
+        self.versioning = versioning
 
         self.current_version = 0
         self.version_to_param: dict[int, list[torch.Tensor]] = {}
@@ -79,6 +83,8 @@ class ComputeEngine:
     # version bookkeeping
     # ------------------------------------------------------------------
     def _snapshot_current(self):
+        if not self.versioning:
+            return
         self.version_to_param[self.current_version] = [
             p.detach().clone() for p in self._params]
         self.version_refs.setdefault(self.current_version, 0)
@@ -87,6 +93,9 @@ class ComputeEngine:
         """Latest parameter snapshot for peers to pull (parity:
         latest_weights_buffer, reference compute.py:47-51)."""
         with self._lock:
+            if not self.versioning:
+                return {"version": self.current_version,
+                        "params": [p.detach().clone() for p in self._params]}
             return {
                 "version": self.current_version,
                 "params": [t.clone() for t in
@@ -111,6 +120,8 @@ class ComputeEngine:
         weight pull)."""
         old = self.current_version
         self.current_version += 1
+        if not self.versioning:
+            return
         self._snapshot_current()
         if self.version_refs.get(old, 0) == 0:
             self.version_refs.pop(old, None)
@@ -229,8 +240,8 @@ class ComputeEngine:
         if self.optimizer is not None and \
                 self.n_backwards % self.update_frequency == 0:
             self.optimizer.step()
-            self.optimizer.zero_grad(set_to_none=False)
-            self.model.zero_grad(set_to_none=False)
+            self.optimizer.zero_grad(set_to_none=True)
+            self.model.zero_grad(set_to_none=True)
             self.bump_version()
             stepped = True
         return stepped
@@ -269,8 +280,8 @@ class ComputeEngine:
         if self.optimizer is not None and \
                 self.n_backwards % self.update_frequency == 0:
             self.optimizer.step()
-            self.optimizer.zero_grad(set_to_none=False)
-            self.model.zero_grad(set_to_none=False)
+            self.optimizer.zero_grad(set_to_none=True)
+            self.model.zero_grad(set_to_none=True)
             self.bump_version()
             stepped = True
             if self.loss_filename:

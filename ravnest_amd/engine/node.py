@@ -159,7 +159,8 @@ class Node:
                                     update_frequency=update_frequency,
                                     criterion=criterion,
                                     loss_filename=loss_filename,
-                                    amp_dtype=amp_dtype)
+                                    amp_dtype=amp_dtype,
+                                    versioning=not self.fused)
 
         # ---- routing precomputation ---------------------------------
         self._build_routing()

@@ -41,6 +41,8 @@ class FusedAdam(Optimizer):
                     st["step"] = 0
                     st["m"] = torch.zeros_like(p, dtype=torch.float32)
                     st["v"] = torch.zeros_like(p, dtype=torch.float32)
+                    if p.dtype == torch.bfloat16 and p.is_cuda:
+                        st["master"] = p.detach().float().clone()
                 st["step"] += 1
                 ps.append(p)
                 gs.append(p.grad)
@@ -54,7 +56,9 @@ class FusedAdam(Optimizer):
             bc2 = 1 - b2 ** step
             if _on_gpu(ps):
                 ext = get_ext(required=True)
-                ext.fused_adam(ps, gs, ms, vs, group["lr"], b1, b2,
+                masters = ([self.state[p]["master"] for p in ps]
+                           if ps[0].dtype == torch.bfloat16 else [])
+                ext.fused_adam(ps, gs, ms, vs, masters, group["lr"], b1, b2,
                                group["eps"], group["weight_decay"], bc1, bc2)
             else:
                 for p, g, m, v in zip(ps, gs, ms, vs):
@@ -89,14 +93,20 @@ class FusedSGD(Optimizer):
                 if mom and "momentum_buffer" not in st:
                     st["momentum_buffer"] = torch.zeros_like(
                         p, dtype=torch.float32)
+                if p.dtype == torch.bfloat16 and p.is_cuda and \
+                        "master" not in st:
+                    st["master"] = p.detach().float().clone()
                 ps.append(p)
                 gs.append(p.grad)
-                bufs.append(st["momentum_buffer"] if mom else p.grad)
+                bufs.append(st["momentum_buffer"] if mom
+                            else torch.zeros(0))
             if not ps:
                 continue
             if _on_gpu(ps):
                 ext = get_ext(required=True)
-                ext.fused_sgd(ps, gs, bufs, group["lr"], mom,
+                masters = ([self.state[p]["master"] for p in ps]
+                           if ps[0].dtype == torch.bfloat16 else [])
+                ext.fused_sgd(ps, gs, bufs, masters, group["lr"], mom,
                               group["weight_decay"],
                               bool(group["nesterov"]))
             else:
@@ -136,6 +146,8 @@ class FusedLAMB(Optimizer):
                     st["step"] = 0
                     st["m"] = torch.zeros_like(p, dtype=torch.float32)
                     st["v"] = torch.zeros_like(p, dtype=torch.float32)
+                    if p.dtype == torch.bfloat16 and p.is_cuda:
+                        st["master"] = p.detach().float().clone()
                 st["step"] += 1
                 ps.append(p)
                 gs.append(p.grad)
@@ -148,7 +160,9 @@ class FusedLAMB(Optimizer):
             bc2 = 1 - b2 ** step
             if _on_gpu(ps):
                 ext = get_ext(required=True)
-                ext.fused_lamb(ps, gs, ms, vs, group["lr"], b1, b2,
+                masters = ([self.state[p]["master"] for p in ps]
+                           if ps[0].dtype == torch.bfloat16 else [])
+                ext.fused_lamb(ps, gs, ms, vs, masters, group["lr"], b1, b2,
                                group["eps"], group["weight_decay"], bc1, bc2,
                                group["clamp_trust"])
             else:

@@ -249,3 +249,30 @@ def test_attention_autograd(dev):
                          (v.grad, v2.grad, "dv")]:
         err = (g1.float() - g2).abs().max().item()
         assert err < 8e-2, f"{name} max err {err}"
+
+
+def test_fused_adam_bf16_master(dev):
+    """bf16 params + bf16 grads with fp32 master vs fp32 reference."""
+    from ravnest_amd.ops import FusedAdam
+    torch.manual_seed(0)
+    ref = [torch.randn(256, 256, device=dev) for _ in range(3)]
+    p_bf = [r.to(torch.bfloat16).requires_grad_(True) for r in ref]
+    p_fp = [r.to(torch.bfloat16).float().requires_grad_(True) for r in ref]
+    o1 = FusedAdam(p_bf, lr=1e-2)
+    o2 = torch.optim.Adam(p_fp, lr=1e-2)
+    for step in range(10):
+        g = [torch.randn_like(r) for r in ref]
+        for p, gg in zip(p_bf, g):
+            p.grad = gg.to(torch.bfloat16)
+        for p, gg in zip(p_fp, g):
+            p.grad = gg.to(torch.bfloat16).float()  # same quantized grads
+        o1.step()
+        o2.step()
+    for a, b in zip(p_bf, p_fp):
+        # bf16 param must equal the bf16-rounded fp32 trajectory
+        assert torch.allclose(a.float(), b.to(torch.bfloat16).float(),
+                              atol=1e-2, rtol=1e-2)
+    # masters track the full-precision trajectory
+    for p, b in zip(p_bf, p_fp):
+        m = o1.state[p]["master"]
+        assert torch.allclose(m, b, atol=1e-4, rtol=1e-3)

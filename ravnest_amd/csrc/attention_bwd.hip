@@ -1,0 +1,369 @@
+// Flash-style fused attention BACKWARD for CDNA4 (gfx950), bf16, D=64.
+//
+// Three kernels (flash2-style split, no atomics):
+//   attn_bwd_delta: Delta[row] = dO[row] . O[row]            (rowwise)
+//   attn_bwd_dkv:   each wave owns a 32-KEY tile, loops Q tiles,
+//                   recomputes P from (Q,K,lse), accumulates dV,dK
+//   attn_bwd_dq:    each wave owns a 32-ROW Q tile, loops KV tiles,
+//                   recomputes P^T, accumulates dQ
+//
+// Both matmul kernels reuse the forward's fragment algebra
+// (attention.hip header comment): the MFMA C layout [row-pattern][col =
+// lane&31] is turned into the next MFMA's A operand by bf16 pair packing
+// + __builtin_amdgcn_permlane32_swap (guide T12/T21 primitives).
+//
+// dS_ij = P_ij * (dP_ij - Delta_i) * scale;  dV = P^T dO; dK = dS^T Q;
+// dQ = dS K.
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+#include "common.h"
+
+typedef __bf16 bf16x8v __attribute__((ext_vector_type(8)));
+typedef float f32x16 __attribute__((ext_vector_type(16)));
+
+namespace {
+
+DEVINL unsigned int pack_bf2b(float lo, float hi) {
+  return (unsigned int)f2us(lo) | ((unsigned int)f2us(hi) << 16);
+}
+
+// transform a [row-pattern][col=lane] f32 acc (16 regs) into the A-operand
+// fragments covering k=rows: pa[step] holds rows 16*step + (lane>>5)*8+e.
+DEVINL void acc_to_afrag(const f32x16& acc, unsigned int pa[2][4]) {
+#pragma unroll
+  for (int step = 0; step < 2; ++step) {
+    const int b0 = step * 8;
+    unsigned int c01 = pack_bf2b(acc[b0 + 0], acc[b0 + 1]);
+    unsigned int c23 = pack_bf2b(acc[b0 + 2], acc[b0 + 3]);
+    unsigned int c45 = pack_bf2b(acc[b0 + 4], acc[b0 + 5]);
+    unsigned int c67 = pack_bf2b(acc[b0 + 6], acc[b0 + 7]);
+    {
+      auto r2 = __builtin_amdgcn_permlane32_swap(c01, c45, false, false);
+      pa[step][0] = r2[0];
+      pa[step][2] = r2[1];
+    }
+    {
+      auto r2 = __builtin_amdgcn_permlane32_swap(c23, c67, false, false);
+      pa[step][1] = r2[0];
+      pa[step][3] = r2[1];
+    }
+  }
+}
+
+// Delta[row] = sum_d dO[row][d] * O[row][d]; one wave per row (D<=128)
+template <int D>
+__global__ void attn_bwd_delta_kernel(const bf16_t* __restrict__ dout,
+                                      const bf16_t* __restrict__ o,
+                                      float* __restrict__ delta, long NR) {
+  const int lane = threadIdx.x & (WAVE - 1);
+  const long row = (long)blockIdx.x * (blockDim.x / WAVE) +
+                   threadIdx.x / WAVE;
+  if (row >= NR) return;
+  float s = 0.f;
+#pragma unroll
+  for (int d = lane; d < D; d += WAVE)
+    s += bf2f(dout[row * D + d]) * bf2f(o[row * D + d]);
+  s = wave_sum(s);
+  if (lane == 0) delta[row] = s;
+}
+
+// -------------------------------------------------------------------
+// dkv kernel: wave owns keys [k0, k0+32); loops q tiles.
+// acc layouts: S/dP = [qrow-pattern][key=lane&31]
+//              dV/dK = [key-pattern][d = half*32 + lane&31]
+// -------------------------------------------------------------------
+__global__ __launch_bounds__(256) void attn_bwd_dkv_kernel(
+    const bf16_t* __restrict__ q, const bf16_t* __restrict__ k,
+    const bf16_t* __restrict__ v, const bf16_t* __restrict__ dout,
+    const float* __restrict__ lse, const float* __restrict__ delta,
+    const float* __restrict__ mask, bf16_t* __restrict__ dk,
+    bf16_t* __restrict__ dv, int S, int causal, float scale, int has_mask,
+    long mask_b_stride) {
+  constexpr int D = 64;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int hi = lane >> 5;
+  const int j32 = lane & 31;
+  const int ktile = blockIdx.x * 4 + (int)(threadIdx.x / WAVE);
+  const int k0 = ktile * 32;
+  if (k0 >= S) return;
+  const long bh = blockIdx.y;
+  const bf16_t* qp = q + bh * (long)S * D;
+  const bf16_t* kp = k + bh * (long)S * D;
+  const bf16_t* vp = v + bh * (long)S * D;
+  const bf16_t* dop = dout + bh * (long)S * D;
+  const float* lsep = lse + bh * (long)S;
+  const float* dltp = delta + bh * (long)S;
+  const float* mp =
+      has_mask ? (mask + (bh / mask_b_stride) * (long)S) : nullptr;
+
+  // B-operand fragments held for the whole loop:
+  //   K^T: B[k=d][j=key]  -> K[key=j32][d=s*16+hi*8+e]   (16B loads)
+  //   V^T: B[k=d][j=key]  -> V[key=j32][d=...]
+  const int key = k0 + j32;
+  const bf16_t* krp = kp + (long)min(key, S - 1) * D;
+  const bf16_t* vrp = vp + (long)min(key, S - 1) * D;
+  bf16x8v kf[4], vf[4];
+#pragma unroll
+  for (int s = 0; s < 4; ++s) {
+    kf[s] = *reinterpret_cast<const bf16x8v*>(krp + s * 16 + hi * 8);
+    vf[s] = *reinterpret_cast<const bf16x8v*>(vrp + s * 16 + hi * 8);
+  }
+  const float mask_val = (has_mask && key < S) ? mp[key] : 0.f;
+
+  f32x16 dv_acc[2], dk_acc[2];
+#pragma unroll
+  for (int h = 0; h < 2; ++h)
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      dv_acc[h][r] = 0.f;
+      dk_acc[h][r] = 0.f;
+    }
+
+  const int q_start = causal ? (k0 / 32) * 32 : 0;
+  for (int q0 = q_start; q0 < S; q0 += 32) {
+    // A-operands for S and dP: Q / dO rows (16B loads per step)
+    const int qrow_l = q0 + j32;
+    const bf16_t* qrp = qp + (long)min(qrow_l, S - 1) * D;
+    const bf16_t* dorp = dop + (long)min(qrow_l, S - 1) * D;
+    f32x16 s_acc, dp_acc;
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      s_acc[r] = 0.f;
+      dp_acc[r] = 0.f;
+    }
+#pragma unroll
+    for (int s = 0; s < 4; ++s) {
+      bf16x8v qf = *reinterpret_cast<const bf16x8v*>(qrp + s * 16 + hi * 8);
+      bf16x8v dof = *reinterpret_cast<const bf16x8v*>(dorp + s * 16 + hi * 8);
+      s_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qf, kf[s], s_acc,
+                                                      0, 0, 0);
+      dp_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dof, vf[s], dp_acc,
+                                                       0, 0, 0);
+    }
+    // P and dS in the [qrow-pattern][key=lane] layout
+    f32x16 p_acc, ds_acc;
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      const int qrow = q0 + (r & 3) + 8 * (r >> 2) + 4 * hi;
+      const int qclmp = min(qrow, S - 1);
+      const float l = lsep[qclmp];
+      float sv = s_acc[r] * scale + mask_val;
+      bool dead = (qrow >= S) || (key >= S) ||
+                  (causal && key > qrow) || !isfinite(l);
+      const float p = dead ? 0.f : __expf(sv - l);
+      p_acc[r] = p;
+      ds_acc[r] = dead ? 0.f : p * (dp_acc[r] - dltp[qclmp]) * scale;
+    }
+    // dV += P^T dO ; dK += dS^T Q : A = transform(acc) over k=qrows,
+    // B[k=qrow][j=d] = dO/Q rows, strided scalar loads
+    unsigned int pa_p[2][4], pa_ds[2][4];
+    acc_to_afrag(p_acc, pa_p);
+    acc_to_afrag(ds_acc, pa_ds);
+#pragma unroll
+    for (int h = 0; h < 2; ++h) {
+#pragma unroll
+      for (int step = 0; step < 2; ++step) {
+        bf16x8v dof, qf;
+#pragma unroll
+        for (int e = 0; e < 8; ++e) {
+          const int qrow = q0 + step * 16 + hi * 8 + e;
+          const long ro = (long)min(qrow, S - 1) * D + h * 32 + j32;
+          const bool live = qrow < S;
+          dof[e] = live ? *reinterpret_cast<const __bf16*>(dop + ro)
+                        : (__bf16)0.f;
+          qf[e] = live ? *reinterpret_cast<const __bf16*>(qp + ro)
+                       : (__bf16)0.f;
+        }
+        dv_acc[h] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+            *reinterpret_cast<const bf16x8v*>(&pa_p[step][0]), dof,
+            dv_acc[h], 0, 0, 0);
+        dk_acc[h] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+            *reinterpret_cast<const bf16x8v*>(&pa_ds[step][0]), qf,
+            dk_acc[h], 0, 0, 0);
+      }
+    }
+  }
+
+  // store dV, dK: [key-pattern][d=h*32+j32]
+#pragma unroll
+  for (int h = 0; h < 2; ++h) {
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      const int krow = k0 + (r & 3) + 8 * (r >> 2) + 4 * hi;
+      if (krow < S) {
+        const long off = bh * (long)S * D + (long)krow * D + h * 32 + j32;
+        dv[off] = f2bf(dv_acc[h][r]);
+        dk[off] = f2bf(dk_acc[h][r]);
+      }
+    }
+  }
+}
+
+// -------------------------------------------------------------------
+// dq kernel: wave owns q rows [q0, q0+32); loops kv tiles (forward
+// orientation: acc = [key-pattern][qrow=lane&31]).
+// -------------------------------------------------------------------
+__global__ __launch_bounds__(256) void attn_bwd_dq_kernel(
+    const bf16_t* __restrict__ q, const bf16_t* __restrict__ k,
+    const bf16_t* __restrict__ v, const bf16_t* __restrict__ dout,
+    const float* __restrict__ lse, const float* __restrict__ delta,
+    const float* __restrict__ mask, bf16_t* __restrict__ dq, int S,
+    int causal, float scale, int has_mask, long mask_b_stride) {
+  constexpr int D = 64;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int hi = lane >> 5;
+  const int j32 = lane & 31;
+  const int qtile = blockIdx.x * 4 + (int)(threadIdx.x / WAVE);
+  const int q0 = qtile * 32;
+  if (q0 >= S) return;
+  const long bh = blockIdx.y;
+  const bf16_t* qp = q + bh * (long)S * D;
+  const bf16_t* kp = k + bh * (long)S * D;
+  const bf16_t* vp = v + bh * (long)S * D;
+  const bf16_t* dop = dout + bh * (long)S * D;
+  const float* mp =
+      has_mask ? (mask + (bh / mask_b_stride) * (long)S) : nullptr;
+
+  const int qrow = q0 + j32;
+  const bf16_t* qrp = qp + (long)min(qrow, S - 1) * D;
+  const bf16_t* dorp = dop + (long)min(qrow, S - 1) * D;
+  bf16x8v qf[4], dof[4];
+#pragma unroll
+  for (int s = 0; s < 4; ++s) {
+    qf[s] = *reinterpret_cast<const bf16x8v*>(qrp + s * 16 + hi * 8);
+    dof[s] = *reinterpret_cast<const bf16x8v*>(dorp + s * 16 + hi * 8);
+  }
+  const float l_row = (qrow < S) ? lse[bh * (long)S + qrow] : -INFINITY;
+  const float dlt_row = (qrow < S) ? delta[bh * (long)S + qrow] : 0.f;
+
+  f32x16 dq_acc[2];
+#pragma unroll
+  for (int h = 0; h < 2; ++h)
+#pragma unroll
+    for (int r = 0; r < 16; ++r) dq_acc[h][r] = 0.f;
+
+  const int kv_end = causal ? min(S, q0 + 32) : S;
+  for (int k0 = 0; k0 < kv_end; k0 += 32) {
+    const int key_l = k0 + j32;
+    const bf16_t* krp = kp + (long)min(key_l, S - 1) * D;
+    const bf16_t* vrp = vp + (long)min(key_l, S - 1) * D;
+    f32x16 s_acc, dp_acc;
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      s_acc[r] = 0.f;
+      dp_acc[r] = 0.f;
+    }
+#pragma unroll
+    for (int s = 0; s < 4; ++s) {
+      bf16x8v kfr = *reinterpret_cast<const bf16x8v*>(krp + s * 16 + hi * 8);
+      bf16x8v vfr = *reinterpret_cast<const bf16x8v*>(vrp + s * 16 + hi * 8);
+      // S^T[key][qrow], dP^T[key][qrow]
+      s_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kfr, qf[s], s_acc,
+                                                      0, 0, 0);
+      dp_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vfr, dof[s], dp_acc,
+                                                       0, 0, 0);
+    }
+    f32x16 ds_acc;
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      const int kk = k0 + (r & 3) + 8 * (r >> 2) + 4 * hi;
+      float sv = s_acc[r] * scale;
+      if (has_mask && kk < S) sv += mp[kk];
+      bool dead = (qrow >= S) || (kk >= S) || (causal && kk > qrow) ||
+                  !isfinite(l_row);
+      const float p = dead ? 0.f : __expf(sv - l_row);
+      ds_acc[r] = dead ? 0.f : p * (dp_acc[r] - dlt_row) * scale;
+    }
+    // dQ += dS K : A = transform(dS^T) over k=keys; B[k=key][j=d] = K rows
+    unsigned int pa_ds[2][4];
+    acc_to_afrag(ds_acc, pa_ds);
+#pragma unroll
+    for (int h = 0; h < 2; ++h) {
+#pragma unroll
+      for (int step = 0; step < 2; ++step) {
+        bf16x8v kcol;
+#pragma unroll
+        for (int e = 0; e < 8; ++e) {
+          const int kk = k0 + step * 16 + hi * 8 + e;
+          kcol[e] = (kk < S)
+              ? *reinterpret_cast<const __bf16*>(
+                    kp + (long)kk * D + h * 32 + j32)
+              : (__bf16)0.f;
+        }
+        dq_acc[h] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+            *reinterpret_cast<const bf16x8v*>(&pa_ds[step][0]), kcol,
+            dq_acc[h], 0, 0, 0);
+      }
+    }
+  }
+
+#pragma unroll
+  for (int h = 0; h < 2; ++h) {
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      const int row = q0 + (r & 3) + 8 * (r >> 2) + 4 * hi;
+      if (row < S)
+        dq[bh * (long)S * D + (long)row * D + h * 32 + j32] =
+            f2bf(dq_acc[h][r]);
+    }
+  }
+}
+
+}  // namespace
+
+std::vector<at::Tensor> attn_bwd(at::Tensor q, at::Tensor k, at::Tensor v,
+                                 at::Tensor o, at::Tensor dout,
+                                 at::Tensor lse, at::Tensor mask, bool causal,
+                                 double scale) {
+  TORCH_CHECK(q.is_cuda() && q.is_contiguous() && k.is_contiguous() &&
+              v.is_contiguous() && o.is_contiguous());
+  TORCH_CHECK(q.scalar_type() == at::kBFloat16);
+  const long B = q.size(0), H = q.size(1);
+  const int S = q.size(2), D = q.size(3);
+  TORCH_CHECK(D == 64, "attn_bwd kernel: head_dim 64");
+  auto dc = dout.contiguous();
+  auto dq = at::empty_like(q);
+  auto dk = at::empty_like(k);
+  auto dv = at::empty_like(v);
+  auto delta = at::empty({B * H * (long)S}, q.options().dtype(at::kFloat));
+
+  const bool has_mask = mask.defined() && mask.numel() > 0;
+  at::Tensor mask_f;
+  const float* mask_ptr = nullptr;
+  if (has_mask) {
+    mask_f = mask.to(at::kFloat).reshape({B, S}).contiguous();
+    mask_ptr = mask_f.data_ptr<float>();
+  }
+
+  auto stream = c10::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
+  const long NR = B * H * (long)S;
+  hipLaunchKernelGGL((attn_bwd_delta_kernel<64>),
+                     dim3((NR + 3) / 4), dim3(256), 0, stream,
+                     reinterpret_cast<const bf16_t*>(dc.data_ptr()),
+                     reinterpret_cast<const bf16_t*>(o.data_ptr()),
+                     delta.data_ptr<float>(), NR);
+  dim3 block(256);
+  dim3 gridk((S + 127) / 128, B * H);
+  hipLaunchKernelGGL(attn_bwd_dkv_kernel, gridk, block, 0, stream,
+                     reinterpret_cast<const bf16_t*>(q.data_ptr()),
+                     reinterpret_cast<const bf16_t*>(k.data_ptr()),
+                     reinterpret_cast<const bf16_t*>(v.data_ptr()),
+                     reinterpret_cast<const bf16_t*>(dc.data_ptr()),
+                     lse.data_ptr<float>(), delta.data_ptr<float>(),
+                     mask_ptr, reinterpret_cast<bf16_t*>(dk.data_ptr()),
+                     reinterpret_cast<bf16_t*>(dv.data_ptr()), S,
+                     causal ? 1 : 0, (float)scale, has_mask ? 1 : 0,
+                     (long)H);
+  hipLaunchKernelGGL(attn_bwd_dq_kernel, gridk, block, 0, stream,
+                     reinterpret_cast<const bf16_t*>(q.data_ptr()),
+                     reinterpret_cast<const bf16_t*>(k.data_ptr()),
+                     reinterpret_cast<const bf16_t*>(v.data_ptr()),
+                     reinterpret_cast<const bf16_t*>(dc.data_ptr()),
+                     lse.data_ptr<float>(), delta.data_ptr<float>(),
+                     mask_ptr, reinterpret_cast<bf16_t*>(dq.data_ptr()), S,
+                     causal ? 1 : 0, (float)scale, has_mask ? 1 : 0,
+                     (long)H);
+  HIP_CHECK_LAST();
+  return {dq, dk, dv};
+}

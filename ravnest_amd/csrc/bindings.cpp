@@ -30,6 +30,10 @@ void fused_lamb(std::vector<at::Tensor> ps, std::vector<at::Tensor> gs,
                 double clamp_trust);
 std::vector<at::Tensor> attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v,
                                  at::Tensor mask, bool causal, double scale);
+std::vector<at::Tensor> attn_bwd(at::Tensor q, at::Tensor k, at::Tensor v,
+                                 at::Tensor o, at::Tensor dout,
+                                 at::Tensor lse, at::Tensor mask, bool causal,
+                                 double scale);
 at::Tensor mfma_probe(at::Tensor a, at::Tensor b);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -45,5 +49,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("fused_sgd", &fused_sgd, "multi-tensor SGD+momentum");
   m.def("fused_lamb", &fused_lamb, "multi-tensor LAMB");
   m.def("attn_fwd", &attn_fwd, "flash attention fwd (bf16, MFMA)");
+  m.def("attn_bwd", &attn_bwd, "flash attention bwd (bf16, MFMA, D=64)");
   m.def("mfma_probe", &mfma_probe, "32x32x16 bf16 MFMA layout probe");
 }

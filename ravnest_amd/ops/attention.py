@@ -54,7 +54,14 @@ class _AttnFn(torch.autograd.Function):
         q, k, v, mask, o, lse = ctx.saved_tensors
         causal, scale = ctx.causal, ctx.scale
         do = do.contiguous()
-        # recompute P from lse: P = exp(S*scale + mask - lse)
+        if q.shape[-1] == 64:
+            ext = get_ext(required=True)
+            dq, dk, dv = ext.attn_bwd(
+                q, k, v, o, do, lse,
+                mask if mask is not None else torch.Tensor(), causal, scale)
+            return dq, dk, dv, None, None, None
+        # head_dim != 64: recompute P from lse via library GEMMs
+        # P = exp(S*scale + mask - lse)
         s = (q @ k.transpose(-2, -1)).float() * scale
         if causal:
             S = q.shape[-2]

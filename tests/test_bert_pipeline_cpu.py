@@ -1,0 +1,98 @@
+"""BERT-tiny 4-stage async pipeline on CPU (gloo): kwargs model inputs,
+attention-mask forwarding from root to every stage (multi-consumer
+routing), CE leaf loss. The hard-parts checklist of SURVEY.md section 7
+(routing-template semantics: root model_inputs forwarded to later stages).
+"""
+import os
+
+import numpy as np
+import pytest
+import torch
+import torch.multiprocessing as mp
+
+from ravnest_amd import clusterize, set_seed
+from ravnest_amd.planner import NodeSpec
+from ravnest_amd.models import BertConfig, BertForMLM
+
+VOCAB = 256
+SEQ = 32
+BATCH = 8
+NBATCH = 6
+
+
+def _data(seed=42):
+    g = np.random.RandomState(seed)
+    ids = g.randint(0, VOCAB, size=(NBATCH * BATCH, SEQ))
+    return torch.tensor(ids, dtype=torch.int64)
+
+
+def _loss_fn(preds, targets):
+    return torch.nn.functional.cross_entropy(
+        preds.reshape(-1, preds.shape[-1]).float(), targets.reshape(-1))
+
+
+class _KwTrainerLoop:
+    """Feeds kwargs microbatches like the reference's BERT custom trainer
+    (examples/bert/bert_trainer.py:11-13)."""
+
+
+def _worker(rank, base_dir, port, out_dir):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    os.chdir(out_dir)
+    set_seed(42)
+    from ravnest_amd import Node, Trainer
+    ids = _data()
+    labels = [ids[i * BATCH:(i + 1) * BATCH] for i in range(NBATCH)]
+    node = Node(name=f"node_{rank}", base_dir=base_dir,
+                optimizer=torch.optim.Adam,
+                optimizer_params={"lr": 3e-3},
+                device=torch.device("cpu"),
+                criterion=_loss_fn,
+                labels=labels * 50,
+                update_frequency=1)
+    node.start()
+    if node.node_type.value == "root":
+        mask = torch.ones(BATCH, SEQ, dtype=torch.int64)
+        for epoch in range(3):
+            for i in range(NBATCH):
+                node.forward_compute(input_ids=labels[i],
+                                     attention_mask=mask)
+            node.wait_for_backwards(timeout=300)
+        node.stop_cluster()
+    else:
+        trainer = Trainer(node=node)
+        trainer.prelim_checks()
+    node.stop()
+
+
+def test_bert_pipeline_4stage(tmp_path):
+    set_seed(42)
+    cfg = BertConfig(vocab_size=VOCAB, hidden=64, layers=4, heads=2,
+                     intermediate=128, max_seq=SEQ, dropout=0.0)
+    model = BertForMLM(cfg)
+    ids = _data()[:2]
+    mask = torch.ones(2, SEQ, dtype=torch.int64)
+    base = str(tmp_path / "node_data")
+    pool = [NodeSpec(name=f"n{i}", ram=10 * 2**20) for i in range(4)]
+    meta = clusterize(model, (ids, mask), node_pool=pool, max_clusters=1,
+                      base_dir=base)
+    assert meta["world_size"] == 4
+
+    port = 29800 + (os.getpid() % 150)
+    ctx = mp.get_context("spawn")
+    procs = [ctx.Process(target=_worker, args=(r, base, port, str(tmp_path)))
+             for r in range(4)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(timeout=300)
+    for p in procs:
+        assert p.exitcode == 0, f"worker exited {p.exitcode}"
+
+    loss_file = tmp_path / "losses.txt"
+    assert loss_file.exists()
+    losses = [float(x) for x in loss_file.read_text().split()]
+    assert len(losses) == 3 * NBATCH
+    # memorizing a 6-batch dataset: loss must drop substantially
+    assert np.mean(losses[-NBATCH:]) < np.mean(losses[:NBATCH]) * 0.9, losses

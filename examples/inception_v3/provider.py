@@ -1,0 +1,42 @@
+"""Inception-V3 (CIFAR-adapted), SGD (parity: reference
+examples/inception_v3/provider.py — CIFAR-10 replaced by synthetic 32x32
+batches: no network in this environment)."""
+import os
+import sys
+
+sys.path.insert(0, os.path.join(os.path.dirname(__file__), "..", ".."))
+from examples.common import node_name  # noqa: E402
+
+import torch  # noqa: E402
+
+from ravnest_amd import Node, Trainer, set_seed  # noqa: E402
+from ravnest_amd.ops import FusedSGD  # noqa: E402
+
+set_seed(42)
+BATCH, NBATCH = 64, 50
+
+
+def synthetic_loader():
+    g = torch.Generator().manual_seed(42)
+    return [(torch.randn(BATCH, 3, 32, 32, generator=g),
+             torch.randint(0, 10, (BATCH,), generator=g))
+            for _ in range(NBATCH)]
+
+
+def loss_fn(preds, targets):
+    return torch.nn.functional.cross_entropy(preds.float(), targets[1])
+
+
+if __name__ == "__main__":
+    name, base_dir = node_name()
+    loader = synthetic_loader()
+    node = Node(name=name, base_dir=base_dir,
+                optimizer=FusedSGD,
+                optimizer_params={"lr": 0.01, "momentum": 0.9,
+                                  "weight_decay": 5e-4},
+                criterion=loss_fn,
+                labels=loader)
+    node.start()
+    trainer = Trainer(node=node, train_loader=loader, epochs=10,
+                      batch_size=BATCH)
+    trainer.train()

@@ -66,10 +66,12 @@ def clusterize(model: torch.nn.Module,
     """
     base = Path(base_dir)
     if base.exists():
-        for child in base.iterdir():  # reference wipes node_data on re-plan
+        # reference wipes node_data SUBFOLDERS on re-plan
+        # (operations/utils.py:390-391) but keeps node_configs.json
+        for child in base.iterdir():
             if child.is_dir():
                 shutil.rmtree(child)
-            else:
+            elif child.suffix == ".json" and child.name != "node_configs.json":
                 child.unlink()
     (base / "nodes").mkdir(parents=True, exist_ok=True)
 

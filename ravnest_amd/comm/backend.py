@@ -62,6 +62,12 @@ def edges_from_plan(base_dir: str) -> list[Edge]:
                 if src.get("kind") == "model_input" and s != 0:
                     edges.add(Edge(ranks[0], ranks[s], "fwd"))
                     edges.add(Edge(ranks[s], ranks[0], "bwd"))
+    # ctrl edges between pipeline neighbors (heartbeat / save / recovery)
+    for cl in plan["clusters"]:
+        ranks = cl["stage_ranks"]
+        for a, b in zip(ranks, ranks[1:]):
+            edges.add(Edge(a, b, "ctrl"))
+            edges.add(Edge(b, a, "ctrl"))
     # ctrl ring across clusters for weight pull / elastic join
     clusters = plan["clusters"]
     if len(clusters) > 1:

@@ -28,7 +28,7 @@ from dataclasses import dataclass, field
 
 import torch
 
-from ..utils import current_rng_states, restore_rng_states
+from ..utils import current_rng_states, restore_rng_states, trace_range
 
 
 @dataclass
@@ -154,7 +154,7 @@ class ComputeEngine:
                           args=[a.detach().clone() if torch.is_tensor(a) else a
                                 for a in args],
                           needs_grad=list(needs_grad))
-        with torch.no_grad(), self._autocast():
+        with torch.no_grad(), self._autocast(), trace_range(f"fwd:{fpid}"):
             out = self.model(*args)
         outputs = out if isinstance(out, tuple) else (out,)
         with self._lock:
@@ -214,7 +214,8 @@ class ComputeEngine:
             outs.append(out)
             grads.append(g.to(out.dtype).to(out.device))
         if outs:
-            torch.autograd.backward(outs, grads)
+            with trace_range(f"bwd:{fpid}"):
+                torch.autograd.backward(outs, grads)
 
         input_grads: list[torch.Tensor | None] = []
         for a, ng in zip(rec.recomputed_args, rec.needs_grad):

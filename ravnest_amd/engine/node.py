@@ -194,6 +194,31 @@ class Node:
 
         self._dispatch_thread: threading.Thread | None = None
         self._started = False
+        self.health_monitor = None
+
+    # ==================================================================
+    # aux subsystems (health, checkpoint) — SURVEY.md section 5 parity+
+    # ==================================================================
+    def start_health_monitor(self, interval: float = 2.0,
+                             timeout: float = 10.0, on_peer_lost=None):
+        from .health import HealthMonitor
+        self.health_monitor = HealthMonitor(self, interval=interval,
+                                            timeout=timeout,
+                                            on_peer_lost=on_peer_lost)
+        self.health_monitor.start()
+        return self.health_monitor
+
+    def health(self) -> dict:
+        return (self.health_monitor.health()
+                if self.health_monitor is not None else {"peers": {}})
+
+    def save_checkpoint(self, path=None):
+        from .checkpoint import save_checkpoint
+        return save_checkpoint(self, path)
+
+    def load_checkpoint(self, path=None):
+        from .checkpoint import load_checkpoint
+        return load_checkpoint(self, path)
 
     # ==================================================================
     # routing tables
@@ -319,7 +344,20 @@ class Node:
         elif msg.action == ActionTypes.SAVE_SUBMODEL:
             self._enqueue(1, ("save_submodel", None, None))
         elif msg.action == ActionTypes.STOP:
-            self._enqueue(3, ("stop_cascade", None, None))
+            if msg.extra == 3:  # health PING -> reply PONG on ctrl
+                if self.health_monitor is not None:
+                    self.health_monitor.note_ping(channel.src)
+                try:
+                    self.comm.send(channel.src, "ctrl", Message(
+                        action=ActionTypes.STOP, fpid=-2, tensors=[],
+                        extra=4))
+                except KeyError:
+                    pass
+            elif msg.extra == 4:  # PONG
+                if self.health_monitor is not None:
+                    self.health_monitor.note_pong(channel.src)
+            else:
+                self._enqueue(3, ("stop_cascade", None, None))
         elif msg.action == ActionTypes.PREDICTION and channel.kind == "ctrl":
             # WEIGHTS protocol over ctrl: extra==1 request, extra==2 reply
             if msg.extra == 1:

@@ -35,6 +35,29 @@ DTYPE_CODES = {
 CODE_DTYPES = {v: k for k, v in DTYPE_CODES.items()}
 
 
+_TRACE = os.environ.get("RAVNEST_TRACE", "0") == "1"
+
+
+class trace_range:
+    """rocTX range (via torch's nvtx shim, which maps to roctx on ROCm).
+    Enabled with RAVNEST_TRACE=1; shows per-stage/microbatch phases in
+    rocprofv3 runtime traces (SURVEY.md section 5 tracing parity)."""
+
+    def __init__(self, name: str):
+        self.name = name
+        self.active = _TRACE and torch.cuda.is_available()
+
+    def __enter__(self):
+        if self.active:
+            torch.cuda.nvtx.range_push(self.name)
+        return self
+
+    def __exit__(self, *a):
+        if self.active:
+            torch.cuda.nvtx.range_pop()
+        return False
+
+
 def set_seed(seed: int) -> None:
     """Deterministic seeding across python/numpy/torch/ROCm.
 

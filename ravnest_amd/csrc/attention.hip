@@ -35,12 +35,19 @@ DEVINL unsigned int pack_bf2(float lo, float hi) {
   return (unsigned int)f2us(lo) | ((unsigned int)f2us(hi) << 16);
 }
 
+// addressing: element offset of (b, h, row) = b*bs + h*hs + row*rs.
+// separate (B,H,S,D) tensors: bs=H*S*D, hs=S*D, rs=D.
+// packed qkv (B,S,3,H,D): bs=S*3*H*D, hs=D (+ part*H*D folded into the
+// base pointer), rs=3*H*D. o packed (B,S,H,D): bs=S*H*D, hs=D, rs=H*D.
+struct Strides { long bs, hs, rs; };
+
 template <int D>  // head_dim: 64 or 128
 __global__ __launch_bounds__(256) void attn_fwd_kernel(
     const bf16_t* __restrict__ q, const bf16_t* __restrict__ k,
     const bf16_t* __restrict__ v, const float* __restrict__ mask,
     bf16_t* __restrict__ o, float* __restrict__ lse, int S, int causal,
-    float scale, int has_mask, long mask_b_stride) {
+    float scale, int has_mask, long H,
+    Strides sq, Strides sk, Strides sv, Strides so) {
   constexpr int DSTEPS = D / 16;   // QK^T k-steps
   constexpr int DHALF = D / 32;    // PV column halves
   const int lane = threadIdx.x & (WAVE - 1);
@@ -52,18 +59,18 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
   const int q0 = qtile * 32;
   if (q0 >= S) return;
   const long bh = blockIdx.y;  // fused batch*head index
-  const bf16_t* qp = q + bh * (long)S * D;
-  const bf16_t* kp = k + bh * (long)S * D;
-  const bf16_t* vp = v + bh * (long)S * D;
-  // mask is (B, S) fp32; mask_b_stride = H so b = bh / H
-  const float* mp =
-      has_mask ? (mask + (bh / mask_b_stride) * (long)S) : nullptr;
+  const long b = bh / H, h = bh % H;
+  const bf16_t* qp = q + b * sq.bs + h * sq.hs;
+  const bf16_t* kp = k + b * sk.bs + h * sk.hs;
+  const bf16_t* vp = v + b * sv.bs + h * sv.hs;
+  bf16_t* op = o + b * so.bs + h * so.hs;
+  const float* mp = has_mask ? (mask + b * (long)S) : nullptr;
 
   // ---- load Q fragments: B-operand, j = qrow = j32, k = d ----
   // frag[s] covers d in [s*16 + hi*8, +8)
   bf16x8v qf[DSTEPS];
   const int qrow = q0 + j32;
-  const bf16_t* qrp = qp + (long)min(qrow, S - 1) * D;
+  const bf16_t* qrp = qp + (long)min(qrow, S - 1) * sq.rs;
 #pragma unroll
   for (int s = 0; s < DSTEPS; ++s)
     qf[s] = *reinterpret_cast<const bf16x8v*>(qrp + s * 16 + hi * 8);
@@ -85,7 +92,7 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
 #pragma unroll
     for (int r = 0; r < 16; ++r) s_acc[r] = 0.f;
     const int krow = k0 + j32;
-    const bf16_t* krp = kp + (long)min(krow, S - 1) * D;
+    const bf16_t* krp = kp + (long)min(krow, S - 1) * sk.rs;
 #pragma unroll
     for (int s = 0; s < DSTEPS; ++s) {
       bf16x8v kf = *reinterpret_cast<const bf16x8v*>(krp + s * 16 + hi * 8);
@@ -166,7 +173,7 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
     // ---- PV: O[qrow][d] += P A-frag x V B-frag ----
     // B[k=key][j=d]: lane reads V[k0 + step*16 + hi*8 + e][dhalf*32 + j32]
 #pragma unroll
-    for (int h = 0; h < DHALF; ++h) {
+    for (int hh = 0; hh < DHALF; ++hh) {
 #pragma unroll
       for (int step = 0; step < 2; ++step) {
         bf16x8v vf;
@@ -174,10 +181,10 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
         for (int e = 0; e < 8; ++e) {
           const int key = k0 + step * 16 + hi * 8 + e;
           vf[e] = *reinterpret_cast<const __bf16*>(
-              vp + (long)min(key, S - 1) * D + h * 32 + j32);
+              vp + (long)min(key, S - 1) * sv.rs + hh * 32 + j32);
         }
-        oacc[h] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-            *reinterpret_cast<const bf16x8v*>(&pa[step][0]), vf, oacc[h],
+        oacc[hh] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+            *reinterpret_cast<const bf16x8v*>(&pa[step][0]), vf, oacc[hh],
             0, 0, 0);
       }
     }
@@ -188,16 +195,16 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
   const float l_safe = (l_run > 0.f) ? l_run : 1.f;
   // per-reg qrow for the PV acc layout
 #pragma unroll
-  for (int h = 0; h < DHALF; ++h) {
+  for (int hh = 0; hh < DHALF; ++hh) {
 #pragma unroll
     for (int r = 0; r < 16; ++r) {
       const int row_local = (r & 3) + 8 * (r >> 2) + 4 * hi;
       // l of that row lives in lane row_local (and row_local+32)
       const float l_row = __shfl(l_safe, row_local, WAVE);
-      const float val = oacc[h][r] / l_row;
+      const float val = oacc[hh][r] / l_row;
       const int row = q0 + row_local;
       if (row < S)
-        o[bh * (long)S * D + (long)row * D + h * 32 + j32] = f2bf(val);
+        op[(long)row * so.rs + hh * 32 + j32] = f2bf(val);
     }
   }
   if (qrow < S && hi == 0)
@@ -250,10 +257,7 @@ std::vector<at::Tensor> attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v,
   const bool has_mask = mask.defined() && mask.numel() > 0;
   at::Tensor mask_f;
   const float* mask_ptr = nullptr;
-  long mask_bh = 1;
   if (has_mask) {
-    // accept (B,1,1,S): expand to per-(b,h) row pointer via stride trick:
-    // kernel indexes mp[key] with mp = mask + (bh/H)*S
     mask_f = mask.to(at::kFloat).reshape({B, S}).contiguous();
     mask_ptr = mask_f.data_ptr<float>();
   }
@@ -261,26 +265,69 @@ std::vector<at::Tensor> attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v,
   auto stream = c10::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
   dim3 block(256);
   dim3 grid((S + 127) / 128, B * H);
+  const Strides sep{H * (long)S * D, (long)S * D, (long)D};
+  const Strides so = sep;
 
-  // NOTE: kernel receives the per-(b,h) mask base via a wrapper lambda —
-  // we pass mask + (bh/H)*S inside the kernel using H, encoded in
-  // mask_b_stride (= H).
+#define LAUNCH_ATTN_FWD(DD, QP, KP, VP, OP, SQ, SK, SV, SO)                 \
+  hipLaunchKernelGGL((attn_fwd_kernel<DD>), grid, block, 0, stream, QP, KP, \
+                     VP, mask_ptr, OP, lse.data_ptr<float>(), S,            \
+                     causal ? 1 : 0, (float)scale, has_mask ? 1 : 0,        \
+                     (long)H, SQ, SK, SV, SO)
+
   if (D == 64) {
-    hipLaunchKernelGGL((attn_fwd_kernel<64>), grid, block, 0, stream,
-                       reinterpret_cast<const bf16_t*>(q.data_ptr()),
-                       reinterpret_cast<const bf16_t*>(k.data_ptr()),
-                       reinterpret_cast<const bf16_t*>(v.data_ptr()),
-                       mask_ptr, reinterpret_cast<bf16_t*>(o.data_ptr()),
-                       lse.data_ptr<float>(), S, causal ? 1 : 0,
-                       (float)scale, has_mask ? 1 : 0, (long)H);
+    LAUNCH_ATTN_FWD(64, reinterpret_cast<const bf16_t*>(q.data_ptr()),
+                    reinterpret_cast<const bf16_t*>(k.data_ptr()),
+                    reinterpret_cast<const bf16_t*>(v.data_ptr()),
+                    reinterpret_cast<bf16_t*>(o.data_ptr()), sep, sep, sep,
+                    so);
   } else {
-    hipLaunchKernelGGL((attn_fwd_kernel<128>), grid, block, 0, stream,
-                       reinterpret_cast<const bf16_t*>(q.data_ptr()),
-                       reinterpret_cast<const bf16_t*>(k.data_ptr()),
-                       reinterpret_cast<const bf16_t*>(v.data_ptr()),
-                       mask_ptr, reinterpret_cast<bf16_t*>(o.data_ptr()),
-                       lse.data_ptr<float>(), S, causal ? 1 : 0,
-                       (float)scale, has_mask ? 1 : 0, (long)H);
+    LAUNCH_ATTN_FWD(128, reinterpret_cast<const bf16_t*>(q.data_ptr()),
+                    reinterpret_cast<const bf16_t*>(k.data_ptr()),
+                    reinterpret_cast<const bf16_t*>(v.data_ptr()),
+                    reinterpret_cast<bf16_t*>(o.data_ptr()), sep, sep, sep,
+                    so);
+  }
+  HIP_CHECK_LAST();
+  return {o, lse};
+}
+
+// packed layout: qkv (B,S,3,H,D) contiguous -> o (B,S,H,D); no
+// permute/contiguous copies around the attention core.
+std::vector<at::Tensor> attn_fwd_qkv(at::Tensor qkv, at::Tensor mask,
+                                     bool causal, double scale) {
+  TORCH_CHECK(qkv.is_cuda() && qkv.is_contiguous());
+  TORCH_CHECK(qkv.dim() == 5 && qkv.size(2) == 3,
+              "qkv must be (B,S,3,H,D)");
+  TORCH_CHECK(qkv.scalar_type() == at::kBFloat16);
+  const long B = qkv.size(0), H = qkv.size(3);
+  const int S = qkv.size(1), D = qkv.size(4);
+  TORCH_CHECK(D == 64 || D == 128, "attn_fwd_qkv: head_dim 64 or 128");
+
+  auto o = at::empty({B, (long)S, H, (long)D}, qkv.options());
+  auto lse = at::empty({B, H, S}, qkv.options().dtype(at::kFloat));
+
+  const bool has_mask = mask.defined() && mask.numel() > 0;
+  at::Tensor mask_f;
+  const float* mask_ptr = nullptr;
+  if (has_mask) {
+    mask_f = mask.to(at::kFloat).reshape({B, S}).contiguous();
+    mask_ptr = mask_f.data_ptr<float>();
+  }
+
+  auto stream = c10::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
+  dim3 block(256);
+  dim3 grid((S + 127) / 128, B * H);
+  const long HD = H * (long)D;
+  const Strides sp{(long)S * 3 * HD, (long)D, 3 * HD};
+  const Strides so{(long)S * HD, (long)D, HD};
+  const bf16_t* base = reinterpret_cast<const bf16_t*>(qkv.data_ptr());
+
+  if (D == 64) {
+    LAUNCH_ATTN_FWD(64, base, base + HD, base + 2 * HD,
+                    reinterpret_cast<bf16_t*>(o.data_ptr()), sp, sp, sp, so);
+  } else {
+    LAUNCH_ATTN_FWD(128, base, base + HD, base + 2 * HD,
+                    reinterpret_cast<bf16_t*>(o.data_ptr()), sp, sp, sp, so);
   }
   HIP_CHECK_LAST();
   return {o, lse};

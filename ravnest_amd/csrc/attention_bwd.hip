@@ -24,6 +24,9 @@ typedef float f32x16 __attribute__((ext_vector_type(16)));
 
 namespace {
 
+// element offset of (b, h, row) = b*bs + h*hs + row*rs (see attention.hip)
+struct StridesB { long bs, hs, rs; };
+
 DEVINL unsigned int pack_bf2b(float lo, float hi) {
   return (unsigned int)f2us(lo) | ((unsigned int)f2us(hi) << 16);
 }
@@ -55,17 +58,21 @@ DEVINL void acc_to_afrag(const f32x16& acc, unsigned int pa[2][4]) {
 template <int D>
 __global__ void attn_bwd_delta_kernel(const bf16_t* __restrict__ dout,
                                       const bf16_t* __restrict__ o,
-                                      float* __restrict__ delta, long NR) {
+                                      float* __restrict__ delta, int S,
+                                      long H, long NR, StridesB sdo) {
   const int lane = threadIdx.x & (WAVE - 1);
-  const long row = (long)blockIdx.x * (blockDim.x / WAVE) +
-                   threadIdx.x / WAVE;
-  if (row >= NR) return;
+  const long id = (long)blockIdx.x * (blockDim.x / WAVE) +
+                  threadIdx.x / WAVE;
+  if (id >= NR) return;
+  const long bh = id / S;
+  const long row = id % S;
+  const long off = (bh / H) * sdo.bs + (bh % H) * sdo.hs + row * sdo.rs;
   float s = 0.f;
 #pragma unroll
   for (int d = lane; d < D; d += WAVE)
-    s += bf2f(dout[row * D + d]) * bf2f(o[row * D + d]);
+    s += bf2f(dout[off + d]) * bf2f(o[off + d]);
   s = wave_sum(s);
-  if (lane == 0) delta[row] = s;
+  if (lane == 0) delta[id] = s;
 }
 
 // -------------------------------------------------------------------
@@ -79,7 +86,8 @@ __global__ __launch_bounds__(256) void attn_bwd_dkv_kernel(
     const float* __restrict__ lse, const float* __restrict__ delta,
     const float* __restrict__ mask, bf16_t* __restrict__ dk,
     bf16_t* __restrict__ dv, int S, int causal, float scale, int has_mask,
-    long mask_b_stride) {
+    long H, StridesB sio, StridesB sdo, StridesB sg) {
+  // sio: strides of q/k/v; sdo: of o/dout; sg: of dk/dv outputs
   constexpr int D = 64;
   const int lane = threadIdx.x & (WAVE - 1);
   const int hi = lane >> 5;
@@ -88,21 +96,23 @@ __global__ __launch_bounds__(256) void attn_bwd_dkv_kernel(
   const int k0 = ktile * 32;
   if (k0 >= S) return;
   const long bh = blockIdx.y;
-  const bf16_t* qp = q + bh * (long)S * D;
-  const bf16_t* kp = k + bh * (long)S * D;
-  const bf16_t* vp = v + bh * (long)S * D;
-  const bf16_t* dop = dout + bh * (long)S * D;
+  const long b = bh / H, h = bh % H;
+  const bf16_t* qp = q + b * sio.bs + h * sio.hs;
+  const bf16_t* kp = k + b * sio.bs + h * sio.hs;
+  const bf16_t* vp = v + b * sio.bs + h * sio.hs;
+  const bf16_t* dop = dout + b * sdo.bs + h * sdo.hs;
+  bf16_t* dkp = dk + b * sg.bs + h * sg.hs;
+  bf16_t* dvp = dv + b * sg.bs + h * sg.hs;
   const float* lsep = lse + bh * (long)S;
   const float* dltp = delta + bh * (long)S;
-  const float* mp =
-      has_mask ? (mask + (bh / mask_b_stride) * (long)S) : nullptr;
+  const float* mp = has_mask ? (mask + b * (long)S) : nullptr;
 
   // B-operand fragments held for the whole loop:
   //   K^T: B[k=d][j=key]  -> K[key=j32][d=s*16+hi*8+e]   (16B loads)
   //   V^T: B[k=d][j=key]  -> V[key=j32][d=...]
   const int key = k0 + j32;
-  const bf16_t* krp = kp + (long)min(key, S - 1) * D;
-  const bf16_t* vrp = vp + (long)min(key, S - 1) * D;
+  const bf16_t* krp = kp + (long)min(key, S - 1) * sio.rs;
+  const bf16_t* vrp = vp + (long)min(key, S - 1) * sio.rs;
   bf16x8v kf[4], vf[4];
 #pragma unroll
   for (int s = 0; s < 4; ++s) {
@@ -124,8 +134,8 @@ __global__ __launch_bounds__(256) void attn_bwd_dkv_kernel(
   for (int q0 = q_start; q0 < S; q0 += 32) {
     // A-operands for S and dP: Q / dO rows (16B loads per step)
     const int qrow_l = q0 + j32;
-    const bf16_t* qrp = qp + (long)min(qrow_l, S - 1) * D;
-    const bf16_t* dorp = dop + (long)min(qrow_l, S - 1) * D;
+    const bf16_t* qrp = qp + (long)min(qrow_l, S - 1) * sio.rs;
+    const bf16_t* dorp = dop + (long)min(qrow_l, S - 1) * sdo.rs;
     f32x16 s_acc, dp_acc;
 #pragma unroll
     for (int r = 0; r < 16; ++r) {
@@ -161,40 +171,41 @@ __global__ __launch_bounds__(256) void attn_bwd_dkv_kernel(
     acc_to_afrag(p_acc, pa_p);
     acc_to_afrag(ds_acc, pa_ds);
 #pragma unroll
-    for (int h = 0; h < 2; ++h) {
+    for (int hh = 0; hh < 2; ++hh) {
 #pragma unroll
       for (int step = 0; step < 2; ++step) {
         bf16x8v dof, qf;
 #pragma unroll
         for (int e = 0; e < 8; ++e) {
           const int qrow = q0 + step * 16 + hi * 8 + e;
-          const long ro = (long)min(qrow, S - 1) * D + h * 32 + j32;
+          const long rq = (long)min(qrow, S - 1) * sio.rs + hh * 32 + j32;
+          const long rd = (long)min(qrow, S - 1) * sdo.rs + hh * 32 + j32;
           const bool live = qrow < S;
-          dof[e] = live ? *reinterpret_cast<const __bf16*>(dop + ro)
+          dof[e] = live ? *reinterpret_cast<const __bf16*>(dop + rd)
                         : (__bf16)0.f;
-          qf[e] = live ? *reinterpret_cast<const __bf16*>(qp + ro)
+          qf[e] = live ? *reinterpret_cast<const __bf16*>(qp + rq)
                        : (__bf16)0.f;
         }
-        dv_acc[h] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+        dv_acc[hh] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
             *reinterpret_cast<const bf16x8v*>(&pa_p[step][0]), dof,
-            dv_acc[h], 0, 0, 0);
-        dk_acc[h] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+            dv_acc[hh], 0, 0, 0);
+        dk_acc[hh] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
             *reinterpret_cast<const bf16x8v*>(&pa_ds[step][0]), qf,
-            dk_acc[h], 0, 0, 0);
+            dk_acc[hh], 0, 0, 0);
       }
     }
   }
 
-  // store dV, dK: [key-pattern][d=h*32+j32]
+  // store dV, dK: [key-pattern][d=hh*32+j32]
 #pragma unroll
-  for (int h = 0; h < 2; ++h) {
+  for (int hh = 0; hh < 2; ++hh) {
 #pragma unroll
     for (int r = 0; r < 16; ++r) {
       const int krow = k0 + (r & 3) + 8 * (r >> 2) + 4 * hi;
       if (krow < S) {
-        const long off = bh * (long)S * D + (long)krow * D + h * 32 + j32;
-        dv[off] = f2bf(dv_acc[h][r]);
-        dk[off] = f2bf(dk_acc[h][r]);
+        const long off = (long)krow * sg.rs + hh * 32 + j32;
+        dvp[off] = f2bf(dv_acc[hh][r]);
+        dkp[off] = f2bf(dk_acc[hh][r]);
       }
     }
   }
@@ -209,7 +220,8 @@ __global__ __launch_bounds__(256) void attn_bwd_dq_kernel(
     const bf16_t* __restrict__ v, const bf16_t* __restrict__ dout,
     const float* __restrict__ lse, const float* __restrict__ delta,
     const float* __restrict__ mask, bf16_t* __restrict__ dq, int S,
-    int causal, float scale, int has_mask, long mask_b_stride) {
+    int causal, float scale, int has_mask, long H, StridesB sio,
+    StridesB sdo, StridesB sg) {
   constexpr int D = 64;
   const int lane = threadIdx.x & (WAVE - 1);
   const int hi = lane >> 5;
@@ -218,16 +230,17 @@ __global__ __launch_bounds__(256) void attn_bwd_dq_kernel(
   const int q0 = qtile * 32;
   if (q0 >= S) return;
   const long bh = blockIdx.y;
-  const bf16_t* qp = q + bh * (long)S * D;
-  const bf16_t* kp = k + bh * (long)S * D;
-  const bf16_t* vp = v + bh * (long)S * D;
-  const bf16_t* dop = dout + bh * (long)S * D;
-  const float* mp =
-      has_mask ? (mask + (bh / mask_b_stride) * (long)S) : nullptr;
+  const long b = bh / H, h = bh % H;
+  const bf16_t* qp = q + b * sio.bs + h * sio.hs;
+  const bf16_t* kp = k + b * sio.bs + h * sio.hs;
+  const bf16_t* vp = v + b * sio.bs + h * sio.hs;
+  const bf16_t* dop = dout + b * sdo.bs + h * sdo.hs;
+  bf16_t* dqp = dq + b * sg.bs + h * sg.hs;
+  const float* mp = has_mask ? (mask + b * (long)S) : nullptr;
 
   const int qrow = q0 + j32;
-  const bf16_t* qrp = qp + (long)min(qrow, S - 1) * D;
-  const bf16_t* dorp = dop + (long)min(qrow, S - 1) * D;
+  const bf16_t* qrp = qp + (long)min(qrow, S - 1) * sio.rs;
+  const bf16_t* dorp = dop + (long)min(qrow, S - 1) * sdo.rs;
   bf16x8v qf[4], dof[4];
 #pragma unroll
   for (int s = 0; s < 4; ++s) {
@@ -246,8 +259,8 @@ __global__ __launch_bounds__(256) void attn_bwd_dq_kernel(
   const int kv_end = causal ? min(S, q0 + 32) : S;
   for (int k0 = 0; k0 < kv_end; k0 += 32) {
     const int key_l = k0 + j32;
-    const bf16_t* krp = kp + (long)min(key_l, S - 1) * D;
-    const bf16_t* vrp = vp + (long)min(key_l, S - 1) * D;
+    const bf16_t* krp = kp + (long)min(key_l, S - 1) * sio.rs;
+    const bf16_t* vrp = vp + (long)min(key_l, S - 1) * sio.rs;
     f32x16 s_acc, dp_acc;
 #pragma unroll
     for (int r = 0; r < 16; ++r) {
@@ -279,7 +292,7 @@ __global__ __launch_bounds__(256) void attn_bwd_dq_kernel(
     unsigned int pa_ds[2][4];
     acc_to_afrag(ds_acc, pa_ds);
 #pragma unroll
-    for (int h = 0; h < 2; ++h) {
+    for (int hh = 0; hh < 2; ++hh) {
 #pragma unroll
       for (int step = 0; step < 2; ++step) {
         bf16x8v kcol;
@@ -288,26 +301,55 @@ __global__ __launch_bounds__(256) void attn_bwd_dq_kernel(
           const int kk = k0 + step * 16 + hi * 8 + e;
           kcol[e] = (kk < S)
               ? *reinterpret_cast<const __bf16*>(
-                    kp + (long)kk * D + h * 32 + j32)
+                    kp + (long)kk * sio.rs + hh * 32 + j32)
               : (__bf16)0.f;
         }
-        dq_acc[h] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+        dq_acc[hh] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
             *reinterpret_cast<const bf16x8v*>(&pa_ds[step][0]), kcol,
-            dq_acc[h], 0, 0, 0);
+            dq_acc[hh], 0, 0, 0);
       }
     }
   }
 
 #pragma unroll
-  for (int h = 0; h < 2; ++h) {
+  for (int hh = 0; hh < 2; ++hh) {
 #pragma unroll
     for (int r = 0; r < 16; ++r) {
       const int row = q0 + (r & 3) + 8 * (r >> 2) + 4 * hi;
       if (row < S)
-        dq[bh * (long)S * D + (long)row * D + h * 32 + j32] =
-            f2bf(dq_acc[h][r]);
+        dqp[(long)row * sg.rs + hh * 32 + j32] = f2bf(dq_acc[hh][r]);
     }
   }
+}
+
+}  // namespace
+
+namespace {
+
+std::vector<at::Tensor> attn_bwd_impl(
+    const bf16_t* qb, const bf16_t* kb, const bf16_t* vb, const bf16_t* ob,
+    const bf16_t* dob, at::Tensor lse, const float* mask_ptr, bool has_mask,
+    bool causal, double scale, long B, long H, int S, StridesB sio,
+    StridesB sdo, StridesB sg, bf16_t* dqb, bf16_t* dkb, bf16_t* dvb,
+    const at::TensorOptions& fopt) {
+  auto delta = at::empty({B * H * (long)S}, fopt);
+  auto stream = c10::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
+  const long NR = B * H * (long)S;
+  hipLaunchKernelGGL((attn_bwd_delta_kernel<64>),
+                     dim3((NR + 3) / 4), dim3(256), 0, stream, dob, ob,
+                     delta.data_ptr<float>(), S, H, NR, sdo);
+  dim3 block(256);
+  dim3 gridk((S + 127) / 128, B * H);
+  hipLaunchKernelGGL(attn_bwd_dkv_kernel, gridk, block, 0, stream, qb, kb,
+                     vb, dob, lse.data_ptr<float>(), delta.data_ptr<float>(),
+                     mask_ptr, dkb, dvb, S, causal ? 1 : 0, (float)scale,
+                     has_mask ? 1 : 0, H, sio, sdo, sg);
+  hipLaunchKernelGGL(attn_bwd_dq_kernel, gridk, block, 0, stream, qb, kb,
+                     vb, dob, lse.data_ptr<float>(), delta.data_ptr<float>(),
+                     mask_ptr, dqb, S, causal ? 1 : 0, (float)scale,
+                     has_mask ? 1 : 0, H, sio, sdo, sg);
+  HIP_CHECK_LAST();
+  return {};
 }
 
 }  // namespace
@@ -326,7 +368,6 @@ std::vector<at::Tensor> attn_bwd(at::Tensor q, at::Tensor k, at::Tensor v,
   auto dq = at::empty_like(q);
   auto dk = at::empty_like(k);
   auto dv = at::empty_like(v);
-  auto delta = at::empty({B * H * (long)S}, q.options().dtype(at::kFloat));
 
   const bool has_mask = mask.defined() && mask.numel() > 0;
   at::Tensor mask_f;
@@ -335,35 +376,49 @@ std::vector<at::Tensor> attn_bwd(at::Tensor q, at::Tensor k, at::Tensor v,
     mask_f = mask.to(at::kFloat).reshape({B, S}).contiguous();
     mask_ptr = mask_f.data_ptr<float>();
   }
-
-  auto stream = c10::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
-  const long NR = B * H * (long)S;
-  hipLaunchKernelGGL((attn_bwd_delta_kernel<64>),
-                     dim3((NR + 3) / 4), dim3(256), 0, stream,
-                     reinterpret_cast<const bf16_t*>(dc.data_ptr()),
-                     reinterpret_cast<const bf16_t*>(o.data_ptr()),
-                     delta.data_ptr<float>(), NR);
-  dim3 block(256);
-  dim3 gridk((S + 127) / 128, B * H);
-  hipLaunchKernelGGL(attn_bwd_dkv_kernel, gridk, block, 0, stream,
-                     reinterpret_cast<const bf16_t*>(q.data_ptr()),
-                     reinterpret_cast<const bf16_t*>(k.data_ptr()),
-                     reinterpret_cast<const bf16_t*>(v.data_ptr()),
-                     reinterpret_cast<const bf16_t*>(dc.data_ptr()),
-                     lse.data_ptr<float>(), delta.data_ptr<float>(),
-                     mask_ptr, reinterpret_cast<bf16_t*>(dk.data_ptr()),
-                     reinterpret_cast<bf16_t*>(dv.data_ptr()), S,
-                     causal ? 1 : 0, (float)scale, has_mask ? 1 : 0,
-                     (long)H);
-  hipLaunchKernelGGL(attn_bwd_dq_kernel, gridk, block, 0, stream,
-                     reinterpret_cast<const bf16_t*>(q.data_ptr()),
-                     reinterpret_cast<const bf16_t*>(k.data_ptr()),
-                     reinterpret_cast<const bf16_t*>(v.data_ptr()),
-                     reinterpret_cast<const bf16_t*>(dc.data_ptr()),
-                     lse.data_ptr<float>(), delta.data_ptr<float>(),
-                     mask_ptr, reinterpret_cast<bf16_t*>(dq.data_ptr()), S,
-                     causal ? 1 : 0, (float)scale, has_mask ? 1 : 0,
-                     (long)H);
-  HIP_CHECK_LAST();
+  const StridesB sep{H * (long)S * D, (long)S * D, (long)D};
+  attn_bwd_impl(reinterpret_cast<const bf16_t*>(q.data_ptr()),
+                reinterpret_cast<const bf16_t*>(k.data_ptr()),
+                reinterpret_cast<const bf16_t*>(v.data_ptr()),
+                reinterpret_cast<const bf16_t*>(o.data_ptr()),
+                reinterpret_cast<const bf16_t*>(dc.data_ptr()), lse,
+                mask_ptr, has_mask, causal, scale, B, H, S, sep, sep, sep,
+                reinterpret_cast<bf16_t*>(dq.data_ptr()),
+                reinterpret_cast<bf16_t*>(dk.data_ptr()),
+                reinterpret_cast<bf16_t*>(dv.data_ptr()),
+                q.options().dtype(at::kFloat));
   return {dq, dk, dv};
+}
+
+// packed layout: qkv (B,S,3,H,D), o/dout (B,S,H,D) -> dqkv (B,S,3,H,D)
+at::Tensor attn_bwd_qkv(at::Tensor qkv, at::Tensor o, at::Tensor dout,
+                        at::Tensor lse, at::Tensor mask, bool causal,
+                        double scale) {
+  TORCH_CHECK(qkv.is_cuda() && qkv.is_contiguous() && o.is_contiguous());
+  TORCH_CHECK(qkv.dim() == 5 && qkv.size(2) == 3);
+  const long B = qkv.size(0), H = qkv.size(3);
+  const int S = qkv.size(1), D = qkv.size(4);
+  TORCH_CHECK(D == 64, "attn_bwd_qkv kernel: head_dim 64");
+  auto dc = dout.contiguous();
+  auto dqkv = at::empty_like(qkv);
+
+  const bool has_mask = mask.defined() && mask.numel() > 0;
+  at::Tensor mask_f;
+  const float* mask_ptr = nullptr;
+  if (has_mask) {
+    mask_f = mask.to(at::kFloat).reshape({B, S}).contiguous();
+    mask_ptr = mask_f.data_ptr<float>();
+  }
+  const long HD = H * (long)D;
+  const StridesB sp{(long)S * 3 * HD, (long)D, 3 * HD};   // qkv & dqkv
+  const StridesB sod{(long)S * HD, (long)D, HD};          // o, dout
+  const bf16_t* base = reinterpret_cast<const bf16_t*>(qkv.data_ptr());
+  bf16_t* dbase = reinterpret_cast<bf16_t*>(dqkv.data_ptr());
+  attn_bwd_impl(base, base + HD, base + 2 * HD,
+                reinterpret_cast<const bf16_t*>(o.data_ptr()),
+                reinterpret_cast<const bf16_t*>(dc.data_ptr()), lse,
+                mask_ptr, has_mask, causal, scale, B, H, S, sp, sod, sp,
+                dbase, dbase + HD, dbase + 2 * HD,
+                qkv.options().dtype(at::kFloat));
+  return dqkv;
 }

@@ -8,6 +8,9 @@ std::vector<at::Tensor> layernorm_fwd(at::Tensor x, at::Tensor w,
 std::vector<at::Tensor> layernorm_bwd(at::Tensor dy, at::Tensor x,
                                       at::Tensor w, at::Tensor mean,
                                       at::Tensor rstd);
+std::vector<at::Tensor> layernorm_add_fwd(at::Tensor a, at::Tensor b,
+                                          at::Tensor w, at::Tensor bias,
+                                          double eps);
 at::Tensor bias_gelu_fwd(at::Tensor x, at::Tensor bias);
 at::Tensor bias_gelu_bwd(at::Tensor dy, at::Tensor x, at::Tensor bias);
 std::vector<at::Tensor> dropout_fwd(at::Tensor x, double p, int64_t seed);
@@ -34,11 +37,18 @@ std::vector<at::Tensor> attn_bwd(at::Tensor q, at::Tensor k, at::Tensor v,
                                  at::Tensor o, at::Tensor dout,
                                  at::Tensor lse, at::Tensor mask, bool causal,
                                  double scale);
+std::vector<at::Tensor> attn_fwd_qkv(at::Tensor qkv, at::Tensor mask,
+                                     bool causal, double scale);
+at::Tensor attn_bwd_qkv(at::Tensor qkv, at::Tensor o, at::Tensor dout,
+                        at::Tensor lse, at::Tensor mask, bool causal,
+                        double scale);
 at::Tensor mfma_probe(at::Tensor a, at::Tensor b);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("layernorm_fwd", &layernorm_fwd, "fused LayerNorm fwd (CDNA4)");
   m.def("layernorm_bwd", &layernorm_bwd, "fused LayerNorm bwd (CDNA4)");
+  m.def("layernorm_add_fwd", &layernorm_add_fwd,
+        "fused residual-add + LayerNorm fwd");
   m.def("bias_gelu_fwd", &bias_gelu_fwd, "fused bias+GELU fwd");
   m.def("bias_gelu_bwd", &bias_gelu_bwd, "fused bias+GELU bwd");
   m.def("dropout_fwd", &dropout_fwd, "philox dropout fwd (replayable)");
@@ -50,5 +60,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("fused_lamb", &fused_lamb, "multi-tensor LAMB");
   m.def("attn_fwd", &attn_fwd, "flash attention fwd (bf16, MFMA)");
   m.def("attn_bwd", &attn_bwd, "flash attention bwd (bf16, MFMA, D=64)");
+  m.def("attn_fwd_qkv", &attn_fwd_qkv,
+        "flash attention fwd, packed (B,S,3,H,D) qkv -> (B,S,H,D)");
+  m.def("attn_bwd_qkv", &attn_bwd_qkv,
+        "flash attention bwd, packed qkv -> dqkv");
   m.def("mfma_probe", &mfma_probe, "32x32x16 bf16 MFMA layout probe");
 }

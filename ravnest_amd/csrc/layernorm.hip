@@ -99,6 +99,91 @@ __global__ void ln_fwd_kernel(const T* __restrict__ x,
   }
 }
 
+// ---------------- fused residual-add + LayerNorm forward -------------
+// y = LN(a + b); also writes s = a + b (saved for backward and as the
+// next block's residual input). Saves one full elementwise add pass
+// through HBM per LN (profile: the separate adds cost ~10ms/7 steps).
+template <typename T, typename PT>
+__global__ void ln_add_fwd_kernel(const T* __restrict__ a,
+                                  const T* __restrict__ b,
+                                  const PT* __restrict__ w,
+                                  const PT* __restrict__ bias,
+                                  T* __restrict__ y, T* __restrict__ s_out,
+                                  float* __restrict__ mean_out,
+                                  float* __restrict__ rstd_out,
+                                  int H, long N, float eps) {
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wid = threadIdx.x / WAVE;
+  const long row = (long)blockIdx.x * (blockDim.x / WAVE) + wid;
+  if (row >= N) return;
+  const T* ar = a + row * H;
+  const T* br = b + row * H;
+  T* yr = y + row * H;
+  T* sr = s_out + row * H;
+
+  float s1 = 0.f, s2 = 0.f;
+  if constexpr (sizeof(T) == 2) {
+    if ((H & 7) == 0) {
+      const int G = H >> 3;
+      for (int g = lane; g < G; g += WAVE) {
+        bf16x8 va = *reinterpret_cast<const bf16x8*>(ar + g * 8);
+        bf16x8 vb = *reinterpret_cast<const bf16x8*>(br + g * 8);
+        bf16x8 vs;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          float f = us2f((unsigned short)va[j]) + us2f((unsigned short)vb[j]);
+          vs[j] = (short)f2us(f);
+          f = us2f((unsigned short)vs[j]);  // quantized sum (bitwise saved)
+          s1 += f;
+          s2 += f * f;
+        }
+        *reinterpret_cast<bf16x8*>(sr + g * 8) = vs;
+      }
+      s1 = wave_sum(s1);
+      s2 = wave_sum(s2);
+      const float mean = s1 / H;
+      const float var = fmaxf(s2 / H - mean * mean, 0.f);
+      const float rstd = rsqrtf(var + eps);
+      if (lane == 0) {
+        mean_out[row] = mean;
+        rstd_out[row] = rstd;
+      }
+      for (int g = lane; g < G; g += WAVE) {
+        bf16x8 vs = *reinterpret_cast<const bf16x8*>(sr + g * 8);
+        bf16x8 o;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          float f = us2f((unsigned short)vs[j]);
+          o[j] = (short)f2us((f - mean) * rstd * load1(w + g * 8 + j) +
+                             load1(bias + g * 8 + j));
+        }
+        *reinterpret_cast<bf16x8*>(yr + g * 8) = o;
+      }
+      return;
+    }
+  }
+  for (int i = lane; i < H; i += WAVE) {
+    float f = load1(ar + i) + load1(br + i);
+    store1(sr + i, f);
+    f = load1(sr + i);
+    s1 += f;
+    s2 += f * f;
+  }
+  s1 = wave_sum(s1);
+  s2 = wave_sum(s2);
+  const float mean = s1 / H;
+  const float var = fmaxf(s2 / H - mean * mean, 0.f);
+  const float rstd = rsqrtf(var + eps);
+  if (lane == 0) {
+    mean_out[row] = mean;
+    rstd_out[row] = rstd;
+  }
+  for (int i = lane; i < H; i += WAVE) {
+    float f = load1(sr + i);
+    store1(yr + i, (f - mean) * rstd * load1(w + i) + load1(bias + i));
+  }
+}
+
 // ---------------- backward: dx ----------------
 template <typename T, typename PT>
 __global__ void ln_bwd_dx_kernel(const T* __restrict__ dy,
@@ -254,4 +339,45 @@ std::vector<at::Tensor> layernorm_bwd(at::Tensor dy, at::Tensor x,
   auto dw_c = dw.to(w.scalar_type());
   auto db_c = db.to(w.scalar_type());
   return {dx, dw_c, db_c};
+}
+
+
+std::vector<at::Tensor> layernorm_add_fwd(at::Tensor a, at::Tensor b,
+                                          at::Tensor w, at::Tensor bias,
+                                          double eps) {
+  TORCH_CHECK(a.is_cuda() && a.is_contiguous() && b.is_contiguous());
+  TORCH_CHECK(a.sizes() == b.sizes() && a.scalar_type() == b.scalar_type());
+  const int H = a.size(-1);
+  const long N = a.numel() / H;
+  auto y = at::empty_like(a);
+  auto s_out = at::empty_like(a);
+  auto mean = at::empty({N}, a.options().dtype(at::kFloat));
+  auto rstd = at::empty({N}, a.options().dtype(at::kFloat));
+  auto stream = c10::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
+  const int waves_per_block = 4;
+  dim3 block(WAVE * waves_per_block);
+  dim3 grid((N + waves_per_block - 1) / waves_per_block);
+
+#define DISPATCH_LN_ADD_FWD(T, PT)                                          \
+  hipLaunchKernelGGL((ln_add_fwd_kernel<T, PT>), grid, block, 0, stream,    \
+                     reinterpret_cast<const T*>(a.data_ptr()),              \
+                     reinterpret_cast<const T*>(b.data_ptr()),              \
+                     reinterpret_cast<const PT*>(w.data_ptr()),             \
+                     reinterpret_cast<const PT*>(bias.data_ptr()),          \
+                     reinterpret_cast<T*>(y.data_ptr()),                    \
+                     reinterpret_cast<T*>(s_out.data_ptr()),                \
+                     mean.data_ptr<float>(), rstd.data_ptr<float>(), H, N,  \
+                     (float)eps)
+
+  if (a.scalar_type() == at::kBFloat16 && w.scalar_type() == at::kFloat) {
+    DISPATCH_LN_ADD_FWD(bf16_t, float);
+  } else if (a.scalar_type() == at::kBFloat16) {
+    DISPATCH_LN_ADD_FWD(bf16_t, bf16_t);
+  } else if (w.scalar_type() == at::kFloat) {
+    DISPATCH_LN_ADD_FWD(float, float);
+  } else {
+    TORCH_CHECK(false, "layernorm_add_fwd: unsupported dtype combo");
+  }
+  HIP_CHECK_LAST();
+  return {y, s_out, mean, rstd};
 }

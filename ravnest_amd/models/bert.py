@@ -14,7 +14,8 @@ from __future__ import annotations
 import torch
 import torch.nn as nn
 
-from ..ops import (AttentionCore, Dropout, FusedLayerNorm, GELU, LinearGelu)
+from ..ops import (AddLayerNorm, AttentionCoreQKV, Dropout, FusedLayerNorm,
+                   GELU, LinearGelu)
 
 
 class BertConfig:
@@ -69,17 +70,17 @@ class BertSelfAttention(nn.Module):
         self.heads = cfg.heads
         self.head_dim = cfg.hidden // cfg.heads
         self.qkv = nn.Linear(cfg.hidden, 3 * cfg.hidden)
-        self.core = AttentionCore(causal=False)
+        self.core = AttentionCoreQKV(causal=False)
         self.out = nn.Linear(cfg.hidden, cfg.hidden)
         self.drop = Dropout(cfg.dropout)
 
     def forward(self, x, mask):
-        # unflatten/flatten keep the fx graph free of .size() scalar nodes
-        # (scalars routed across stage boundaries cost a message each)
+        # packed (B,S,3,H,D) qkv: the attention kernel reads the
+        # projection's natural layout and returns token-major O — no
+        # permute/contiguous copies (unflatten also keeps the fx graph
+        # free of .size() scalar nodes)
         qkv = self.qkv(x).unflatten(-1, (3, self.heads, self.head_dim))
-        qkv = qkv.permute(2, 0, 3, 1, 4)  # (3, B, H, S, D)
-        o = self.core(qkv[0], qkv[1], qkv[2], mask)
-        o = o.transpose(1, 2).flatten(2)
+        o = self.core(qkv, mask)
         return self.drop(self.out(o))
 
 
@@ -87,15 +88,15 @@ class BertLayer(nn.Module):
     def __init__(self, cfg: BertConfig):
         super().__init__()
         self.attn = BertSelfAttention(cfg)
-        self.ln1 = FusedLayerNorm(cfg.hidden, cfg.layer_norm_eps)
+        self.ln1 = AddLayerNorm(cfg.hidden, cfg.layer_norm_eps)
         self.mlp_in = LinearGelu(cfg.hidden, cfg.intermediate)
         self.mlp_out = nn.Linear(cfg.intermediate, cfg.hidden)
         self.drop = Dropout(cfg.dropout)
-        self.ln2 = FusedLayerNorm(cfg.hidden, cfg.layer_norm_eps)
+        self.ln2 = AddLayerNorm(cfg.hidden, cfg.layer_norm_eps)
 
     def forward(self, x, mask):
-        x = self.ln1(x + self.attn(x, mask))
-        x = self.ln2(x + self.drop(self.mlp_out(self.mlp_in(x))))
+        x = self.ln1(x, self.attn(x, mask))
+        x = self.ln2(x, self.drop(self.mlp_out(self.mlp_in(x))))
         return x
 
 

@@ -12,7 +12,7 @@ from __future__ import annotations
 import torch
 import torch.nn as nn
 
-from ..ops import AttentionCore, Dropout, FusedLayerNorm, LinearGelu
+from ..ops import AttentionCoreQKV, Dropout, FusedLayerNorm, LinearGelu
 
 
 class GPTConfig:
@@ -59,7 +59,7 @@ class GPTBlock(nn.Module):
         self.head_dim = cfg.n_embd // cfg.n_head
         self.ln1 = FusedLayerNorm(cfg.n_embd)
         self.qkv = nn.Linear(cfg.n_embd, 3 * cfg.n_embd)
-        self.core = AttentionCore(causal=True)
+        self.core = AttentionCoreQKV(causal=True)
         self.proj = nn.Linear(cfg.n_embd, cfg.n_embd)
         self.attn_drop = Dropout(cfg.dropout)
         self.ln2 = FusedLayerNorm(cfg.n_embd)
@@ -70,9 +70,7 @@ class GPTBlock(nn.Module):
     def forward(self, x):
         h = self.ln1(x)
         qkv = self.qkv(h).unflatten(-1, (3, self.n_head, self.head_dim))
-        qkv = qkv.permute(2, 0, 3, 1, 4)
-        o = self.core(qkv[0], qkv[1], qkv[2])
-        o = o.transpose(1, 2).flatten(2)
+        o = self.core(qkv)
         x = x + self.attn_drop(self.proj(o))
         x = x + self.mlp_drop(self.mlp_out(self.mlp_in(self.ln2(x))))
         return x

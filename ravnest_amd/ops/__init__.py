@@ -1,17 +1,17 @@
 from ._ext import get_ext, has_ext
-from .layernorm import FusedLayerNorm, layer_norm
+from .layernorm import (FusedLayerNorm, AddLayerNorm, layer_norm, add_layer_norm)
 from .activation import GELU, LinearGelu, gelu, bias_gelu
 from .dropout import Dropout, dropout
-from .attention import AttentionCore, attention
+from .attention import (AttentionCore, AttentionCoreQKV, attention, attention_qkv)
 from .losses import CrossEntropyLoss, cross_entropy
 from .optim import FusedAdam, FusedSGD, FusedLAMB
 
 __all__ = [
     "get_ext", "has_ext",
-    "FusedLayerNorm", "layer_norm",
+    "FusedLayerNorm", "AddLayerNorm", "layer_norm", "add_layer_norm",
     "GELU", "LinearGelu", "gelu", "bias_gelu",
     "Dropout", "dropout",
-    "AttentionCore", "attention",
+    "AttentionCore", "AttentionCoreQKV", "attention", "attention_qkv",
     "CrossEntropyLoss", "cross_entropy",
     "FusedAdam", "FusedSGD", "FusedLAMB",
 ]

@@ -99,3 +99,56 @@ class AttentionCore(nn.Module):
 
     def forward(self, q, k, v, mask=None):
         return attention(q, k, v, mask=mask, causal=self.causal)
+
+
+class _AttnQKVFn(torch.autograd.Function):
+    """Packed path: qkv (B,S,3,H,D) -> o (B,S,H,D). No permute/contiguous
+    copies: the kernel reads the projection's natural layout and writes O
+    in token-major order (the next Linear's input layout)."""
+
+    @staticmethod
+    def forward(ctx, qkv, mask, causal, scale):
+        ext = get_ext(required=True)
+        o, lse = ext.attn_fwd_qkv(qkv,
+                                  mask if mask is not None else torch.Tensor(),
+                                  causal, scale)
+        ctx.save_for_backward(qkv, mask if mask is not None else None, o, lse)
+        ctx.causal = causal
+        ctx.scale = scale
+        return o
+
+    @staticmethod
+    def backward(ctx, do):
+        qkv, mask, o, lse = ctx.saved_tensors
+        ext = get_ext(required=True)
+        dqkv = ext.attn_bwd_qkv(
+            qkv, o, do.contiguous(), lse,
+            mask if mask is not None else torch.Tensor(),
+            ctx.causal, ctx.scale)
+        return dqkv, None, None, None
+
+
+def attention_qkv(qkv, mask=None, causal=False, scale=None):
+    """qkv (B,S,3,H,D) -> (B,S,H*D)."""
+    H, D = qkv.shape[-2], qkv.shape[-1]
+    if scale is None:
+        scale = 1.0 / math.sqrt(D)
+    if qkv.is_cuda and D == 64:
+        o = _AttnQKVFn.apply(qkv.contiguous(), mask, causal, scale)
+        return o.flatten(2)
+    # fallback: unpack + (custom or math) attention
+    q, k, v = (qkv.permute(2, 0, 3, 1, 4)[i] for i in range(3))
+    o = attention(q.contiguous(), k.contiguous(), v.contiguous(),
+                  mask=mask, causal=causal, scale=scale)
+    return o.transpose(1, 2).flatten(2)
+
+
+class AttentionCoreQKV(nn.Module):
+    _is_leaf_module = True
+
+    def __init__(self, causal: bool = False):
+        super().__init__()
+        self.causal = causal
+
+    def forward(self, qkv, mask=None):
+        return attention_qkv(qkv, mask=mask, causal=self.causal)

@@ -49,3 +49,42 @@ class FusedLayerNorm(nn.Module):
 
     def forward(self, x):
         return layer_norm(x, self.weight, self.bias, self.eps)
+
+
+class _AddLayerNormFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, a, b, weight, bias, eps):
+        ext = get_ext(required=True)
+        y, s, mean, rstd = ext.layernorm_add_fwd(a, b, weight, bias, eps)
+        ctx.save_for_backward(s, weight, mean, rstd)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        s, weight, mean, rstd = ctx.saved_tensors
+        ext = get_ext(required=True)
+        dx, dw, db = ext.layernorm_bwd(dy.contiguous(), s, weight, mean,
+                                       rstd)
+        return dx, dx, dw, db, None
+
+
+def add_layer_norm(a, b, weight, bias, eps: float = 1e-12):
+    """y = LayerNorm(a + b) with the residual add fused into the LN read
+    pass (one HBM round-trip instead of two)."""
+    if a.is_cuda:
+        return _AddLayerNormFn.apply(a.contiguous(), b.contiguous(),
+                                     weight, bias, eps)
+    return F.layer_norm(a + b, (a.shape[-1],), weight, bias, eps)
+
+
+class AddLayerNorm(nn.Module):
+    _is_leaf_module = True
+
+    def __init__(self, hidden: int, eps: float = 1e-12):
+        super().__init__()
+        self.weight = nn.Parameter(torch.ones(hidden))
+        self.bias = nn.Parameter(torch.zeros(hidden))
+        self.eps = eps
+
+    def forward(self, a, b):
+        return add_layer_norm(a, b, self.weight, self.bias, self.eps)

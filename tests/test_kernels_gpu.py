@@ -320,3 +320,51 @@ def test_attention_bwd_kernel(dev, causal, masked, S):
                          (v.grad, v2.grad, "dv")]:
         err = (g1.float() - g2).abs().max().item()
         assert err < 8e-2, f"{name} max err {err} ({causal},{masked},{S})"
+
+
+def test_attention_qkv_packed(dev):
+    """Packed (B,S,3,H,D) path vs the separate-tensor kernel path."""
+    from ravnest_amd.ops import attention, attention_qkv
+    torch.manual_seed(0)
+    B, S, H, D = 2, 256, 4, 64
+    qkv = torch.randn(B, S, 3, H, D, device=dev, dtype=torch.bfloat16,
+                      requires_grad=True)
+    mask4 = None
+    o = attention_qkv(qkv, mask=mask4, causal=True)  # (B,S,H*D)
+    do = torch.randn_like(o)
+    o.backward(do)
+
+    qkv2 = qkv.detach().clone().requires_grad_(True)
+    q, k, v = (t.contiguous() for t in qkv2.permute(2, 0, 3, 1, 4))
+    o2 = attention(q, k, v, causal=True)  # (B,H,S,D)
+    o2 = o2.transpose(1, 2).flatten(2)
+    o2.backward(do)
+    assert torch.allclose(o.float(), o2.float(), atol=2e-2, rtol=2e-2)
+    err = (qkv.grad.float() - qkv2.grad.float()).abs().max().item()
+    assert err < 5e-2, f"dqkv err {err}"
+
+
+def test_add_layer_norm(dev):
+    from ravnest_amd.ops import add_layer_norm
+    torch.manual_seed(0)
+    H, N = 768, 512
+    a = torch.randn(N, H, device=dev, dtype=torch.bfloat16,
+                    requires_grad=True)
+    b = torch.randn(N, H, device=dev, dtype=torch.bfloat16,
+                    requires_grad=True)
+    w = torch.randn(H, device=dev, requires_grad=True)
+    bias = torch.randn(H, device=dev, requires_grad=True)
+    y = add_layer_norm(a, b, w, bias, 1e-5)
+    dy = torch.randn_like(y)
+    y.backward(dy)
+
+    a2 = a.detach().float().clone().requires_grad_(True)
+    b2 = b.detach().float().clone().requires_grad_(True)
+    w2 = w.detach().clone().requires_grad_(True)
+    bias2 = bias.detach().clone().requires_grad_(True)
+    y2 = torch.nn.functional.layer_norm(a2 + b2, (H,), w2, bias2, 1e-5)
+    y2.backward(dy.float())
+    assert torch.allclose(y.float(), y2, atol=5e-2, rtol=5e-2)
+    assert torch.allclose(a.grad.float(), a2.grad, atol=8e-2, rtol=8e-2)
+    assert torch.allclose(b.grad.float(), b2.grad, atol=8e-2, rtol=8e-2)
+    assert torch.allclose(w.grad, w2.grad, atol=1.0, rtol=3e-2)

@@ -151,8 +151,9 @@ __global__ __launch_bounds__(256) void attn_bwd_dkv_kernel(
       dp_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dof, vf[s], dp_acc,
                                                        0, 0, 0);
     }
-    // P and dS in the [qrow-pattern][key=lane] layout
-    f32x16 p_acc, ds_acc;
+    // P and dS in the [qrow-pattern][key=lane] layout — computed IN
+    // PLACE into s_acc / dp_acc (the extra register sets cost an
+    // occupancy tier: 180 VGPRs = 2 waves/SIMD)
 #pragma unroll
     for (int r = 0; r < 16; ++r) {
       const int qrow = q0 + (r & 3) + 8 * (r >> 2) + 4 * hi;
@@ -162,14 +163,15 @@ __global__ __launch_bounds__(256) void attn_bwd_dkv_kernel(
       bool dead = (qrow >= S) || (key >= S) ||
                   (causal && key > qrow) || !isfinite(l);
       const float p = dead ? 0.f : __expf(sv - l);
-      p_acc[r] = p;
-      ds_acc[r] = dead ? 0.f : p * (dp_acc[r] - dltp[qclmp]) * scale;
+      s_acc[r] = p;                     // s_acc now holds P
+      dp_acc[r] = dead ? 0.f
+                       : p * (dp_acc[r] - dltp[qclmp]) * scale;  // dS
     }
     // dV += P^T dO ; dK += dS^T Q : A = transform(acc) over k=qrows,
     // B[k=qrow][j=d] = dO/Q rows, strided scalar loads
     unsigned int pa_p[2][4], pa_ds[2][4];
-    acc_to_afrag(p_acc, pa_p);
-    acc_to_afrag(ds_acc, pa_ds);
+    acc_to_afrag(s_acc, pa_p);
+    acc_to_afrag(dp_acc, pa_ds);
 #pragma unroll
     for (int hh = 0; hh < 2; ++hh) {
 #pragma unroll
@@ -215,7 +217,7 @@ __global__ __launch_bounds__(256) void attn_bwd_dkv_kernel(
 // dq kernel: wave owns q rows [q0, q0+32); loops kv tiles (forward
 // orientation: acc = [key-pattern][qrow=lane&31]).
 // -------------------------------------------------------------------
-__global__ __launch_bounds__(256) void attn_bwd_dq_kernel(
+__global__ __launch_bounds__(256, 3) void attn_bwd_dq_kernel(
     const bf16_t* __restrict__ q, const bf16_t* __restrict__ k,
     const bf16_t* __restrict__ v, const bf16_t* __restrict__ dout,
     const float* __restrict__ lse, const float* __restrict__ delta,
@@ -277,7 +279,6 @@ __global__ __launch_bounds__(256) void attn_bwd_dq_kernel(
       dp_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vfr, dof[s], dp_acc,
                                                        0, 0, 0);
     }
-    f32x16 ds_acc;
 #pragma unroll
     for (int r = 0; r < 16; ++r) {
       const int kk = k0 + (r & 3) + 8 * (r >> 2) + 4 * hi;
@@ -286,11 +287,11 @@ __global__ __launch_bounds__(256) void attn_bwd_dq_kernel(
       bool dead = (qrow >= S) || (kk >= S) || (causal && kk > qrow) ||
                   !isfinite(l_row);
       const float p = dead ? 0.f : __expf(sv - l_row);
-      ds_acc[r] = dead ? 0.f : p * (dp_acc[r] - dlt_row) * scale;
+      dp_acc[r] = dead ? 0.f : p * (dp_acc[r] - dlt_row) * scale;  // dS
     }
     // dQ += dS K : A = transform(dS^T) over k=keys; B[k=key][j=d] = K rows
     unsigned int pa_ds[2][4];
-    acc_to_afrag(ds_acc, pa_ds);
+    acc_to_afrag(dp_acc, pa_ds);
 #pragma unroll
     for (int hh = 0; hh < 2; ++hh) {
 #pragma unroll

@@ -57,7 +57,7 @@ def main():
     ap.add_argument("--steps", type=int, default=20)
     ap.add_argument("--warmup", type=int, default=5)
     ap.add_argument("--model", default="bert-base")
-    ap.add_argument("--micro-batch", type=int, default=16)
+    ap.add_argument("--micro-batch", type=int, default=64)
     ap.add_argument("--seq", type=int, default=512)
     ap.add_argument("--reduce-factor", type=int, default=4,
                     help="average DP replicas every N steps")

@@ -214,6 +214,208 @@ __global__ __launch_bounds__(256) void attn_bwd_dkv_kernel(
 }
 
 // -------------------------------------------------------------------
+// dv kernel: wave owns keys; S -> P -> dV only (~110 VGPR, 4 waves/SIMD;
+// the combined dkv kernel sat at 180 VGPR = 2 waves/SIMD and 56% of wave
+// cycles parked on memory waits — PMC profile r01)
+// -------------------------------------------------------------------
+__global__ __launch_bounds__(256, 4) void attn_bwd_dv_kernel(
+    const bf16_t* __restrict__ q, const bf16_t* __restrict__ k,
+    const bf16_t* __restrict__ dout, const float* __restrict__ lse,
+    const float* __restrict__ mask, bf16_t* __restrict__ dv, int S,
+    int causal, float scale, int has_mask, long H, StridesB sio,
+    StridesB sdo, StridesB sg) {
+  constexpr int D = 64;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int hi = lane >> 5;
+  const int j32 = lane & 31;
+  const int k0 = (blockIdx.x * 4 + (int)(threadIdx.x / WAVE)) * 32;
+  if (k0 >= S) return;
+  const long bh = blockIdx.y;
+  const long b = bh / H, h = bh % H;
+  const bf16_t* qp = q + b * sio.bs + h * sio.hs;
+  const bf16_t* kp = k + b * sio.bs + h * sio.hs;
+  const bf16_t* dop = dout + b * sdo.bs + h * sdo.hs;
+  bf16_t* dvp = dv + b * sg.bs + h * sg.hs;
+  const float* lsep = lse + bh * (long)S;
+  const float* mp = has_mask ? (mask + b * (long)S) : nullptr;
+
+  const int key = k0 + j32;
+  const bf16_t* krp = kp + (long)min(key, S - 1) * sio.rs;
+  bf16x8v kf[4];
+#pragma unroll
+  for (int s = 0; s < 4; ++s)
+    kf[s] = *reinterpret_cast<const bf16x8v*>(krp + s * 16 + hi * 8);
+  const float mask_val = (has_mask && key < S) ? mp[key] : 0.f;
+
+  f32x16 dv_acc[2];
+#pragma unroll
+  for (int hh = 0; hh < 2; ++hh)
+#pragma unroll
+    for (int r = 0; r < 16; ++r) dv_acc[hh][r] = 0.f;
+
+  const int q_start = causal ? k0 : 0;
+  for (int q0 = q_start; q0 < S; q0 += 32) {
+    const int qrow_l = q0 + j32;
+    const bf16_t* qrp = qp + (long)min(qrow_l, S - 1) * sio.rs;
+    f32x16 s_acc;
+#pragma unroll
+    for (int r = 0; r < 16; ++r) s_acc[r] = 0.f;
+#pragma unroll
+    for (int s = 0; s < 4; ++s) {
+      bf16x8v qf = *reinterpret_cast<const bf16x8v*>(qrp + s * 16 + hi * 8);
+      s_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qf, kf[s], s_acc,
+                                                      0, 0, 0);
+    }
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      const int qrow = q0 + (r & 3) + 8 * (r >> 2) + 4 * hi;
+      const int qclmp = min(qrow, S - 1);
+      const float l = lsep[qclmp];
+      float sv = s_acc[r] * scale + mask_val;
+      bool dead = (qrow >= S) || (key >= S) || (causal && key > qrow) ||
+                  !isfinite(l);
+      s_acc[r] = dead ? 0.f : __expf(sv - l);  // P
+    }
+    unsigned int pa_p[2][4];
+    acc_to_afrag(s_acc, pa_p);
+#pragma unroll
+    for (int hh = 0; hh < 2; ++hh) {
+#pragma unroll
+      for (int step = 0; step < 2; ++step) {
+        bf16x8v dof;
+#pragma unroll
+        for (int e = 0; e < 8; ++e) {
+          const int qrow = q0 + step * 16 + hi * 8 + e;
+          dof[e] = (qrow < S)
+              ? *reinterpret_cast<const __bf16*>(
+                    dop + (long)qrow * sdo.rs + hh * 32 + j32)
+              : (__bf16)0.f;
+        }
+        dv_acc[hh] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+            *reinterpret_cast<const bf16x8v*>(&pa_p[step][0]), dof,
+            dv_acc[hh], 0, 0, 0);
+      }
+    }
+  }
+#pragma unroll
+  for (int hh = 0; hh < 2; ++hh) {
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      const int krow = k0 + (r & 3) + 8 * (r >> 2) + 4 * hi;
+      if (krow < S)
+        dvp[(long)krow * sg.rs + hh * 32 + j32] = f2bf(dv_acc[hh][r]);
+    }
+  }
+}
+
+// -------------------------------------------------------------------
+// dk kernel: wave owns keys; S, dP -> dS -> dK (~150 VGPR, 3 waves/SIMD)
+// -------------------------------------------------------------------
+__global__ __launch_bounds__(256) void attn_bwd_dk_kernel(
+    const bf16_t* __restrict__ q, const bf16_t* __restrict__ k,
+    const bf16_t* __restrict__ v, const bf16_t* __restrict__ dout,
+    const float* __restrict__ lse, const float* __restrict__ delta,
+    const float* __restrict__ mask, bf16_t* __restrict__ dk, int S,
+    int causal, float scale, int has_mask, long H, StridesB sio,
+    StridesB sdo, StridesB sg) {
+  constexpr int D = 64;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int hi = lane >> 5;
+  const int j32 = lane & 31;
+  const int k0 = (blockIdx.x * 4 + (int)(threadIdx.x / WAVE)) * 32;
+  if (k0 >= S) return;
+  const long bh = blockIdx.y;
+  const long b = bh / H, h = bh % H;
+  const bf16_t* qp = q + b * sio.bs + h * sio.hs;
+  const bf16_t* kp = k + b * sio.bs + h * sio.hs;
+  const bf16_t* vp = v + b * sio.bs + h * sio.hs;
+  const bf16_t* dop = dout + b * sdo.bs + h * sdo.hs;
+  bf16_t* dkp = dk + b * sg.bs + h * sg.hs;
+  const float* lsep = lse + bh * (long)S;
+  const float* dltp = delta + bh * (long)S;
+  const float* mp = has_mask ? (mask + b * (long)S) : nullptr;
+
+  const int key = k0 + j32;
+  const bf16_t* krp = kp + (long)min(key, S - 1) * sio.rs;
+  const bf16_t* vrp = vp + (long)min(key, S - 1) * sio.rs;
+  bf16x8v kf[4], vf[4];
+#pragma unroll
+  for (int s = 0; s < 4; ++s) {
+    kf[s] = *reinterpret_cast<const bf16x8v*>(krp + s * 16 + hi * 8);
+    vf[s] = *reinterpret_cast<const bf16x8v*>(vrp + s * 16 + hi * 8);
+  }
+  const float mask_val = (has_mask && key < S) ? mp[key] : 0.f;
+
+  f32x16 dk_acc[2];
+#pragma unroll
+  for (int hh = 0; hh < 2; ++hh)
+#pragma unroll
+    for (int r = 0; r < 16; ++r) dk_acc[hh][r] = 0.f;
+
+  const int q_start = causal ? k0 : 0;
+  for (int q0 = q_start; q0 < S; q0 += 32) {
+    const int qrow_l = q0 + j32;
+    const bf16_t* qrp = qp + (long)min(qrow_l, S - 1) * sio.rs;
+    const bf16_t* dorp = dop + (long)min(qrow_l, S - 1) * sdo.rs;
+    f32x16 s_acc, dp_acc;
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      s_acc[r] = 0.f;
+      dp_acc[r] = 0.f;
+    }
+#pragma unroll
+    for (int s = 0; s < 4; ++s) {
+      bf16x8v qf = *reinterpret_cast<const bf16x8v*>(qrp + s * 16 + hi * 8);
+      bf16x8v dof = *reinterpret_cast<const bf16x8v*>(dorp + s * 16 + hi * 8);
+      s_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qf, kf[s], s_acc,
+                                                      0, 0, 0);
+      dp_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dof, vf[s], dp_acc,
+                                                       0, 0, 0);
+    }
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      const int qrow = q0 + (r & 3) + 8 * (r >> 2) + 4 * hi;
+      const int qclmp = min(qrow, S - 1);
+      const float l = lsep[qclmp];
+      float sv = s_acc[r] * scale + mask_val;
+      bool dead = (qrow >= S) || (key >= S) || (causal && key > qrow) ||
+                  !isfinite(l);
+      const float p = dead ? 0.f : __expf(sv - l);
+      dp_acc[r] = dead ? 0.f : p * (dp_acc[r] - dltp[qclmp]) * scale;
+    }
+    unsigned int pa_ds[2][4];
+    acc_to_afrag(dp_acc, pa_ds);
+#pragma unroll
+    for (int hh = 0; hh < 2; ++hh) {
+#pragma unroll
+      for (int step = 0; step < 2; ++step) {
+        bf16x8v qf2;
+#pragma unroll
+        for (int e = 0; e < 8; ++e) {
+          const int qrow = q0 + step * 16 + hi * 8 + e;
+          qf2[e] = (qrow < S)
+              ? *reinterpret_cast<const __bf16*>(
+                    qp + (long)qrow * sio.rs + hh * 32 + j32)
+              : (__bf16)0.f;
+        }
+        dk_acc[hh] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+            *reinterpret_cast<const bf16x8v*>(&pa_ds[step][0]), qf2,
+            dk_acc[hh], 0, 0, 0);
+      }
+    }
+  }
+#pragma unroll
+  for (int hh = 0; hh < 2; ++hh) {
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      const int krow = k0 + (r & 3) + 8 * (r >> 2) + 4 * hi;
+      if (krow < S)
+        dkp[(long)krow * sg.rs + hh * 32 + j32] = f2bf(dk_acc[hh][r]);
+    }
+  }
+}
+
+// -------------------------------------------------------------------
 // dq kernel: wave owns q rows [q0, q0+32); loops kv tiles (forward
 // orientation: acc = [key-pattern][qrow=lane&31]).
 // -------------------------------------------------------------------
@@ -341,9 +543,13 @@ std::vector<at::Tensor> attn_bwd_impl(
                      delta.data_ptr<float>(), S, H, NR, sdo);
   dim3 block(256);
   dim3 gridk((S + 127) / 128, B * H);
-  hipLaunchKernelGGL(attn_bwd_dkv_kernel, gridk, block, 0, stream, qb, kb,
+  hipLaunchKernelGGL(attn_bwd_dv_kernel, gridk, block, 0, stream, qb, kb,
+                     dob, lse.data_ptr<float>(), mask_ptr, dvb, S,
+                     causal ? 1 : 0, (float)scale, has_mask ? 1 : 0, H, sio,
+                     sdo, sg);
+  hipLaunchKernelGGL(attn_bwd_dk_kernel, gridk, block, 0, stream, qb, kb,
                      vb, dob, lse.data_ptr<float>(), delta.data_ptr<float>(),
-                     mask_ptr, dkb, dvb, S, causal ? 1 : 0, (float)scale,
+                     mask_ptr, dkb, S, causal ? 1 : 0, (float)scale,
                      has_mask ? 1 : 0, H, sio, sdo, sg);
   hipLaunchKernelGGL(attn_bwd_dq_kernel, gridk, block, 0, stream, qb, kb,
                      vb, dob, lse.data_ptr<float>(), delta.data_ptr<float>(),

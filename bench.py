@@ -210,7 +210,12 @@ def main():
     if comm is not None:
         comm.stop()
         torch.distributed.barrier()
-        torch.distributed.destroy_process_group()
+        sys.stdout.flush()
+        sys.stderr.flush()
+        # gloo/RCCL teardown with live helper threads can abort on exit;
+        # results are printed, so leave hard (single-rank keeps the clean
+        # exit that rocprof finalization needs)
+        os._exit(0)
 
 
 if __name__ == "__main__":

@@ -30,6 +30,12 @@ class GPTConfig:
         return cls(vocab_size, block_size, n_layer=3, n_head=3, n_embd=48)
 
     @classmethod
+    def nano64(cls, vocab_size=16, block_size=16):
+        """gpt-nano scale with head_dim 64 (the GPU attention kernel's
+        native size)."""
+        return cls(vocab_size, block_size, n_layer=3, n_head=3, n_embd=192)
+
+    @classmethod
     def gpt2_small(cls, vocab_size=50257, block_size=1024):
         return cls(vocab_size, block_size, n_layer=12, n_head=12, n_embd=768)
 

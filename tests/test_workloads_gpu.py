@@ -81,21 +81,23 @@ def test_bert_loss_decreases(dev):
     """Native-kernel BERT must actually learn (memorize a small batch)."""
     from ravnest_amd import set_seed
     from ravnest_amd.models import BertConfig, BertForMLM
-    from ravnest_amd.ops import FusedLAMB, cross_entropy
+    from ravnest_amd.ops import FusedAdam, cross_entropy
     set_seed(0)
     cfg = BertConfig.tiny(max_seq=64)
     cfg.dropout = 0.0
     m = BertForMLM(cfg).to(dev).to(torch.bfloat16)
-    opt = FusedLAMB(m.parameters(), lr=2e-3)
+    opt = FusedAdam(m.parameters(), lr=2e-3)
     ids = torch.randint(0, cfg.vocab_size, (16, 64), device=dev)
     mask = torch.ones_like(ids)
     losses = []
-    for step in range(60):
+    for step in range(80):
         logits = m(ids, mask)
         loss = cross_entropy(logits.reshape(-1, logits.size(-1)),
                              ids.reshape(-1))
         loss.backward()
         opt.step()
         opt.zero_grad(set_to_none=True)
-        losses.append(float(loss))
-    assert losses[-1] < losses[0] * 0.5, losses[::10]
+        losses.append(float(loss.detach()))
+    # CPU fp32 reference trajectory reaches ~0.01 by step 80; allow bf16
+    # headroom but require real memorization
+    assert losses[-1] < 1.0, losses[::10]

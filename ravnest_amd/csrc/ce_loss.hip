@@ -19,8 +19,17 @@ DEVINL float ld<bf16_t>(const bf16_t* p) { return bf2f(*p); }
 template <>
 DEVINL float ld<float>(const float* p) { return *p; }
 
-// one block per row: two passes over V (max, then sum-exp) with
-// vectorized loads; writes lse and the row's loss contribution.
+// one block per row, ONE pass over V: per-thread online (max, sum)
+// state merged wave- then block-wide ((m,s) pairs combine as
+// s = s1*exp(m1-m) + s2*exp(m2-m)). Halves the V reads of the naive
+// two-pass version (the 30522-vocab read is the whole cost).
+DEVINL void ms_merge(float& m, float& s, float m2, float s2) {
+  const float mn = fmaxf(m, m2);
+  if (mn == -INFINITY) { m = mn; return; }
+  s = s * __expf(m - mn) + s2 * __expf(m2 - mn);
+  m = mn;
+}
+
 template <typename T>
 __global__ void ce_fwd_kernel(const T* __restrict__ logits,
                               const long* __restrict__ targets,
@@ -28,21 +37,58 @@ __global__ void ce_fwd_kernel(const T* __restrict__ logits,
                               float* __restrict__ loss_accum,
                               int* __restrict__ valid_accum, int V, long N,
                               long ignore_index) {
-  __shared__ float scratch[16];
+  __shared__ float scratch_m[16];
+  __shared__ float scratch_s[16];
   const long row = blockIdx.x;
   if (row >= N) return;
   const T* lr = logits + row * V;
   const long tgt = targets[row];
 
-  float m = -INFINITY;
-  for (int i = threadIdx.x; i < V; i += blockDim.x)
-    m = fmaxf(m, ld(lr + i));
-  m = block_max(m, scratch);
-
-  float s = 0.f;
-  for (int i = threadIdx.x; i < V; i += blockDim.x)
-    s += __expf(ld(lr + i) - m);
-  s = block_sum(s, scratch);
+  float m = -INFINITY, s = 0.f;
+  for (int i = threadIdx.x; i < V; i += blockDim.x) {
+    const float v = ld(lr + i);
+    if (v > m) {
+      s = s * __expf(m - v) + 1.f;
+      m = v;
+    } else {
+      s += __expf(v - m);
+    }
+  }
+  // wave merge
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) {
+    float m2 = __shfl_down(m, off, WAVE);
+    float s2 = __shfl_down(s, off, WAVE);
+    ms_merge(m, s, m2, s2);
+  }
+  // block merge via LDS
+  {
+    const int lane = threadIdx.x & (WAVE - 1);
+    const int wid = threadIdx.x / WAVE;
+    const int nw = (blockDim.x + WAVE - 1) / WAVE;
+    if (lane == 0) {
+      scratch_m[wid] = m;
+      scratch_s[wid] = s;
+    }
+    __syncthreads();
+    if (wid == 0) {
+      m = (threadIdx.x < nw) ? scratch_m[threadIdx.x] : -INFINITY;
+      s = (threadIdx.x < nw) ? scratch_s[threadIdx.x] : 0.f;
+#pragma unroll
+      for (int off = 32; off > 0; off >>= 1) {
+        float m2 = __shfl_down(m, off, WAVE);
+        float s2 = __shfl_down(s, off, WAVE);
+        ms_merge(m, s, m2, s2);
+      }
+      if (threadIdx.x == 0) {
+        scratch_m[0] = m;
+        scratch_s[0] = s;
+      }
+    }
+    __syncthreads();
+    m = scratch_m[0];
+    s = scratch_s[0];
+  }
   const float lse = m + __logf(s);
 
   if (threadIdx.x == 0) {

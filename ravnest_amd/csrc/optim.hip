@@ -48,18 +48,69 @@ DEVINL void stv(void* p, int i, float v) {
 }
 
 template <typename T>
+DEVINL void adam_one(float& p, float g, float& m, float& v, float lr,
+                     float b1, float b2, float eps, float wd, float bc1,
+                     float bc2) {
+  if (wd != 0.f) g += wd * p;
+  m = b1 * m + (1.f - b1) * g;
+  v = b2 * v + (1.f - b2) * g * g;
+  p -= lr * (m / bc1) / (sqrtf(v / bc2) + eps);
+}
+
+// vectorized: 4 elements/thread/iter with 16-byte fp32 state accesses
+// (m/v/master are the traffic; chunks are 64Ki elements so the body is
+// always aligned and the tail only exists in the last chunk)
+template <typename T>
 __global__ void adam_mt_kernel(const Chunk* __restrict__ chunks, float lr,
                                float b1, float b2, float eps, float wd,
                                float bc1, float bc2) {
   const Chunk c = chunks[blockIdx.x];
-  for (int i = threadIdx.x; i < c.n; i += blockDim.x) {
+  const int n4 = c.n >> 2;
+  float4* m4 = reinterpret_cast<float4*>(c.m);
+  float4* v4 = reinterpret_cast<float4*>(c.v);
+  float4* w4 = reinterpret_cast<float4*>(c.master);
+  for (int i = threadIdx.x; i < n4; i += blockDim.x) {
+    float4 m = m4[i], v = v4[i];
+    float4 p;
+    float g[4];
+    if constexpr (sizeof(T) == 2) {
+      unsigned long long graw =
+          reinterpret_cast<const unsigned long long*>(c.g)[i];
+#pragma unroll
+      for (int j = 0; j < 4; ++j)
+        g[j] = us2f((unsigned short)(graw >> (16 * j)));
+      p = w4[i];
+    } else {
+      const float4 gv = reinterpret_cast<const float4*>(c.g)[i];
+      g[0] = gv.x; g[1] = gv.y; g[2] = gv.z; g[3] = gv.w;
+      p = reinterpret_cast<const float4*>(c.p)[i];
+    }
+    adam_one<T>(p.x, g[0], m.x, v.x, lr, b1, b2, eps, wd, bc1, bc2);
+    adam_one<T>(p.y, g[1], m.y, v.y, lr, b1, b2, eps, wd, bc1, bc2);
+    adam_one<T>(p.z, g[2], m.z, v.z, lr, b1, b2, eps, wd, bc1, bc2);
+    adam_one<T>(p.w, g[3], m.w, v.w, lr, b1, b2, eps, wd, bc1, bc2);
+    m4[i] = m;
+    v4[i] = v;
+    if constexpr (sizeof(T) == 2) {
+      w4[i] = p;
+      unsigned long long praw =
+          ((unsigned long long)f2us(p.x)) |
+          ((unsigned long long)f2us(p.y) << 16) |
+          ((unsigned long long)f2us(p.z) << 32) |
+          ((unsigned long long)f2us(p.w) << 48);
+      reinterpret_cast<unsigned long long*>(c.p)[i] = praw;
+    } else {
+      reinterpret_cast<float4*>(c.p)[i] = p;
+    }
+  }
+  // scalar tail
+  for (int i = (n4 << 2) + threadIdx.x; i < c.n; i += blockDim.x) {
     float g = ldv<T>(c.g, i);
     float p = c.master ? c.master[i] : ldv<T>(c.p, i);
-    if (wd != 0.f) g += wd * p;
-    float m = c.m[i] = b1 * c.m[i] + (1.f - b1) * g;
-    float v = c.v[i] = b2 * c.v[i] + (1.f - b2) * g * g;
-    const float denom = sqrtf(v / bc2) + eps;
-    p -= lr * (m / bc1) / denom;
+    float m = c.m[i], v = c.v[i];
+    adam_one<T>(p, g, m, v, lr, b1, b2, eps, wd, bc1, bc2);
+    c.m[i] = m;
+    c.v[i] = v;
     if (c.master) c.master[i] = p;
     stv<T>(c.p, i, p);
   }

@@ -217,6 +217,186 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
         (l_run > 0.f) ? m_run + __logf(l_run) : -INFINITY;
 }
 
+// ---------------------------------------------------------------------
+// fwd v2 (D=64): K/V tiles staged once per BLOCK through LDS.
+// The v1 kernel's four waves each re-read every K/V tile from L2 (16B
+// fragment loads + 32 scalar V loads per tile) and sat 38% of wave
+// cycles in memory waits (PMC r01). Here 256 threads cooperatively load
+// the 4KB K and V tiles, K stored with the ((row&15)<<4) XOR swizzle so
+// the wave's ds_read_b128 fragment reads are bank-conflict-free (guide
+// G4/T2: the b128 16-lane groups read rows distinct mod 16), V stored
+// linear and consumed as 2B LDS reads (16 consecutive banks per group).
+// ---------------------------------------------------------------------
+__global__ __launch_bounds__(256) void attn_fwd_lds_kernel(
+    const bf16_t* __restrict__ q, const bf16_t* __restrict__ k,
+    const bf16_t* __restrict__ v, const float* __restrict__ mask,
+    bf16_t* __restrict__ o, float* __restrict__ lse, int S, int causal,
+    float scale, int has_mask, long H,
+    Strides sq, Strides sk, Strides sv, Strides so) {
+  constexpr int D = 64;
+  __shared__ __attribute__((aligned(16))) char smem[8192];
+  bf16_t* k_lds = reinterpret_cast<bf16_t*>(smem);          // [32][64] swz
+  bf16_t* v_lds = reinterpret_cast<bf16_t*>(smem + 4096);   // [32][64]
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wid = threadIdx.x / WAVE;
+  const int hi = lane >> 5;
+  const int j32 = lane & 31;
+
+  const int q0 = (blockIdx.x * 4 + wid) * 32;
+  const long bh = blockIdx.y;
+  const long b = bh / H, h = bh % H;
+  const bf16_t* qp = q + b * sq.bs + h * sq.hs;
+  const bf16_t* kp = k + b * sk.bs + h * sk.hs;
+  const bf16_t* vp = v + b * sv.bs + h * sv.hs;
+  bf16_t* op = o + b * so.bs + h * so.hs;
+  const float* mp = has_mask ? (mask + b * (long)S) : nullptr;
+  const bool live_wave = q0 < S;
+
+  bf16x8v qf[4];
+  const int qrow = q0 + j32;
+  if (live_wave) {
+    const bf16_t* qrp = qp + (long)min(qrow, S - 1) * sq.rs;
+#pragma unroll
+    for (int s = 0; s < 4; ++s)
+      qf[s] = *reinterpret_cast<const bf16x8v*>(qrp + s * 16 + hi * 8);
+  }
+
+  f32x16 oacc[2];
+#pragma unroll
+  for (int hh = 0; hh < 2; ++hh)
+#pragma unroll
+    for (int r = 0; r < 16; ++r) oacc[hh][r] = 0.f;
+  float m_run = -INFINITY, l_run = 0.f;
+
+  // staging indices: thread t loads 16B of row (t>>3), col bytes (t&7)*16
+  const int st_row = threadIdx.x >> 3;
+  const int st_c16 = (threadIdx.x & 7) * 16;  // byte col
+  const int k_dst = (st_row * 128 + st_c16) ^ ((st_row & 15) << 4);
+  const int v_dst = st_row * 128 + st_c16;
+
+  // every wave must loop over ALL tiles (barriers are block-wide); a
+  // wave past the causal horizon just skips its compute
+  const int kv_all = causal ? min(S, blockIdx.x * 128 + 128) : S;
+  for (int k0 = 0; k0 < kv_all; k0 += 32) {
+    // ---- cooperative stage ----
+    const int grow = min(k0 + st_row, S - 1);
+    bf16x8v kstage = *reinterpret_cast<const bf16x8v*>(
+        kp + (long)grow * sk.rs + st_c16 / 2);
+    bf16x8v vstage = *reinterpret_cast<const bf16x8v*>(
+        vp + (long)grow * sv.rs + st_c16 / 2);
+    __syncthreads();  // previous tile fully consumed
+    *reinterpret_cast<bf16x8v*>(
+        reinterpret_cast<char*>(k_lds) + k_dst) = kstage;
+    *reinterpret_cast<bf16x8v*>(
+        reinterpret_cast<char*>(v_lds) + v_dst) = vstage;
+    __syncthreads();
+    const bool compute = live_wave && (!causal || k0 <= q0 + 31);
+    if (compute) {
+      // ---- QK^T: K A-frags from swizzled LDS (ds_read_b128) ----
+      f32x16 s_acc;
+#pragma unroll
+      for (int r = 0; r < 16; ++r) s_acc[r] = 0.f;
+#pragma unroll
+      for (int s = 0; s < 4; ++s) {
+        const int koff = (j32 * 128 + (s * 16 + hi * 8) * 2) ^
+                         ((j32 & 15) << 4);
+        bf16x8v kf = *reinterpret_cast<const bf16x8v*>(
+            reinterpret_cast<char*>(k_lds) + koff);
+        s_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qf[s], s_acc,
+                                                        0, 0, 0);
+      }
+      float tile_max = -INFINITY;
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int key = k0 + (r & 3) + 8 * (r >> 2) + 4 * hi;
+        float sv2 = s_acc[r] * scale;
+        if (key >= S) sv2 = -INFINITY;
+        if (causal && key > qrow) sv2 = -INFINITY;
+        if (has_mask) sv2 += mp[min(key, S - 1)];
+        s_acc[r] = sv2;
+        tile_max = fmaxf(tile_max, sv2);
+      }
+      tile_max = fmaxf(tile_max, __shfl_xor(tile_max, 32, WAVE));
+      const float m_new = fmaxf(m_run, tile_max);
+      float alpha = 1.f;
+      if (m_new != m_run) {
+        alpha = (m_run == -INFINITY) ? 0.f : __expf(m_run - m_new);
+        m_run = m_new;
+      }
+      l_run *= alpha;
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int row_local = (r & 3) + 8 * (r >> 2) + 4 * hi;
+        const float a_r = __shfl(alpha, row_local, WAVE);
+#pragma unroll
+        for (int hh = 0; hh < 2; ++hh) oacc[hh][r] *= a_r;
+      }
+      float psum = 0.f;
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        float pv = (s_acc[r] == -INFINITY) ? 0.f : __expf(s_acc[r] - m_run);
+        s_acc[r] = pv;
+        psum += pv;
+      }
+      psum += __shfl_xor(psum, 32, WAVE);
+      l_run += psum;
+
+      unsigned int pa[2][4];
+#pragma unroll
+      for (int step = 0; step < 2; ++step) {
+        const int b0 = step * 8;
+        unsigned int c01 = pack_bf2(s_acc[b0 + 0], s_acc[b0 + 1]);
+        unsigned int c23 = pack_bf2(s_acc[b0 + 2], s_acc[b0 + 3]);
+        unsigned int c45 = pack_bf2(s_acc[b0 + 4], s_acc[b0 + 5]);
+        unsigned int c67 = pack_bf2(s_acc[b0 + 6], s_acc[b0 + 7]);
+        {
+          auto r2 = __builtin_amdgcn_permlane32_swap(c01, c45, false, false);
+          pa[step][0] = r2[0];
+          pa[step][2] = r2[1];
+        }
+        {
+          auto r2 = __builtin_amdgcn_permlane32_swap(c23, c67, false, false);
+          pa[step][1] = r2[0];
+          pa[step][3] = r2[1];
+        }
+      }
+      // ---- PV: V B-frags from LDS (2B reads, 16-bank groups) ----
+#pragma unroll
+      for (int hh = 0; hh < 2; ++hh) {
+#pragma unroll
+        for (int step = 0; step < 2; ++step) {
+          bf16x8v vf;
+#pragma unroll
+          for (int e = 0; e < 8; ++e) {
+            const int key_l = step * 16 + hi * 8 + e;
+            vf[e] = v_lds[key_l * 64 + hh * 32 + j32];
+          }
+          oacc[hh] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              *reinterpret_cast<const bf16x8v*>(&pa[step][0]), vf,
+              oacc[hh], 0, 0, 0);
+        }
+      }
+    }
+  }
+
+  if (!live_wave) return;
+  const float l_safe = (l_run > 0.f) ? l_run : 1.f;
+#pragma unroll
+  for (int hh = 0; hh < 2; ++hh) {
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      const int row_local = (r & 3) + 8 * (r >> 2) + 4 * hi;
+      const float l_row = __shfl(l_safe, row_local, WAVE);
+      const int row = q0 + row_local;
+      if (row < S)
+        op[(long)row * so.rs + hh * 32 + j32] = f2bf(oacc[hh][r] / l_row);
+    }
+  }
+  if (qrow < S && hi == 0)
+    lse[bh * (long)S + qrow] =
+        (l_run > 0.f) ? m_run + __logf(l_run) : -INFINITY;
+}
+
 // ---- layout probe: one wave computes one 32x32x16 tile --------------
 __global__ void mfma_probe_kernel(const bf16_t* __restrict__ a,
                                   const bf16_t* __restrict__ b,
@@ -244,6 +424,11 @@ __global__ void mfma_probe_kernel(const bf16_t* __restrict__ a,
 }
 
 }  // namespace
+
+static const bool use_lds_fwd = [] {
+  const char* e = getenv("RAVNEST_ATTN_FWD_V1");
+  return !(e && e[0] == '1');  // default: the LDS kernel for D=64
+}();
 
 std::vector<at::Tensor> attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v,
                                  at::Tensor mask, bool causal, double scale) {
@@ -274,10 +459,18 @@ std::vector<at::Tensor> attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v,
   const Strides so = sep;
 
 #define LAUNCH_ATTN_FWD(DD, QP, KP, VP, OP, SQ, SK, SV, SO)                 \
-  hipLaunchKernelGGL((attn_fwd_kernel<DD>), grid, block, 0, stream, QP, KP, \
-                     VP, mask_ptr, OP, lse.data_ptr<float>(), S,            \
-                     causal ? 1 : 0, (float)scale, has_mask ? 1 : 0,        \
-                     (long)H, SQ, SK, SV, SO)
+  do {                                                                      \
+    if (DD == 64 && use_lds_fwd)                                            \
+      hipLaunchKernelGGL(attn_fwd_lds_kernel, grid, block, 0, stream, QP,   \
+                         KP, VP, mask_ptr, OP, lse.data_ptr<float>(), S,    \
+                         causal ? 1 : 0, (float)scale, has_mask ? 1 : 0,    \
+                         (long)H, SQ, SK, SV, SO);                          \
+    else                                                                    \
+      hipLaunchKernelGGL((attn_fwd_kernel<DD>), grid, block, 0, stream, QP, \
+                         KP, VP, mask_ptr, OP, lse.data_ptr<float>(), S,    \
+                         causal ? 1 : 0, (float)scale, has_mask ? 1 : 0,    \
+                         (long)H, SQ, SK, SV, SO);                          \
+  } while (0)
 
   if (D == 64) {
     LAUNCH_ATTN_FWD(64, reinterpret_cast<const bf16_t*>(q.data_ptr()),

@@ -225,11 +225,18 @@ __global__ __launch_bounds__(256, 4) void attn_bwd_dv_kernel(
     int causal, float scale, int has_mask, long H, StridesB sio,
     StridesB sdo, StridesB sg) {
   constexpr int D = 64;
+  // block-shared dO tile: the B-operand of the dV MFMA needs
+  // row-scattered 2B elements — staged once per block (4KB) instead of
+  // 32 strided global scalars per wave per tile (fwd-v2 recipe)
+  __shared__ __attribute__((aligned(16))) char smem[4096];
+  bf16_t* do_lds = reinterpret_cast<bf16_t*>(smem);
+  const int st_row = threadIdx.x >> 3;
+  const int st_c16 = (threadIdx.x & 7) * 16;
   const int lane = threadIdx.x & (WAVE - 1);
   const int hi = lane >> 5;
   const int j32 = lane & 31;
   const int k0 = (blockIdx.x * 4 + (int)(threadIdx.x / WAVE)) * 32;
-  if (k0 >= S) return;
+  const bool live_wave = k0 < S;
   const long bh = blockIdx.y;
   const long b = bh / H, h = bh % H;
   const bf16_t* qp = q + b * sio.bs + h * sio.hs;
@@ -242,10 +249,12 @@ __global__ __launch_bounds__(256, 4) void attn_bwd_dv_kernel(
   const int key = k0 + j32;
   const bf16_t* krp = kp + (long)min(key, S - 1) * sio.rs;
   bf16x8v kf[4];
+  if (live_wave) {
 #pragma unroll
-  for (int s = 0; s < 4; ++s)
-    kf[s] = *reinterpret_cast<const bf16x8v*>(krp + s * 16 + hi * 8);
-  const float mask_val = (has_mask && key < S) ? mp[key] : 0.f;
+    for (int s = 0; s < 4; ++s)
+      kf[s] = *reinterpret_cast<const bf16x8v*>(krp + s * 16 + hi * 8);
+  }
+  const float mask_val = (has_mask && key < S) ? mp[min(key, S - 1)] : 0.f;
 
   f32x16 dv_acc[2];
 #pragma unroll
@@ -253,8 +262,17 @@ __global__ __launch_bounds__(256, 4) void attn_bwd_dv_kernel(
 #pragma unroll
     for (int r = 0; r < 16; ++r) dv_acc[hh][r] = 0.f;
 
-  const int q_start = causal ? k0 : 0;
+  const int q_start = causal ? blockIdx.x * 128 : 0;
   for (int q0 = q_start; q0 < S; q0 += 32) {
+    const int srow = min(q0 + st_row, S - 1);
+    bf16x8v dstage = *reinterpret_cast<const bf16x8v*>(
+        dop + (long)srow * sdo.rs + st_c16 / 2);
+    __syncthreads();
+    *reinterpret_cast<bf16x8v*>(
+        reinterpret_cast<char*>(do_lds) + st_row * 128 + st_c16) = dstage;
+    __syncthreads();
+    const bool compute = live_wave && (!causal || q0 + 31 >= k0);
+    if (!compute) continue;
     const int qrow_l = q0 + j32;
     const bf16_t* qrp = qp + (long)min(qrow_l, S - 1) * sio.rs;
     f32x16 s_acc;
@@ -284,19 +302,15 @@ __global__ __launch_bounds__(256, 4) void attn_bwd_dv_kernel(
       for (int step = 0; step < 2; ++step) {
         bf16x8v dof;
 #pragma unroll
-        for (int e = 0; e < 8; ++e) {
-          const int qrow = q0 + step * 16 + hi * 8 + e;
-          dof[e] = (qrow < S)
-              ? *reinterpret_cast<const __bf16*>(
-                    dop + (long)qrow * sdo.rs + hh * 32 + j32)
-              : (__bf16)0.f;
-        }
+        for (int e = 0; e < 8; ++e)
+          dof[e] = do_lds[(step * 16 + hi * 8 + e) * 64 + hh * 32 + j32];
         dv_acc[hh] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
             *reinterpret_cast<const bf16x8v*>(&pa_p[step][0]), dof,
             dv_acc[hh], 0, 0, 0);
       }
     }
   }
+  if (!live_wave) return;
 #pragma unroll
   for (int hh = 0; hh < 2; ++hh) {
 #pragma unroll
@@ -319,11 +333,15 @@ __global__ __launch_bounds__(256) void attn_bwd_dk_kernel(
     int causal, float scale, int has_mask, long H, StridesB sio,
     StridesB sdo, StridesB sg) {
   constexpr int D = 64;
+  __shared__ __attribute__((aligned(16))) char smem[4096];
+  bf16_t* q_lds = reinterpret_cast<bf16_t*>(smem);  // B-operand Q tile
+  const int st_row = threadIdx.x >> 3;
+  const int st_c16 = (threadIdx.x & 7) * 16;
   const int lane = threadIdx.x & (WAVE - 1);
   const int hi = lane >> 5;
   const int j32 = lane & 31;
   const int k0 = (blockIdx.x * 4 + (int)(threadIdx.x / WAVE)) * 32;
-  if (k0 >= S) return;
+  const bool live_wave = k0 < S;
   const long bh = blockIdx.y;
   const long b = bh / H, h = bh % H;
   const bf16_t* qp = q + b * sio.bs + h * sio.hs;
@@ -339,12 +357,14 @@ __global__ __launch_bounds__(256) void attn_bwd_dk_kernel(
   const bf16_t* krp = kp + (long)min(key, S - 1) * sio.rs;
   const bf16_t* vrp = vp + (long)min(key, S - 1) * sio.rs;
   bf16x8v kf[4], vf[4];
+  if (live_wave) {
 #pragma unroll
-  for (int s = 0; s < 4; ++s) {
-    kf[s] = *reinterpret_cast<const bf16x8v*>(krp + s * 16 + hi * 8);
-    vf[s] = *reinterpret_cast<const bf16x8v*>(vrp + s * 16 + hi * 8);
+    for (int s = 0; s < 4; ++s) {
+      kf[s] = *reinterpret_cast<const bf16x8v*>(krp + s * 16 + hi * 8);
+      vf[s] = *reinterpret_cast<const bf16x8v*>(vrp + s * 16 + hi * 8);
+    }
   }
-  const float mask_val = (has_mask && key < S) ? mp[key] : 0.f;
+  const float mask_val = (has_mask && key < S) ? mp[min(key, S - 1)] : 0.f;
 
   f32x16 dk_acc[2];
 #pragma unroll
@@ -352,8 +372,17 @@ __global__ __launch_bounds__(256) void attn_bwd_dk_kernel(
 #pragma unroll
     for (int r = 0; r < 16; ++r) dk_acc[hh][r] = 0.f;
 
-  const int q_start = causal ? k0 : 0;
+  const int q_start = causal ? blockIdx.x * 128 : 0;
   for (int q0 = q_start; q0 < S; q0 += 32) {
+    const int srow = min(q0 + st_row, S - 1);
+    bf16x8v qstage = *reinterpret_cast<const bf16x8v*>(
+        qp + (long)srow * sio.rs + st_c16 / 2);
+    __syncthreads();
+    *reinterpret_cast<bf16x8v*>(
+        reinterpret_cast<char*>(q_lds) + st_row * 128 + st_c16) = qstage;
+    __syncthreads();
+    const bool compute = live_wave && (!causal || q0 + 31 >= k0);
+    if (!compute) continue;
     const int qrow_l = q0 + j32;
     const bf16_t* qrp = qp + (long)min(qrow_l, S - 1) * sio.rs;
     const bf16_t* dorp = dop + (long)min(qrow_l, S - 1) * sdo.rs;
@@ -391,19 +420,15 @@ __global__ __launch_bounds__(256) void attn_bwd_dk_kernel(
       for (int step = 0; step < 2; ++step) {
         bf16x8v qf2;
 #pragma unroll
-        for (int e = 0; e < 8; ++e) {
-          const int qrow = q0 + step * 16 + hi * 8 + e;
-          qf2[e] = (qrow < S)
-              ? *reinterpret_cast<const __bf16*>(
-                    qp + (long)qrow * sio.rs + hh * 32 + j32)
-              : (__bf16)0.f;
-        }
+        for (int e = 0; e < 8; ++e)
+          qf2[e] = q_lds[(step * 16 + hi * 8 + e) * 64 + hh * 32 + j32];
         dk_acc[hh] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
             *reinterpret_cast<const bf16x8v*>(&pa_ds[step][0]), qf2,
             dk_acc[hh], 0, 0, 0);
       }
     }
   }
+  if (!live_wave) return;
 #pragma unroll
   for (int hh = 0; hh < 2; ++hh) {
 #pragma unroll
@@ -427,12 +452,16 @@ __global__ __launch_bounds__(256, 3) void attn_bwd_dq_kernel(
     int causal, float scale, int has_mask, long H, StridesB sio,
     StridesB sdo, StridesB sg) {
   constexpr int D = 64;
+  __shared__ __attribute__((aligned(16))) char smem[4096];
+  bf16_t* k_ldsb = reinterpret_cast<bf16_t*>(smem);  // B-operand K tile
+  const int st_row = threadIdx.x >> 3;
+  const int st_c16 = (threadIdx.x & 7) * 16;
   const int lane = threadIdx.x & (WAVE - 1);
   const int hi = lane >> 5;
   const int j32 = lane & 31;
   const int qtile = blockIdx.x * 4 + (int)(threadIdx.x / WAVE);
   const int q0 = qtile * 32;
-  if (q0 >= S) return;
+  const bool live_wave = q0 < S;
   const long bh = blockIdx.y;
   const long b = bh / H, h = bh % H;
   const bf16_t* qp = q + b * sio.bs + h * sio.hs;
@@ -446,10 +475,12 @@ __global__ __launch_bounds__(256, 3) void attn_bwd_dq_kernel(
   const bf16_t* qrp = qp + (long)min(qrow, S - 1) * sio.rs;
   const bf16_t* dorp = dop + (long)min(qrow, S - 1) * sdo.rs;
   bf16x8v qf[4], dof[4];
+  if (live_wave) {
 #pragma unroll
-  for (int s = 0; s < 4; ++s) {
-    qf[s] = *reinterpret_cast<const bf16x8v*>(qrp + s * 16 + hi * 8);
-    dof[s] = *reinterpret_cast<const bf16x8v*>(dorp + s * 16 + hi * 8);
+    for (int s = 0; s < 4; ++s) {
+      qf[s] = *reinterpret_cast<const bf16x8v*>(qrp + s * 16 + hi * 8);
+      dof[s] = *reinterpret_cast<const bf16x8v*>(dorp + s * 16 + hi * 8);
+    }
   }
   const float l_row = (qrow < S) ? lse[bh * (long)S + qrow] : -INFINITY;
   const float dlt_row = (qrow < S) ? delta[bh * (long)S + qrow] : 0.f;
@@ -460,8 +491,18 @@ __global__ __launch_bounds__(256, 3) void attn_bwd_dq_kernel(
 #pragma unroll
     for (int r = 0; r < 16; ++r) dq_acc[h][r] = 0.f;
 
-  const int kv_end = causal ? min(S, q0 + 32) : S;
+  // block-wide kv range (barriers are block-wide); waves guard compute
+  const int kv_end = causal ? min(S, blockIdx.x * 128 + 128) : S;
   for (int k0 = 0; k0 < kv_end; k0 += 32) {
+    const int srow = min(k0 + st_row, S - 1);
+    bf16x8v kstage = *reinterpret_cast<const bf16x8v*>(
+        kp + (long)srow * sio.rs + st_c16 / 2);
+    __syncthreads();
+    *reinterpret_cast<bf16x8v*>(
+        reinterpret_cast<char*>(k_ldsb) + st_row * 128 + st_c16) = kstage;
+    __syncthreads();
+    const bool compute = live_wave && (!causal || k0 <= q0 + 31);
+    if (!compute) continue;
     const int key_l = k0 + j32;
     const bf16_t* krp = kp + (long)min(key_l, S - 1) * sio.rs;
     const bf16_t* vrp = vp + (long)min(key_l, S - 1) * sio.rs;
@@ -500,19 +541,15 @@ __global__ __launch_bounds__(256, 3) void attn_bwd_dq_kernel(
       for (int step = 0; step < 2; ++step) {
         bf16x8v kcol;
 #pragma unroll
-        for (int e = 0; e < 8; ++e) {
-          const int kk = k0 + step * 16 + hi * 8 + e;
-          kcol[e] = (kk < S)
-              ? *reinterpret_cast<const __bf16*>(
-                    kp + (long)kk * sio.rs + hh * 32 + j32)
-              : (__bf16)0.f;
-        }
+        for (int e = 0; e < 8; ++e)
+          kcol[e] = k_ldsb[(step * 16 + hi * 8 + e) * 64 + hh * 32 + j32];
         dq_acc[hh] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
             *reinterpret_cast<const bf16x8v*>(&pa_ds[step][0]), kcol,
             dq_acc[hh], 0, 0, 0);
       }
     }
   }
+  if (!live_wave) return;
 
 #pragma unroll
   for (int hh = 0; hh < 2; ++hh) {

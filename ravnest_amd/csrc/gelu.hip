@@ -15,16 +15,23 @@ DEVINL float load_pt<float>(const float* p) { return *p; }
 template <>
 DEVINL float load_pt<bf16_t>(const bf16_t* p) { return bf2f(*p); }
 
+// tanh via the fast hardware exp (libm tanhf is a long VALU chain;
+// these kernels are otherwise HBM-bound)
+DEVINL float fast_tanh(float x) {
+  const float t = __expf(2.f * x);
+  return (t - 1.f) / (t + 1.f);
+}
+
 DEVINL float gelu_f(float x) {
   const float k = 0.7978845608028654f;  // sqrt(2/pi)
-  float t = tanhf(k * (x + 0.044715f * x * x * x));
+  float t = fast_tanh(k * (x + 0.044715f * x * x * x));
   return 0.5f * x * (1.f + t);
 }
 
 DEVINL float gelu_grad_f(float x) {
   const float k = 0.7978845608028654f;
   float x2 = x * x;
-  float th = tanhf(k * (x + 0.044715f * x * x2));
+  float th = fast_tanh(k * (x + 0.044715f * x * x2));
   float sech2 = 1.f - th * th;
   return 0.5f * (1.f + th) + 0.5f * x * sech2 * k * (1.f + 3.f * 0.044715f * x2);
 }
@@ -39,11 +46,11 @@ __global__ void bias_gelu_fwd_kernel(const T* __restrict__ x,
     const long nv = n >> 3;
     for (long i = i0; i < nv; i += stride) {
       bf16x8 v = *reinterpret_cast<const bf16x8*>(x + i * 8);
-      const int hb = (int)((i * 8) % H);
+      const int hb = (int)((i * 8) % H);  // H%8==0: the 8 never wrap
       bf16x8 o;
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
-        float b = load_pt(bias + (hb + j) % H);
+        float b = load_pt(bias + hb + j);
         o[j] = (short)f2us(gelu_f(us2f((unsigned short)v[j]) + b));
       }
       *reinterpret_cast<bf16x8*>(y + i * 8) = o;
@@ -71,11 +78,11 @@ __global__ void bias_gelu_bwd_kernel(const T* __restrict__ dy,
     for (long i = i0; i < nv; i += stride) {
       bf16x8 v = *reinterpret_cast<const bf16x8*>(x + i * 8);
       bf16x8 g = *reinterpret_cast<const bf16x8*>(dy + i * 8);
-      const int hb = (int)((i * 8) % H);
+      const int hb = (int)((i * 8) % H);  // H%8==0: the 8 never wrap
       bf16x8 o;
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
-        float b = load_pt(bias + (hb + j) % H);
+        float b = load_pt(bias + hb + j);
         float xf = us2f((unsigned short)v[j]) + b;
         o[j] = (short)f2us(us2f((unsigned short)g[j]) * gelu_grad_f(xf));
       }
@@ -87,6 +94,21 @@ __global__ void bias_gelu_bwd_kernel(const T* __restrict__ dy,
     for (long i = i0; i < n; i += stride)
       dx[i] = dy[i] * gelu_grad_f(x[i] + load_pt(bias + i % H));
   }
+}
+
+// column sum of a bf16 (N,H) tensor -> fp32 (H), chunked rows +
+// atomics (the LN dwdb recipe). Used for the fused bias grad.
+__global__ void colsum_kernel(const bf16_t* __restrict__ x,
+                              float* __restrict__ out, int H, long N,
+                              long rows_per_chunk) {
+  const int col = blockIdx.x * blockDim.x + threadIdx.x;
+  const int chunk = blockIdx.y;
+  if (col >= H) return;
+  const long r0 = (long)chunk * rows_per_chunk;
+  const long r1 = min(N, r0 + rows_per_chunk);
+  float s = 0.f;
+  for (long r = r0; r < r1; ++r) s += bf2f(x[r * H + col]);
+  atomicAdd(&out[col], s);
 }
 
 }  // namespace
@@ -151,4 +173,23 @@ at::Tensor bias_gelu_bwd(at::Tensor dy, at::Tensor x, at::Tensor bias) {
   }
   HIP_CHECK_LAST();
   return dx;
+}
+
+
+at::Tensor colsum_bf16(at::Tensor x) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous());
+  TORCH_CHECK(x.scalar_type() == at::kBFloat16);
+  const int H = x.size(-1);
+  const long N = x.numel() / H;
+  auto out = at::zeros({H}, x.options().dtype(at::kFloat));
+  auto stream = c10::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
+  const int col_blocks = (H + 255) / 256;
+  int nchunks = (int)std::min<long>((N + 63) / 64,
+                                    std::max(1L, (long)(768 / col_blocks)));
+  const long rows_per_chunk = (N + nchunks - 1) / nchunks;
+  hipLaunchKernelGGL(colsum_kernel, dim3(col_blocks, nchunks), dim3(256), 0,
+                     stream, reinterpret_cast<const bf16_t*>(x.data_ptr()),
+                     out.data_ptr<float>(), H, N, rows_per_chunk);
+  HIP_CHECK_LAST();
+  return out;
 }

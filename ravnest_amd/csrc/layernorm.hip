@@ -202,6 +202,37 @@ __global__ void ln_bwd_dx_kernel(const T* __restrict__ dy,
   const float mu = mean[row], rs = rstd[row];
 
   float c1 = 0.f, c2 = 0.f;
+  if constexpr (sizeof(T) == 2) {
+    if ((H & 7) == 0) {
+      const int G = H >> 3;
+      for (int g8 = lane; g8 < G; g8 += WAVE) {
+        bf16x8 vd = *reinterpret_cast<const bf16x8*>(dyr + g8 * 8);
+        bf16x8 vx = *reinterpret_cast<const bf16x8*>(xr + g8 * 8);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          float g = us2f((unsigned short)vd[j]) * load1(w + g8 * 8 + j);
+          float xhat = (us2f((unsigned short)vx[j]) - mu) * rs;
+          c1 += g * xhat;
+          c2 += g;
+        }
+      }
+      c1 = wave_sum(c1) / H;
+      c2 = wave_sum(c2) / H;
+      for (int g8 = lane; g8 < G; g8 += WAVE) {
+        bf16x8 vd = *reinterpret_cast<const bf16x8*>(dyr + g8 * 8);
+        bf16x8 vx = *reinterpret_cast<const bf16x8*>(xr + g8 * 8);
+        bf16x8 o;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          float g = us2f((unsigned short)vd[j]) * load1(w + g8 * 8 + j);
+          float xhat = (us2f((unsigned short)vx[j]) - mu) * rs;
+          o[j] = (short)f2us(rs * (g - xhat * c1 - c2));
+        }
+        *reinterpret_cast<bf16x8*>(dxr + g8 * 8) = o;
+      }
+      return;
+    }
+  }
   for (int i = lane; i < H; i += WAVE) {
     float g = load1(dyr + i) * load1(w + i);
     float xhat = (load1(xr + i) - mu) * rs;

@@ -34,8 +34,11 @@ class _BiasGeluFn(torch.autograd.Function):
         x, bias = ctx.saved_tensors
         ext = get_ext(required=True)
         dx = ext.bias_gelu_bwd(dy.contiguous(), x, bias)
-        # bias grad: reduce over all but last dim (fp32 accumulation)
-        db = dx.float().reshape(-1, dx.shape[-1]).sum(0).to(bias.dtype)
+        # bias grad: fused bf16 column sum (no fp32 materialization)
+        if dx.dtype == torch.bfloat16:
+            db = ext.colsum_bf16(dx.reshape(-1, dx.shape[-1])).to(bias.dtype)
+        else:
+            db = dx.float().reshape(-1, dx.shape[-1]).sum(0).to(bias.dtype)
         return dx, db
 
 

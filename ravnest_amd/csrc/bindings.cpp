@@ -32,6 +32,7 @@ void fused_lamb(std::vector<at::Tensor> ps, std::vector<at::Tensor> gs,
                 std::vector<at::Tensor> masters, double lr, double b1,
                 double b2, double eps, double wd, double bc1, double bc2,
                 double clamp_trust);
+void fused_copy(std::vector<at::Tensor> srcs, std::vector<at::Tensor> dsts);
 std::vector<at::Tensor> attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v,
                                  at::Tensor mask, bool causal, double scale);
 std::vector<at::Tensor> attn_bwd(at::Tensor q, at::Tensor k, at::Tensor v,
@@ -60,6 +61,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("fused_adam", &fused_adam, "multi-tensor Adam");
   m.def("fused_sgd", &fused_sgd, "multi-tensor SGD+momentum");
   m.def("fused_lamb", &fused_lamb, "multi-tensor LAMB");
+  m.def("fused_copy", &fused_copy, "multi-tensor device copy");
   m.def("attn_fwd", &attn_fwd, "flash attention fwd (bf16, MFMA)");
   m.def("attn_bwd", &attn_bwd, "flash attention bwd (bf16, MFMA, D=64)");
   m.def("attn_fwd_qkv", &attn_fwd_qkv,

@@ -228,6 +228,32 @@ at::Tensor build_chunks(const std::vector<at::Tensor>& ps,
   return host.to(ps[0].device());  // blocking copy: host buffer dies here
 }
 
+// multi-tensor device copy: ONE launch clones every parameter into its
+// snapshot slot (the versioning engine's per-step param snapshot was ~200
+// separate hipMemcpy calls — SURVEY.md section 2.3 "multi-tensor
+// clone/copy kernels ... versioned weight arena").
+__global__ void copy_mt_kernel(const Chunk* __restrict__ chunks) {
+  const Chunk c = chunks[blockIdx.x];
+  // p = src, m = dst (fp32 elems) OR 2-byte mode via tensor_idx flag
+  if (c.tensor_idx == 2) {  // 2-byte elements
+    const int n4 = c.n >> 3;  // 16B groups of 8 bf16
+    const ulong2* src = reinterpret_cast<const ulong2*>(c.p);
+    ulong2* dst = reinterpret_cast<ulong2*>(c.g);
+    for (int i = threadIdx.x; i < n4; i += blockDim.x) dst[i] = src[i];
+    for (int i = (n4 << 3) + threadIdx.x; i < c.n; i += blockDim.x)
+      reinterpret_cast<bf16_t*>(c.g)[i] =
+          reinterpret_cast<const bf16_t*>(c.p)[i];
+  } else {
+    const int n4 = c.n >> 2;
+    const float4* src = reinterpret_cast<const float4*>(c.p);
+    float4* dst = reinterpret_cast<float4*>(c.g);
+    for (int i = threadIdx.x; i < n4; i += blockDim.x) dst[i] = src[i];
+    for (int i = (n4 << 2) + threadIdx.x; i < c.n; i += blockDim.x)
+      reinterpret_cast<float*>(c.g)[i] =
+          reinterpret_cast<const float*>(c.p)[i];
+  }
+}
+
 }  // namespace
 
 void fused_adam(std::vector<at::Tensor> ps, std::vector<at::Tensor> gs,
@@ -302,5 +328,37 @@ void fused_lamb(std::vector<at::Tensor> ps, std::vector<at::Tensor> gs,
     LAMB_LAUNCH(bf16_t);
   else
     LAMB_LAUNCH(float);
+  HIP_CHECK_LAST();
+}
+
+
+void fused_copy(std::vector<at::Tensor> srcs, std::vector<at::Tensor> dsts) {
+  if (srcs.empty()) return;
+  std::vector<Chunk> chunks;
+  for (size_t t = 0; t < srcs.size(); ++t) {
+    TORCH_CHECK(srcs[t].is_contiguous() && dsts[t].is_contiguous());
+    TORCH_CHECK(srcs[t].scalar_type() == dsts[t].scalar_type());
+    const int es = srcs[t].element_size();
+    TORCH_CHECK(es == 2 || es == 4, "fused_copy: 2- or 4-byte elems");
+    long n = srcs[t].numel();
+    char* sp = (char*)srcs[t].data_ptr();
+    char* dp = (char*)dsts[t].data_ptr();
+    for (long off = 0; off < n; off += CHUNK) {
+      Chunk c{};
+      c.p = sp + off * es;
+      c.g = dp + off * es;
+      c.n = (int)std::min<long>(CHUNK, n - off);
+      c.tensor_idx = es;  // element size flag
+      chunks.push_back(c);
+    }
+  }
+  auto host = at::from_blob(chunks.data(),
+                            {(long)(chunks.size() * sizeof(Chunk))},
+                            at::TensorOptions().dtype(at::kByte));
+  auto dev_chunks = host.to(srcs[0].device());
+  auto stream = c10::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
+  hipLaunchKernelGGL(copy_mt_kernel, dim3((int)chunks.size()), dim3(256), 0,
+                     stream,
+                     reinterpret_cast<const Chunk*>(dev_chunks.data_ptr()));
   HIP_CHECK_LAST();
 }

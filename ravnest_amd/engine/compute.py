@@ -85,9 +85,22 @@ class ComputeEngine:
     def _snapshot_current(self):
         if not self.versioning:
             return
-        self.version_to_param[self.current_version] = [
-            p.detach().clone() for p in self._params]
+        snap = [torch.empty_like(p) for p in self._params]
+        self._fast_copy([p.detach() for p in self._params], snap)
+        self.version_to_param[self.current_version] = snap
         self.version_refs.setdefault(self.current_version, 0)
+
+    def _fast_copy(self, srcs, dsts):
+        """Multi-tensor device copy: one kernel launch for the whole
+        parameter set (vs ~#params hipMemcpy calls)."""
+        if srcs and srcs[0].is_cuda:
+            from ..ops import get_ext
+            ext = get_ext(required=False)
+            if ext is not None:
+                ext.fused_copy(srcs, dsts)
+                return
+        for s, d in zip(srcs, dsts):
+            d.copy_(s)
 
     def latest_state_snapshot(self) -> dict:
         """Latest parameter snapshot for peers to pull (parity:

@@ -90,6 +90,12 @@ class Node:
                  compression: bool = False,
                  wire_dtype: torch.dtype | None = None,
                  amp_dtype: torch.dtype | None = None):
+        # parity: the reference's fp16 wire compression (utils.py:184-194)
+        # becomes an optional on-the-wire cast; bf16 is the natural MI355X
+        # wire dtype (no clamping needed)
+        if compression and wire_dtype is None:
+            wire_dtype = torch.bfloat16
+        self.wire_dtype = wire_dtype
         self.base_dir = base_dir
         if config is None:
             config = load_node_json_configs(name, base_dir)
@@ -482,6 +488,8 @@ class Node:
             out = outputs[k]
             if torch.is_tensor(out):
                 out = out.detach()
+                if self.wire_dtype is not None and out.is_floating_point():
+                    out = out.to(self.wire_dtype)
             else:  # routed python scalar (a .size() value): ship as int64
                 out = torch.tensor([int(out)], dtype=torch.int64,
                                    device=self.device)

@@ -275,21 +275,29 @@ __global__ __launch_bounds__(256) void attn_fwd_lds_kernel(
   const int v_dst = st_row * 128 + st_c16;
 
   // every wave must loop over ALL tiles (barriers are block-wide); a
-  // wave past the causal horizon just skips its compute
+  // wave past the causal horizon just skips its compute.
+  // T14 split (guide G15): tile t+1's global loads ISSUE before tile t's
+  // MFMA/softmax phase so HBM/L2 latency hides under compute; the
+  // register stage is written to LDS after the consume barrier.
   const int kv_all = causal ? min(S, blockIdx.x * 128 + 128) : S;
+  bf16x8v kstage = *reinterpret_cast<const bf16x8v*>(
+      kp + (long)min(st_row, S - 1) * sk.rs + st_c16 / 2);
+  bf16x8v vstage = *reinterpret_cast<const bf16x8v*>(
+      vp + (long)min(st_row, S - 1) * sv.rs + st_c16 / 2);
   for (int k0 = 0; k0 < kv_all; k0 += 32) {
-    // ---- cooperative stage ----
-    const int grow = min(k0 + st_row, S - 1);
-    bf16x8v kstage = *reinterpret_cast<const bf16x8v*>(
-        kp + (long)grow * sk.rs + st_c16 / 2);
-    bf16x8v vstage = *reinterpret_cast<const bf16x8v*>(
-        vp + (long)grow * sv.rs + st_c16 / 2);
     __syncthreads();  // previous tile fully consumed
     *reinterpret_cast<bf16x8v*>(
         reinterpret_cast<char*>(k_lds) + k_dst) = kstage;
     *reinterpret_cast<bf16x8v*>(
         reinterpret_cast<char*>(v_lds) + v_dst) = vstage;
     __syncthreads();
+    if (k0 + 32 < kv_all) {  // issue next tile's loads under this compute
+      const int nrow = min(k0 + 32 + st_row, S - 1);
+      kstage = *reinterpret_cast<const bf16x8v*>(
+          kp + (long)nrow * sk.rs + st_c16 / 2);
+      vstage = *reinterpret_cast<const bf16x8v*>(
+          vp + (long)nrow * sv.rs + st_c16 / 2);
+    }
     const bool compute = live_wave && (!causal || k0 <= q0 + 31);
     if (compute) {
       // ---- QK^T: K A-frags from swizzled LDS (ds_read_b128) ----

@@ -225,11 +225,12 @@ __global__ __launch_bounds__(256, 4) void attn_bwd_dv_kernel(
     int causal, float scale, int has_mask, long H, StridesB sio,
     StridesB sdo, StridesB sg) {
   constexpr int D = 64;
-  // block-shared dO tile: the B-operand of the dV MFMA needs
-  // row-scattered 2B elements — staged once per block (4KB) instead of
-  // 32 strided global scalars per wave per tile (fwd-v2 recipe)
-  __shared__ __attribute__((aligned(16))) char smem[4096];
+  // block-shared tiles: dO (linear: 2B B-operand reads) and Q
+  // (((row&15)<<4)-swizzled: conflict-free b128 A-fragment reads).
+  // After staging, a tile iteration does NO per-wave global reads.
+  __shared__ __attribute__((aligned(16))) char smem[8192];
   bf16_t* do_lds = reinterpret_cast<bf16_t*>(smem);
+  bf16_t* q_lds = reinterpret_cast<bf16_t*>(smem + 4096);
   const int st_row = threadIdx.x >> 3;
   const int st_c16 = (threadIdx.x & 7) * 16;
   const int lane = threadIdx.x & (WAVE - 1);
@@ -263,24 +264,35 @@ __global__ __launch_bounds__(256, 4) void attn_bwd_dv_kernel(
     for (int r = 0; r < 16; ++r) dv_acc[hh][r] = 0.f;
 
   const int q_start = causal ? blockIdx.x * 128 : 0;
+  const int q_dst = (st_row * 128 + st_c16) ^ ((st_row & 15) << 4);
+  bf16x8v dstage = *reinterpret_cast<const bf16x8v*>(
+      dop + (long)min(q_start + st_row, S - 1) * sdo.rs + st_c16 / 2);
+  bf16x8v qstage = *reinterpret_cast<const bf16x8v*>(
+      qp + (long)min(q_start + st_row, S - 1) * sio.rs + st_c16 / 2);
   for (int q0 = q_start; q0 < S; q0 += 32) {
-    const int srow = min(q0 + st_row, S - 1);
-    bf16x8v dstage = *reinterpret_cast<const bf16x8v*>(
-        dop + (long)srow * sdo.rs + st_c16 / 2);
     __syncthreads();
     *reinterpret_cast<bf16x8v*>(
         reinterpret_cast<char*>(do_lds) + st_row * 128 + st_c16) = dstage;
+    *reinterpret_cast<bf16x8v*>(
+        reinterpret_cast<char*>(q_lds) + q_dst) = qstage;
     __syncthreads();
+    if (q0 + 32 < S) {  // T14: next tile's loads under this compute
+      dstage = *reinterpret_cast<const bf16x8v*>(
+          dop + (long)min(q0 + 32 + st_row, S - 1) * sdo.rs + st_c16 / 2);
+      qstage = *reinterpret_cast<const bf16x8v*>(
+          qp + (long)min(q0 + 32 + st_row, S - 1) * sio.rs + st_c16 / 2);
+    }
     const bool compute = live_wave && (!causal || q0 + 31 >= k0);
     if (!compute) continue;
-    const int qrow_l = q0 + j32;
-    const bf16_t* qrp = qp + (long)min(qrow_l, S - 1) * sio.rs;
     f32x16 s_acc;
 #pragma unroll
     for (int r = 0; r < 16; ++r) s_acc[r] = 0.f;
 #pragma unroll
     for (int s = 0; s < 4; ++s) {
-      bf16x8v qf = *reinterpret_cast<const bf16x8v*>(qrp + s * 16 + hi * 8);
+      const int qoff = (j32 * 128 + (s * 16 + hi * 8) * 2) ^
+                       ((j32 & 15) << 4);
+      bf16x8v qf = *reinterpret_cast<const bf16x8v*>(
+          reinterpret_cast<char*>(q_lds) + qoff);
       s_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qf, kf[s], s_acc,
                                                       0, 0, 0);
     }
@@ -333,8 +345,11 @@ __global__ __launch_bounds__(256) void attn_bwd_dk_kernel(
     int causal, float scale, int has_mask, long H, StridesB sio,
     StridesB sdo, StridesB sg) {
   constexpr int D = 64;
-  __shared__ __attribute__((aligned(16))) char smem[4096];
-  bf16_t* q_lds = reinterpret_cast<bf16_t*>(smem);  // B-operand Q tile
+  // Q tile swizzled (b128 A-frags for S, XOR-adjusted u16 B-reads for
+  // dK) + dO tile swizzled (b128 A-frags for dP)
+  __shared__ __attribute__((aligned(16))) char smem[8192];
+  bf16_t* q_lds = reinterpret_cast<bf16_t*>(smem);
+  bf16_t* do_lds = reinterpret_cast<bf16_t*>(smem + 4096);
   const int st_row = threadIdx.x >> 3;
   const int st_c16 = (threadIdx.x & 7) * 16;
   const int lane = threadIdx.x & (WAVE - 1);
@@ -373,19 +388,26 @@ __global__ __launch_bounds__(256) void attn_bwd_dk_kernel(
     for (int r = 0; r < 16; ++r) dk_acc[hh][r] = 0.f;
 
   const int q_start = causal ? blockIdx.x * 128 : 0;
+  const int swz_dst = (st_row * 128 + st_c16) ^ ((st_row & 15) << 4);
+  bf16x8v qstage = *reinterpret_cast<const bf16x8v*>(
+      qp + (long)min(q_start + st_row, S - 1) * sio.rs + st_c16 / 2);
+  bf16x8v dstage = *reinterpret_cast<const bf16x8v*>(
+      dop + (long)min(q_start + st_row, S - 1) * sdo.rs + st_c16 / 2);
   for (int q0 = q_start; q0 < S; q0 += 32) {
-    const int srow = min(q0 + st_row, S - 1);
-    bf16x8v qstage = *reinterpret_cast<const bf16x8v*>(
-        qp + (long)srow * sio.rs + st_c16 / 2);
     __syncthreads();
     *reinterpret_cast<bf16x8v*>(
-        reinterpret_cast<char*>(q_lds) + st_row * 128 + st_c16) = qstage;
+        reinterpret_cast<char*>(q_lds) + swz_dst) = qstage;
+    *reinterpret_cast<bf16x8v*>(
+        reinterpret_cast<char*>(do_lds) + swz_dst) = dstage;
     __syncthreads();
+    if (q0 + 32 < S) {  // T14: next tile's loads under this compute
+      qstage = *reinterpret_cast<const bf16x8v*>(
+          qp + (long)min(q0 + 32 + st_row, S - 1) * sio.rs + st_c16 / 2);
+      dstage = *reinterpret_cast<const bf16x8v*>(
+          dop + (long)min(q0 + 32 + st_row, S - 1) * sdo.rs + st_c16 / 2);
+    }
     const bool compute = live_wave && (!causal || q0 + 31 >= k0);
     if (!compute) continue;
-    const int qrow_l = q0 + j32;
-    const bf16_t* qrp = qp + (long)min(qrow_l, S - 1) * sio.rs;
-    const bf16_t* dorp = dop + (long)min(qrow_l, S - 1) * sdo.rs;
     f32x16 s_acc, dp_acc;
 #pragma unroll
     for (int r = 0; r < 16; ++r) {
@@ -394,8 +416,12 @@ __global__ __launch_bounds__(256) void attn_bwd_dk_kernel(
     }
 #pragma unroll
     for (int s = 0; s < 4; ++s) {
-      bf16x8v qf = *reinterpret_cast<const bf16x8v*>(qrp + s * 16 + hi * 8);
-      bf16x8v dof = *reinterpret_cast<const bf16x8v*>(dorp + s * 16 + hi * 8);
+      const int foff = (j32 * 128 + (s * 16 + hi * 8) * 2) ^
+                       ((j32 & 15) << 4);
+      bf16x8v qf = *reinterpret_cast<const bf16x8v*>(
+          reinterpret_cast<char*>(q_lds) + foff);
+      bf16x8v dof = *reinterpret_cast<const bf16x8v*>(
+          reinterpret_cast<char*>(do_lds) + foff);
       s_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qf, kf[s], s_acc,
                                                       0, 0, 0);
       dp_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dof, vf[s], dp_acc,
@@ -420,8 +446,13 @@ __global__ __launch_bounds__(256) void attn_bwd_dk_kernel(
       for (int step = 0; step < 2; ++step) {
         bf16x8v qf2;
 #pragma unroll
-        for (int e = 0; e < 8; ++e)
-          qf2[e] = q_lds[(step * 16 + hi * 8 + e) * 64 + hh * 32 + j32];
+        for (int e = 0; e < 8; ++e) {
+          const int row = step * 16 + hi * 8 + e;
+          const int boff = (row * 128 + (hh * 32 + j32) * 2) ^
+                           ((row & 15) << 4);
+          qf2[e] = *reinterpret_cast<const bf16_t*>(
+              reinterpret_cast<const char*>(q_lds) + boff);
+        }
         dk_acc[hh] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
             *reinterpret_cast<const bf16x8v*>(&pa_ds[step][0]), qf2,
             dk_acc[hh], 0, 0, 0);
@@ -452,8 +483,11 @@ __global__ __launch_bounds__(256, 3) void attn_bwd_dq_kernel(
     int causal, float scale, int has_mask, long H, StridesB sio,
     StridesB sdo, StridesB sg) {
   constexpr int D = 64;
-  __shared__ __attribute__((aligned(16))) char smem[4096];
-  bf16_t* k_ldsb = reinterpret_cast<bf16_t*>(smem);  // B-operand K tile
+  // K tile swizzled (b128 A-frags for S^T, XOR-adjusted u16 B-reads for
+  // dQ) + V tile swizzled (b128 A-frags for dP^T)
+  __shared__ __attribute__((aligned(16))) char smem[8192];
+  bf16_t* k_ldsb = reinterpret_cast<bf16_t*>(smem);
+  bf16_t* v_ldsb = reinterpret_cast<bf16_t*>(smem + 4096);
   const int st_row = threadIdx.x >> 3;
   const int st_c16 = (threadIdx.x & 7) * 16;
   const int lane = threadIdx.x & (WAVE - 1);
@@ -493,19 +527,26 @@ __global__ __launch_bounds__(256, 3) void attn_bwd_dq_kernel(
 
   // block-wide kv range (barriers are block-wide); waves guard compute
   const int kv_end = causal ? min(S, blockIdx.x * 128 + 128) : S;
+  const int swz_dst = (st_row * 128 + st_c16) ^ ((st_row & 15) << 4);
+  bf16x8v kstage = *reinterpret_cast<const bf16x8v*>(
+      kp + (long)min(st_row, S - 1) * sio.rs + st_c16 / 2);
+  bf16x8v vstage = *reinterpret_cast<const bf16x8v*>(
+      vp + (long)min(st_row, S - 1) * sio.rs + st_c16 / 2);
   for (int k0 = 0; k0 < kv_end; k0 += 32) {
-    const int srow = min(k0 + st_row, S - 1);
-    bf16x8v kstage = *reinterpret_cast<const bf16x8v*>(
-        kp + (long)srow * sio.rs + st_c16 / 2);
     __syncthreads();
     *reinterpret_cast<bf16x8v*>(
-        reinterpret_cast<char*>(k_ldsb) + st_row * 128 + st_c16) = kstage;
+        reinterpret_cast<char*>(k_ldsb) + swz_dst) = kstage;
+    *reinterpret_cast<bf16x8v*>(
+        reinterpret_cast<char*>(v_ldsb) + swz_dst) = vstage;
     __syncthreads();
+    if (k0 + 32 < kv_end) {  // T14: next tile's loads under this compute
+      kstage = *reinterpret_cast<const bf16x8v*>(
+          kp + (long)min(k0 + 32 + st_row, S - 1) * sio.rs + st_c16 / 2);
+      vstage = *reinterpret_cast<const bf16x8v*>(
+          vp + (long)min(k0 + 32 + st_row, S - 1) * sio.rs + st_c16 / 2);
+    }
     const bool compute = live_wave && (!causal || k0 <= q0 + 31);
     if (!compute) continue;
-    const int key_l = k0 + j32;
-    const bf16_t* krp = kp + (long)min(key_l, S - 1) * sio.rs;
-    const bf16_t* vrp = vp + (long)min(key_l, S - 1) * sio.rs;
     f32x16 s_acc, dp_acc;
 #pragma unroll
     for (int r = 0; r < 16; ++r) {
@@ -514,8 +555,12 @@ __global__ __launch_bounds__(256, 3) void attn_bwd_dq_kernel(
     }
 #pragma unroll
     for (int s = 0; s < 4; ++s) {
-      bf16x8v kfr = *reinterpret_cast<const bf16x8v*>(krp + s * 16 + hi * 8);
-      bf16x8v vfr = *reinterpret_cast<const bf16x8v*>(vrp + s * 16 + hi * 8);
+      const int foff = (j32 * 128 + (s * 16 + hi * 8) * 2) ^
+                       ((j32 & 15) << 4);
+      bf16x8v kfr = *reinterpret_cast<const bf16x8v*>(
+          reinterpret_cast<char*>(k_ldsb) + foff);
+      bf16x8v vfr = *reinterpret_cast<const bf16x8v*>(
+          reinterpret_cast<char*>(v_ldsb) + foff);
       // S^T[key][qrow], dP^T[key][qrow]
       s_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kfr, qf[s], s_acc,
                                                       0, 0, 0);
@@ -541,8 +586,13 @@ __global__ __launch_bounds__(256, 3) void attn_bwd_dq_kernel(
       for (int step = 0; step < 2; ++step) {
         bf16x8v kcol;
 #pragma unroll
-        for (int e = 0; e < 8; ++e)
-          kcol[e] = k_ldsb[(step * 16 + hi * 8 + e) * 64 + hh * 32 + j32];
+        for (int e = 0; e < 8; ++e) {
+          const int row = step * 16 + hi * 8 + e;
+          const int boff = (row * 128 + (hh * 32 + j32) * 2) ^
+                           ((row & 15) << 4);
+          kcol[e] = *reinterpret_cast<const bf16_t*>(
+              reinterpret_cast<const char*>(k_ldsb) + boff);
+        }
         dq_acc[hh] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
             *reinterpret_cast<const bf16x8v*>(&pa_ds[step][0]), kcol,
             dq_acc[hh], 0, 0, 0);

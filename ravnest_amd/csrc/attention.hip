@@ -227,7 +227,7 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
 // G4/T2: the b128 16-lane groups read rows distinct mod 16), V stored
 // linear and consumed as 2B LDS reads (16 consecutive banks per group).
 // ---------------------------------------------------------------------
-__global__ __launch_bounds__(256) void attn_fwd_lds_kernel(
+__global__ __launch_bounds__(256, 4) void attn_fwd_lds_kernel(
     const bf16_t* __restrict__ q, const bf16_t* __restrict__ k,
     const bf16_t* __restrict__ v, const float* __restrict__ mask,
     bf16_t* __restrict__ o, float* __restrict__ lse, int S, int causal,

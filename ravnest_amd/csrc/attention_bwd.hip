@@ -76,144 +76,6 @@ __global__ void attn_bwd_delta_kernel(const bf16_t* __restrict__ dout,
 }
 
 // -------------------------------------------------------------------
-// dkv kernel: wave owns keys [k0, k0+32); loops q tiles.
-// acc layouts: S/dP = [qrow-pattern][key=lane&31]
-//              dV/dK = [key-pattern][d = half*32 + lane&31]
-// -------------------------------------------------------------------
-__global__ __launch_bounds__(256) void attn_bwd_dkv_kernel(
-    const bf16_t* __restrict__ q, const bf16_t* __restrict__ k,
-    const bf16_t* __restrict__ v, const bf16_t* __restrict__ dout,
-    const float* __restrict__ lse, const float* __restrict__ delta,
-    const float* __restrict__ mask, bf16_t* __restrict__ dk,
-    bf16_t* __restrict__ dv, int S, int causal, float scale, int has_mask,
-    long H, StridesB sio, StridesB sdo, StridesB sg) {
-  // sio: strides of q/k/v; sdo: of o/dout; sg: of dk/dv outputs
-  constexpr int D = 64;
-  const int lane = threadIdx.x & (WAVE - 1);
-  const int hi = lane >> 5;
-  const int j32 = lane & 31;
-  const int ktile = blockIdx.x * 4 + (int)(threadIdx.x / WAVE);
-  const int k0 = ktile * 32;
-  if (k0 >= S) return;
-  const long bh = blockIdx.y;
-  const long b = bh / H, h = bh % H;
-  const bf16_t* qp = q + b * sio.bs + h * sio.hs;
-  const bf16_t* kp = k + b * sio.bs + h * sio.hs;
-  const bf16_t* vp = v + b * sio.bs + h * sio.hs;
-  const bf16_t* dop = dout + b * sdo.bs + h * sdo.hs;
-  bf16_t* dkp = dk + b * sg.bs + h * sg.hs;
-  bf16_t* dvp = dv + b * sg.bs + h * sg.hs;
-  const float* lsep = lse + bh * (long)S;
-  const float* dltp = delta + bh * (long)S;
-  const float* mp = has_mask ? (mask + b * (long)S) : nullptr;
-
-  // B-operand fragments held for the whole loop:
-  //   K^T: B[k=d][j=key]  -> K[key=j32][d=s*16+hi*8+e]   (16B loads)
-  //   V^T: B[k=d][j=key]  -> V[key=j32][d=...]
-  const int key = k0 + j32;
-  const bf16_t* krp = kp + (long)min(key, S - 1) * sio.rs;
-  const bf16_t* vrp = vp + (long)min(key, S - 1) * sio.rs;
-  bf16x8v kf[4], vf[4];
-#pragma unroll
-  for (int s = 0; s < 4; ++s) {
-    kf[s] = *reinterpret_cast<const bf16x8v*>(krp + s * 16 + hi * 8);
-    vf[s] = *reinterpret_cast<const bf16x8v*>(vrp + s * 16 + hi * 8);
-  }
-  const float mask_val = (has_mask && key < S) ? mp[key] : 0.f;
-
-  f32x16 dv_acc[2], dk_acc[2];
-#pragma unroll
-  for (int h = 0; h < 2; ++h)
-#pragma unroll
-    for (int r = 0; r < 16; ++r) {
-      dv_acc[h][r] = 0.f;
-      dk_acc[h][r] = 0.f;
-    }
-
-  const int q_start = causal ? (k0 / 32) * 32 : 0;
-  for (int q0 = q_start; q0 < S; q0 += 32) {
-    // A-operands for S and dP: Q / dO rows (16B loads per step)
-    const int qrow_l = q0 + j32;
-    const bf16_t* qrp = qp + (long)min(qrow_l, S - 1) * sio.rs;
-    const bf16_t* dorp = dop + (long)min(qrow_l, S - 1) * sdo.rs;
-    f32x16 s_acc, dp_acc;
-#pragma unroll
-    for (int r = 0; r < 16; ++r) {
-      s_acc[r] = 0.f;
-      dp_acc[r] = 0.f;
-    }
-#pragma unroll
-    for (int s = 0; s < 4; ++s) {
-      bf16x8v qf = *reinterpret_cast<const bf16x8v*>(qrp + s * 16 + hi * 8);
-      bf16x8v dof = *reinterpret_cast<const bf16x8v*>(dorp + s * 16 + hi * 8);
-      s_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qf, kf[s], s_acc,
-                                                      0, 0, 0);
-      dp_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dof, vf[s], dp_acc,
-                                                       0, 0, 0);
-    }
-    // P and dS in the [qrow-pattern][key=lane] layout — computed IN
-    // PLACE into s_acc / dp_acc (the extra register sets cost an
-    // occupancy tier: 180 VGPRs = 2 waves/SIMD)
-#pragma unroll
-    for (int r = 0; r < 16; ++r) {
-      const int qrow = q0 + (r & 3) + 8 * (r >> 2) + 4 * hi;
-      const int qclmp = min(qrow, S - 1);
-      const float l = lsep[qclmp];
-      float sv = s_acc[r] * scale + mask_val;
-      bool dead = (qrow >= S) || (key >= S) ||
-                  (causal && key > qrow) || !isfinite(l);
-      const float p = dead ? 0.f : __expf(sv - l);
-      s_acc[r] = p;                     // s_acc now holds P
-      dp_acc[r] = dead ? 0.f
-                       : p * (dp_acc[r] - dltp[qclmp]) * scale;  // dS
-    }
-    // dV += P^T dO ; dK += dS^T Q : A = transform(acc) over k=qrows,
-    // B[k=qrow][j=d] = dO/Q rows, strided scalar loads
-    unsigned int pa_p[2][4], pa_ds[2][4];
-    acc_to_afrag(s_acc, pa_p);
-    acc_to_afrag(dp_acc, pa_ds);
-#pragma unroll
-    for (int hh = 0; hh < 2; ++hh) {
-#pragma unroll
-      for (int step = 0; step < 2; ++step) {
-        bf16x8v dof, qf;
-#pragma unroll
-        for (int e = 0; e < 8; ++e) {
-          const int qrow = q0 + step * 16 + hi * 8 + e;
-          const long rq = (long)min(qrow, S - 1) * sio.rs + hh * 32 + j32;
-          const long rd = (long)min(qrow, S - 1) * sdo.rs + hh * 32 + j32;
-          const bool live = qrow < S;
-          dof[e] = live ? *reinterpret_cast<const __bf16*>(dop + rd)
-                        : (__bf16)0.f;
-          qf[e] = live ? *reinterpret_cast<const __bf16*>(qp + rq)
-                       : (__bf16)0.f;
-        }
-        dv_acc[hh] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-            *reinterpret_cast<const bf16x8v*>(&pa_p[step][0]), dof,
-            dv_acc[hh], 0, 0, 0);
-        dk_acc[hh] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-            *reinterpret_cast<const bf16x8v*>(&pa_ds[step][0]), qf,
-            dk_acc[hh], 0, 0, 0);
-      }
-    }
-  }
-
-  // store dV, dK: [key-pattern][d=hh*32+j32]
-#pragma unroll
-  for (int hh = 0; hh < 2; ++hh) {
-#pragma unroll
-    for (int r = 0; r < 16; ++r) {
-      const int krow = k0 + (r & 3) + 8 * (r >> 2) + 4 * hi;
-      if (krow < S) {
-        const long off = (long)krow * sg.rs + hh * 32 + j32;
-        dvp[off] = f2bf(dv_acc[hh][r]);
-        dkp[off] = f2bf(dk_acc[hh][r]);
-      }
-    }
-  }
-}
-
-// -------------------------------------------------------------------
 // dv kernel: wave owns keys; S -> P -> dV only (~110 VGPR, 4 waves/SIMD;
 // the combined dkv kernel sat at 180 VGPR = 2 waves/SIMD and 56% of wave
 // cycles parked on memory waits — PMC profile r01)
@@ -337,7 +199,7 @@ __global__ __launch_bounds__(256, 4) void attn_bwd_dv_kernel(
 // -------------------------------------------------------------------
 // dk kernel: wave owns keys; S, dP -> dS -> dK (~150 VGPR, 3 waves/SIMD)
 // -------------------------------------------------------------------
-__global__ __launch_bounds__(256) void attn_bwd_dk_kernel(
+__global__ __launch_bounds__(256, 3) void attn_bwd_dk_kernel(
     const bf16_t* __restrict__ q, const bf16_t* __restrict__ k,
     const bf16_t* __restrict__ v, const bf16_t* __restrict__ dout,
     const float* __restrict__ lse, const float* __restrict__ delta,

@@ -12,6 +12,8 @@
 
 namespace {
 
+// vectorized: 8 elements/thread/iter (two philox quads, 16B x/y for
+// bf16, 8B mask store); scalar tail for the last partial octet
 template <typename T>
 __global__ void dropout_fwd_kernel(const T* __restrict__ x,
                                    T* __restrict__ y,
@@ -22,29 +24,56 @@ __global__ void dropout_fwd_kernel(const T* __restrict__ x,
   const long i0 = (long)(blockIdx.x * blockDim.x + threadIdx.x);
   const long stride = (long)gridDim.x * blockDim.x;
   const float thresh = p;
-  const long nq = (n + 3) / 4;  // philox quads
-  for (long q = i0; q < nq; q += stride) {
-    unsigned int r[4];
-    ph.gen((unsigned long long)q, 0ull, r);
-    const long base = q * 4;
+  const long no = n >> 3;  // full octets
+  for (long o = i0; o < no; o += stride) {
+    unsigned int r[8];
+    ph.gen((unsigned long long)(o * 2), 0ull, r);
+    ph.gen((unsigned long long)(o * 2 + 1), 0ull, r + 4);
+    unsigned long long mbits = 0;
+    float vals[8];
+    if constexpr (sizeof(T) == 2) {
+      bf16x8 v = *reinterpret_cast<const bf16x8*>(x + o * 8);
 #pragma unroll
-    for (int j = 0; j < 4; ++j) {
-      const long i = base + j;
-      if (i >= n) break;
-      const float u = (r[j] >> 8) * (1.0f / 16777216.0f);  // [0,1)
-      const bool keep = u >= thresh;
-      mask[i] = keep ? 1 : 0;
-      float v = 0.f;
-      if constexpr (sizeof(T) == 2)
-        v = bf2f(x[i]);
-      else
-        v = x[i];
-      v = keep ? v * scale : 0.f;
-      if constexpr (sizeof(T) == 2)
-        y[i] = f2bf(v);
-      else
-        y[i] = v;
+      for (int j = 0; j < 8; ++j) vals[j] = us2f((unsigned short)v[j]);
+    } else {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) vals[j] = ((const float*)x)[o * 8 + j];
     }
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const float u = (r[j] >> 8) * (1.0f / 16777216.0f);
+      const bool keep = u >= thresh;
+      mbits |= ((unsigned long long)(keep ? 1 : 0)) << (8 * j);
+      vals[j] = keep ? vals[j] * scale : 0.f;
+    }
+    *reinterpret_cast<unsigned long long*>(mask + o * 8) = mbits;
+    if constexpr (sizeof(T) == 2) {
+      bf16x8 ov;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) ov[j] = (short)f2us(vals[j]);
+      *reinterpret_cast<bf16x8*>(y + o * 8) = ov;
+    } else {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) ((float*)y)[o * 8 + j] = vals[j];
+    }
+  }
+  // tail (same philox indexing as the vector body)
+  for (long i = (no << 3) + i0; i < n; i += stride) {
+    unsigned int r[4];
+    ph.gen((unsigned long long)(i >> 2), 0ull, r);
+    const float u = (r[i & 3] >> 8) * (1.0f / 16777216.0f);
+    const bool keep = u >= thresh;
+    mask[i] = keep ? 1 : 0;
+    float v;
+    if constexpr (sizeof(T) == 2)
+      v = bf2f(x[i]);
+    else
+      v = x[i];
+    v = keep ? v * scale : 0.f;
+    if constexpr (sizeof(T) == 2)
+      y[i] = f2bf(v);
+    else
+      y[i] = v;
   }
 }
 
@@ -54,7 +83,29 @@ __global__ void dropout_bwd_kernel(const T* __restrict__ dy,
                                    T* __restrict__ dx, long n, float scale) {
   const long i0 = (long)(blockIdx.x * blockDim.x + threadIdx.x);
   const long stride = (long)gridDim.x * blockDim.x;
-  for (long i = i0; i < n; i += stride) {
+  const long no = n >> 3;
+  for (long o = i0; o < no; o += stride) {
+    const unsigned long long mbits =
+        *reinterpret_cast<const unsigned long long*>(mask + o * 8);
+    if constexpr (sizeof(T) == 2) {
+      bf16x8 v = *reinterpret_cast<const bf16x8*>(dy + o * 8);
+      bf16x8 ov;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float f = us2f((unsigned short)v[j]);
+        ov[j] = (short)f2us(((mbits >> (8 * j)) & 1) ? f * scale : 0.f);
+      }
+      *reinterpret_cast<bf16x8*>(dx + o * 8) = ov;
+    } else {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float f = ((const float*)dy)[o * 8 + j];
+        ((float*)dx)[o * 8 + j] =
+            ((mbits >> (8 * j)) & 1) ? f * scale : 0.f;
+      }
+    }
+  }
+  for (long i = (no << 3) + i0; i < n; i += stride) {
     float v;
     if constexpr (sizeof(T) == 2)
       v = bf2f(dy[i]);

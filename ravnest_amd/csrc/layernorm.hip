@@ -266,7 +266,21 @@ __global__ void ln_bwd_dwdb_partial(const T* __restrict__ dy,
   const long r0 = (long)chunk * rows_per_chunk;
   const long r1 = min(N, r0 + rows_per_chunk);
   float sw = 0.f, sb = 0.f;
-  for (long r = r0; r < r1; ++r) {
+  long r = r0;
+  for (; r + 4 <= r1; r += 4) {  // keep several loads in flight
+    float g[4], xh[4];
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      g[j] = load1(dy + (r + j) * H + col);
+      xh[j] = (load1(x + (r + j) * H + col) - mean[r + j]) * rstd[r + j];
+    }
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      sw += g[j] * xh[j];
+      sb += g[j];
+    }
+  }
+  for (; r < r1; ++r) {
     float g = load1(dy + r * H + col);
     float xhat = (load1(x + r * H + col) - mean[r]) * rstd[r];
     sw += g * xhat;

@@ -49,6 +49,11 @@ def build_model(name: str, seq: int):
         return BertForMLM(BertConfig.tiny(max_seq=seq)), "bert"
     if name == "gpt2-small":
         return GPT(GPTConfig.gpt2_small(block_size=seq)), "gpt"
+    if name == "resnet50":
+        from ravnest_amd.models import resnet50
+        m = resnet50(num_classes=200)
+        m.cfg = type("C", (), {"vocab_size": 200})()  # criterion range
+        return m, "resnet"
     raise ValueError(name)
 
 
@@ -66,7 +71,7 @@ def main():
     ap.add_argument("--steps", type=int, default=20)
     ap.add_argument("--warmup", type=int, default=5)
     ap.add_argument("--model", default="bert-base")
-    ap.add_argument("--micro-batch", type=int, default=64)
+    ap.add_argument("--micro-batch", type=int, default=128)
     ap.add_argument("--seq", type=int, default=512)
     ap.add_argument("--reduce-factor", type=int, default=4,
                     help="average DP replicas every N steps")
@@ -119,9 +124,16 @@ def main():
             y[drop] = -100
             batches.append({"input_ids": ids, "attention_mask": mask})
             labels.append(y)
-        else:
+        elif family == "gpt":
             batches.append({"idx": ids})
             y = torch.roll(ids, -1, dims=1)
+            labels.append(y)
+        else:  # resnet: TinyImageNet-shaped synthetic images
+            X = torch.randn(args.micro_batch, 3, 64, 64, device=device)
+            if args.dtype == "bf16" and on_gpu:
+                X = X.to(torch.bfloat16)
+            y = torch.randint(0, 200, (args.micro_batch,), device=device)
+            batches.append({"x": X})
             labels.append(y)
 
     comm = None
@@ -130,7 +142,7 @@ def main():
                            dp_groups=[list(range(world))], device=device)
 
     mi_names = (["input_ids", "attention_mask"] if family == "bert"
-                else ["idx"])
+                else ["x"] if family == "resnet" else ["idx"])
     cfg = {
         "rank": rank, "world_size": world, "cluster_id": rank, "stage": 0,
         "n_stages": 1, "cluster_length": 1, "stage_ranks": [rank],

@@ -75,7 +75,7 @@ def main():
     ap.add_argument("--seq", type=int, default=512)
     ap.add_argument("--reduce-factor", type=int, default=4,
                     help="average DP replicas every N steps")
-    ap.add_argument("--dtype", default="bf16", choices=["bf16", "fp32"])
+    ap.add_argument("--dtype", default=None, choices=["bf16", "fp32"])
     ap.add_argument("--cpu", action="store_true",
                     help="CPU sanity mode (tiny model)")
     args = ap.parse_args()
@@ -87,6 +87,11 @@ def main():
         os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
         os.environ.setdefault("MASTER_PORT", "29531")
 
+    if args.dtype is None:
+        # transformers run bf16-native; ResNet stays fp32 NCHW (measured:
+        # MIOpen bf16 convs are ~4x slower than fp32 on this stack, and
+        # channels_last is ~15x slower — see BASELINE.md)
+        args.dtype = "fp32" if args.model == "resnet50" else "bf16"
     on_gpu = torch.cuda.is_available() and not args.cpu
     if not on_gpu and args.model == "bert-base" and args.cpu:
         args.model = "bert-tiny"
@@ -105,6 +110,9 @@ def main():
     # autocast dtype ping-pong); the fused optimizers keep fp32 masters.
     if args.dtype == "bf16" and on_gpu:
         model = model.to(torch.bfloat16)
+    if family == "resnet" and on_gpu and \
+            os.environ.get("RAVNEST_CHANNELS_LAST", "0") == "1":
+        model = model.to(memory_format=torch.channels_last)
     amp = None
 
     # ---- synthetic data of the workload's shape (no network: random ids,
@@ -132,6 +140,8 @@ def main():
             X = torch.randn(args.micro_batch, 3, 64, 64, device=device)
             if args.dtype == "bf16" and on_gpu:
                 X = X.to(torch.bfloat16)
+            if on_gpu and os.environ.get("RAVNEST_CHANNELS_LAST", "0") == "1":
+                X = X.to(memory_format=torch.channels_last)
             y = torch.randint(0, 200, (args.micro_batch,), device=device)
             batches.append({"x": X})
             labels.append(y)

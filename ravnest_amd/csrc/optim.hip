@@ -193,7 +193,16 @@ at::Tensor build_chunks(const std::vector<at::Tensor>& ps,
   std::vector<Chunk> chunks;
   esize = ps[0].scalar_type() == at::kBFloat16 ? 2 : 4;
   for (size_t t = 0; t < ps.size(); ++t) {
-    TORCH_CHECK(ps[t].is_contiguous() && gs[t].is_contiguous());
+    // any DENSE layout is fine (channels_last conv weights included):
+    // the update is elementwise over raw storage, so p/g/state only need
+    // to share one layout
+    const bool dense =
+        ps[t].is_contiguous() ||
+        ps[t].is_contiguous(at::MemoryFormat::ChannelsLast) ||
+        ps[t].is_contiguous(at::MemoryFormat::ChannelsLast3d);
+    TORCH_CHECK(dense, "fused optimizer: param must be dense");
+    TORCH_CHECK(gs[t].strides() == ps[t].strides(),
+                "grad layout must match param layout");
     TORCH_CHECK(gs[t].scalar_type() == ps[t].scalar_type(),
                 "grad dtype must match param dtype");
     long n = ps[t].numel();
@@ -336,7 +345,8 @@ void fused_copy(std::vector<at::Tensor> srcs, std::vector<at::Tensor> dsts) {
   if (srcs.empty()) return;
   std::vector<Chunk> chunks;
   for (size_t t = 0; t < srcs.size(); ++t) {
-    TORCH_CHECK(srcs[t].is_contiguous() && dsts[t].is_contiguous());
+    TORCH_CHECK(srcs[t].strides() == dsts[t].strides(),
+                "fused_copy: layouts must match");
     TORCH_CHECK(srcs[t].scalar_type() == dsts[t].scalar_type());
     const int es = srcs[t].element_size();
     TORCH_CHECK(es == 2 || es == 4, "fused_copy: 2- or 4-byte elems");

@@ -569,17 +569,24 @@ class Node:
         if self.stage + 1 < self.n_stages:
             nxt = self.stage_ranks[self.stage + 1]
             try:
-                ch = self.comm.channel(self.rank, nxt, "fwd")
-                ch.send_sync(Message(action=ActionTypes.STOP, fpid=0,
-                                     tensors=[]))
+                # queued (not send_sync): must not overtake in-flight
+                # forward/eval messages on the same FIFO channel
+                self.comm.send(nxt, "fwd", Message(
+                    action=ActionTypes.STOP, fpid=0, tensors=[]))
             except KeyError:
                 pass
 
-    def stop_cluster(self):
-        """Root: drain, cascade STOP down the pipeline, stop this node."""
+    def stop_cluster(self, timeout: float = 120.0):
+        """Root: drain, cascade STOP down the pipeline, stop this node.
+        The cascade goes THROUGH the dispatch queue at the lowest priority
+        so queued work (e.g. pending eval forwards) runs first — a direct
+        send here would overtake it on the FIFO channels."""
         if self.node_type == NodeTypes.ROOT and not self.fused:
             self.wait_for_backwards()
-        self._forward_stop()
+        self._enqueue(3, ("stop_cascade", None, None))
+        deadline = time.monotonic() + timeout
+        while not self._stop.is_set() and time.monotonic() < deadline:
+            time.sleep(0.01)
         self._stop.set()
 
     def update_with_latest_weights(self, src_rank: int | None = None,

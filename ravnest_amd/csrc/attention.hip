@@ -218,7 +218,9 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
 }
 
 // ---------------------------------------------------------------------
-// fwd v2 (D=64): K/V tiles staged once per BLOCK through LDS.
+// fwd v2 (D=64): K/V tiles staged once per BLOCK through double-buffered
+// LDS (one barrier per tile; the next tile's stage write lands in the
+// other slot under the current tile's compute).
 // The v1 kernel's four waves each re-read every K/V tile from L2 (16B
 // fragment loads + 32 scalar V loads per tile) and sat 38% of wave
 // cycles in memory waits (PMC r01). Here 256 threads cooperatively load
@@ -234,9 +236,8 @@ __global__ __launch_bounds__(256, 4) void attn_fwd_lds_kernel(
     float scale, int has_mask, long H,
     Strides sq, Strides sk, Strides sv, Strides so) {
   constexpr int D = 64;
-  __shared__ __attribute__((aligned(16))) char smem[8192];
-  bf16_t* k_lds = reinterpret_cast<bf16_t*>(smem);          // [32][64] swz
-  bf16_t* v_lds = reinterpret_cast<bf16_t*>(smem + 4096);   // [32][64]
+  __shared__ __attribute__((aligned(16))) char smem[16384];
+  // two 8KB slots: [slot][K swz 4KB | V linear 4KB]
   const int lane = threadIdx.x & (WAVE - 1);
   const int wid = threadIdx.x / WAVE;
   const int hi = lane >> 5;
@@ -277,21 +278,25 @@ __global__ __launch_bounds__(256, 4) void attn_fwd_lds_kernel(
   // every wave must loop over ALL tiles (barriers are block-wide); a
   // wave past the causal horizon just skips its compute.
   // T14 split (guide G15): tile t+1's global loads ISSUE before tile t's
-  // MFMA/softmax phase so HBM/L2 latency hides under compute; the
-  // register stage is written to LDS after the consume barrier.
+  // MFMA/softmax phase; double-buffered slots make the LDS write safe
+  // after compute (the slot's last readers passed the barrier at the top
+  // of THIS iteration).
   const int kv_all = causal ? min(S, blockIdx.x * 128 + 128) : S;
   bf16x8v kstage = *reinterpret_cast<const bf16x8v*>(
       kp + (long)min(st_row, S - 1) * sk.rs + st_c16 / 2);
   bf16x8v vstage = *reinterpret_cast<const bf16x8v*>(
       vp + (long)min(st_row, S - 1) * sv.rs + st_c16 / 2);
+  {  // prologue: stage tile 0 into slot 0
+    *reinterpret_cast<bf16x8v*>(smem + k_dst) = kstage;
+    *reinterpret_cast<bf16x8v*>(smem + 4096 + v_dst) = vstage;
+  }
   for (int k0 = 0; k0 < kv_all; k0 += 32) {
-    __syncthreads();  // previous tile fully consumed
-    *reinterpret_cast<bf16x8v*>(
-        reinterpret_cast<char*>(k_lds) + k_dst) = kstage;
-    *reinterpret_cast<bf16x8v*>(
-        reinterpret_cast<char*>(v_lds) + v_dst) = vstage;
-    __syncthreads();
-    if (k0 + 32 < kv_all) {  // issue next tile's loads under this compute
+    const int slot = (k0 >> 5) & 1;
+    char* k_lds = smem + slot * 8192;
+    char* v_lds = k_lds + 4096;
+    __syncthreads();  // this slot staged; other slot's readers done
+    const bool have_next = k0 + 32 < kv_all;
+    if (have_next) {  // issue next tile's loads under this compute
       const int nrow = min(k0 + 32 + st_row, S - 1);
       kstage = *reinterpret_cast<const bf16x8v*>(
           kp + (long)nrow * sk.rs + st_c16 / 2);
@@ -308,8 +313,7 @@ __global__ __launch_bounds__(256, 4) void attn_fwd_lds_kernel(
       for (int s = 0; s < 4; ++s) {
         const int koff = (j32 * 128 + (s * 16 + hi * 8) * 2) ^
                          ((j32 & 15) << 4);
-        bf16x8v kf = *reinterpret_cast<const bf16x8v*>(
-            reinterpret_cast<char*>(k_lds) + koff);
+        bf16x8v kf = *reinterpret_cast<const bf16x8v*>(k_lds + koff);
         s_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qf[s], s_acc,
                                                         0, 0, 0);
       }
@@ -377,13 +381,20 @@ __global__ __launch_bounds__(256, 4) void attn_fwd_lds_kernel(
 #pragma unroll
           for (int e = 0; e < 8; ++e) {
             const int key_l = step * 16 + hi * 8 + e;
-            vf[e] = v_lds[key_l * 64 + hh * 32 + j32];
+            vf[e] = *reinterpret_cast<const bf16_t*>(
+                v_lds + key_l * 128 + (hh * 32 + j32) * 2);
           }
           oacc[hh] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
               *reinterpret_cast<const bf16x8v*>(&pa[step][0]), vf,
               oacc[hh], 0, 0, 0);
         }
       }
+    }
+    if (have_next) {  // stage t+1 into the other slot (safe: its last
+                      // readers passed this iteration's barrier)
+      char* nk = smem + (slot ^ 1) * 8192;
+      *reinterpret_cast<bf16x8v*>(nk + k_dst) = kstage;
+      *reinterpret_cast<bf16x8v*>(nk + 4096 + v_dst) = vstage;
     }
   }
 

@@ -90,9 +90,8 @@ __global__ __launch_bounds__(256, 4) void attn_bwd_dv_kernel(
   // block-shared tiles: dO (linear: 2B B-operand reads) and Q
   // (((row&15)<<4)-swizzled: conflict-free b128 A-fragment reads).
   // After staging, a tile iteration does NO per-wave global reads.
-  __shared__ __attribute__((aligned(16))) char smem[8192];
-  bf16_t* do_lds = reinterpret_cast<bf16_t*>(smem);
-  bf16_t* q_lds = reinterpret_cast<bf16_t*>(smem + 4096);
+  __shared__ __attribute__((aligned(16))) char smem[16384];
+  // two 8KB slots: [slot][dO linear 4KB | Q swz 4KB]
   const int st_row = threadIdx.x >> 3;
   const int st_c16 = (threadIdx.x & 7) * 16;
   const int lane = threadIdx.x & (WAVE - 1);
@@ -126,26 +125,28 @@ __global__ __launch_bounds__(256, 4) void attn_bwd_dv_kernel(
     for (int r = 0; r < 16; ++r) dv_acc[hh][r] = 0.f;
 
   const int q_start = causal ? blockIdx.x * 128 : 0;
-  const int q_dst = (st_row * 128 + st_c16) ^ ((st_row & 15) << 4);
+  const int lin_dst = st_row * 128 + st_c16;
+  const int q_dst = lin_dst ^ ((st_row & 15) << 4);
   bf16x8v dstage = *reinterpret_cast<const bf16x8v*>(
       dop + (long)min(q_start + st_row, S - 1) * sdo.rs + st_c16 / 2);
   bf16x8v qstage = *reinterpret_cast<const bf16x8v*>(
       qp + (long)min(q_start + st_row, S - 1) * sio.rs + st_c16 / 2);
+  *reinterpret_cast<bf16x8v*>(smem + lin_dst) = dstage;
+  *reinterpret_cast<bf16x8v*>(smem + 4096 + q_dst) = qstage;
   for (int q0 = q_start; q0 < S; q0 += 32) {
+    const int slot = ((q0 - q_start) >> 5) & 1;
+    const char* do_lds = smem + slot * 8192;
+    const char* q_lds = do_lds + 4096;
     __syncthreads();
-    *reinterpret_cast<bf16x8v*>(
-        reinterpret_cast<char*>(do_lds) + st_row * 128 + st_c16) = dstage;
-    *reinterpret_cast<bf16x8v*>(
-        reinterpret_cast<char*>(q_lds) + q_dst) = qstage;
-    __syncthreads();
-    if (q0 + 32 < S) {  // T14: next tile's loads under this compute
+    const bool have_next = q0 + 32 < S;
+    if (have_next) {  // T14: next tile's loads under this compute
       dstage = *reinterpret_cast<const bf16x8v*>(
           dop + (long)min(q0 + 32 + st_row, S - 1) * sdo.rs + st_c16 / 2);
       qstage = *reinterpret_cast<const bf16x8v*>(
           qp + (long)min(q0 + 32 + st_row, S - 1) * sio.rs + st_c16 / 2);
     }
     const bool compute = live_wave && (!causal || q0 + 31 >= k0);
-    if (!compute) continue;
+    if (compute) {
     f32x16 s_acc;
 #pragma unroll
     for (int r = 0; r < 16; ++r) s_acc[r] = 0.f;
@@ -153,8 +154,7 @@ __global__ __launch_bounds__(256, 4) void attn_bwd_dv_kernel(
     for (int s = 0; s < 4; ++s) {
       const int qoff = (j32 * 128 + (s * 16 + hi * 8) * 2) ^
                        ((j32 & 15) << 4);
-      bf16x8v qf = *reinterpret_cast<const bf16x8v*>(
-          reinterpret_cast<char*>(q_lds) + qoff);
+      bf16x8v qf = *reinterpret_cast<const bf16x8v*>(q_lds + qoff);
       s_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qf, kf[s], s_acc,
                                                       0, 0, 0);
     }
@@ -177,11 +177,19 @@ __global__ __launch_bounds__(256, 4) void attn_bwd_dv_kernel(
         bf16x8v dof;
 #pragma unroll
         for (int e = 0; e < 8; ++e)
-          dof[e] = do_lds[(step * 16 + hi * 8 + e) * 64 + hh * 32 + j32];
+          dof[e] = *reinterpret_cast<const bf16_t*>(
+              do_lds + (step * 16 + hi * 8 + e) * 128 +
+              (hh * 32 + j32) * 2);
         dv_acc[hh] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
             *reinterpret_cast<const bf16x8v*>(&pa_p[step][0]), dof,
             dv_acc[hh], 0, 0, 0);
       }
+    }
+    }  // compute
+    if (have_next) {
+      char* nb = smem + (slot ^ 1) * 8192;
+      *reinterpret_cast<bf16x8v*>(nb + lin_dst) = dstage;
+      *reinterpret_cast<bf16x8v*>(nb + 4096 + q_dst) = qstage;
     }
   }
   if (!live_wave) return;
@@ -208,10 +216,8 @@ __global__ __launch_bounds__(256, 3) void attn_bwd_dk_kernel(
     StridesB sdo, StridesB sg) {
   constexpr int D = 64;
   // Q tile swizzled (b128 A-frags for S, XOR-adjusted u16 B-reads for
-  // dK) + dO tile swizzled (b128 A-frags for dP)
-  __shared__ __attribute__((aligned(16))) char smem[8192];
-  bf16_t* q_lds = reinterpret_cast<bf16_t*>(smem);
-  bf16_t* do_lds = reinterpret_cast<bf16_t*>(smem + 4096);
+  // dK) + dO tile swizzled (b128 A-frags for dP); double-buffered
+  __shared__ __attribute__((aligned(16))) char smem[16384];
   const int st_row = threadIdx.x >> 3;
   const int st_c16 = (threadIdx.x & 7) * 16;
   const int lane = threadIdx.x & (WAVE - 1);
@@ -255,21 +261,22 @@ __global__ __launch_bounds__(256, 3) void attn_bwd_dk_kernel(
       qp + (long)min(q_start + st_row, S - 1) * sio.rs + st_c16 / 2);
   bf16x8v dstage = *reinterpret_cast<const bf16x8v*>(
       dop + (long)min(q_start + st_row, S - 1) * sdo.rs + st_c16 / 2);
+  *reinterpret_cast<bf16x8v*>(smem + swz_dst) = qstage;
+  *reinterpret_cast<bf16x8v*>(smem + 4096 + swz_dst) = dstage;
   for (int q0 = q_start; q0 < S; q0 += 32) {
+    const int slot = ((q0 - q_start) >> 5) & 1;
+    const char* q_lds = smem + slot * 8192;
+    const char* do_lds = q_lds + 4096;
     __syncthreads();
-    *reinterpret_cast<bf16x8v*>(
-        reinterpret_cast<char*>(q_lds) + swz_dst) = qstage;
-    *reinterpret_cast<bf16x8v*>(
-        reinterpret_cast<char*>(do_lds) + swz_dst) = dstage;
-    __syncthreads();
-    if (q0 + 32 < S) {  // T14: next tile's loads under this compute
+    const bool have_next = q0 + 32 < S;
+    if (have_next) {  // T14: next tile's loads under this compute
       qstage = *reinterpret_cast<const bf16x8v*>(
           qp + (long)min(q0 + 32 + st_row, S - 1) * sio.rs + st_c16 / 2);
       dstage = *reinterpret_cast<const bf16x8v*>(
           dop + (long)min(q0 + 32 + st_row, S - 1) * sdo.rs + st_c16 / 2);
     }
     const bool compute = live_wave && (!causal || q0 + 31 >= k0);
-    if (!compute) continue;
+    if (compute) {
     f32x16 s_acc, dp_acc;
 #pragma unroll
     for (int r = 0; r < 16; ++r) {
@@ -280,10 +287,8 @@ __global__ __launch_bounds__(256, 3) void attn_bwd_dk_kernel(
     for (int s = 0; s < 4; ++s) {
       const int foff = (j32 * 128 + (s * 16 + hi * 8) * 2) ^
                        ((j32 & 15) << 4);
-      bf16x8v qf = *reinterpret_cast<const bf16x8v*>(
-          reinterpret_cast<char*>(q_lds) + foff);
-      bf16x8v dof = *reinterpret_cast<const bf16x8v*>(
-          reinterpret_cast<char*>(do_lds) + foff);
+      bf16x8v qf = *reinterpret_cast<const bf16x8v*>(q_lds + foff);
+      bf16x8v dof = *reinterpret_cast<const bf16x8v*>(do_lds + foff);
       s_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qf, kf[s], s_acc,
                                                       0, 0, 0);
       dp_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dof, vf[s], dp_acc,
@@ -312,13 +317,18 @@ __global__ __launch_bounds__(256, 3) void attn_bwd_dk_kernel(
           const int row = step * 16 + hi * 8 + e;
           const int boff = (row * 128 + (hh * 32 + j32) * 2) ^
                            ((row & 15) << 4);
-          qf2[e] = *reinterpret_cast<const bf16_t*>(
-              reinterpret_cast<const char*>(q_lds) + boff);
+          qf2[e] = *reinterpret_cast<const bf16_t*>(q_lds + boff);
         }
         dk_acc[hh] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
             *reinterpret_cast<const bf16x8v*>(&pa_ds[step][0]), qf2,
             dk_acc[hh], 0, 0, 0);
       }
+    }
+    }  // compute
+    if (have_next) {
+      char* nb = smem + (slot ^ 1) * 8192;
+      *reinterpret_cast<bf16x8v*>(nb + swz_dst) = qstage;
+      *reinterpret_cast<bf16x8v*>(nb + 4096 + swz_dst) = dstage;
     }
   }
   if (!live_wave) return;
@@ -346,10 +356,8 @@ __global__ __launch_bounds__(256, 3) void attn_bwd_dq_kernel(
     StridesB sdo, StridesB sg) {
   constexpr int D = 64;
   // K tile swizzled (b128 A-frags for S^T, XOR-adjusted u16 B-reads for
-  // dQ) + V tile swizzled (b128 A-frags for dP^T)
-  __shared__ __attribute__((aligned(16))) char smem[8192];
-  bf16_t* k_ldsb = reinterpret_cast<bf16_t*>(smem);
-  bf16_t* v_ldsb = reinterpret_cast<bf16_t*>(smem + 4096);
+  // dQ) + V tile swizzled (b128 A-frags for dP^T); double-buffered
+  __shared__ __attribute__((aligned(16))) char smem[16384];
   const int st_row = threadIdx.x >> 3;
   const int st_c16 = (threadIdx.x & 7) * 16;
   const int lane = threadIdx.x & (WAVE - 1);
@@ -394,21 +402,22 @@ __global__ __launch_bounds__(256, 3) void attn_bwd_dq_kernel(
       kp + (long)min(st_row, S - 1) * sio.rs + st_c16 / 2);
   bf16x8v vstage = *reinterpret_cast<const bf16x8v*>(
       vp + (long)min(st_row, S - 1) * sio.rs + st_c16 / 2);
+  *reinterpret_cast<bf16x8v*>(smem + swz_dst) = kstage;
+  *reinterpret_cast<bf16x8v*>(smem + 4096 + swz_dst) = vstage;
   for (int k0 = 0; k0 < kv_end; k0 += 32) {
+    const int slot = (k0 >> 5) & 1;
+    const char* k_ldsb = smem + slot * 8192;
+    const char* v_ldsb = k_ldsb + 4096;
     __syncthreads();
-    *reinterpret_cast<bf16x8v*>(
-        reinterpret_cast<char*>(k_ldsb) + swz_dst) = kstage;
-    *reinterpret_cast<bf16x8v*>(
-        reinterpret_cast<char*>(v_ldsb) + swz_dst) = vstage;
-    __syncthreads();
-    if (k0 + 32 < kv_end) {  // T14: next tile's loads under this compute
+    const bool have_next = k0 + 32 < kv_end;
+    if (have_next) {  // T14: next tile's loads under this compute
       kstage = *reinterpret_cast<const bf16x8v*>(
           kp + (long)min(k0 + 32 + st_row, S - 1) * sio.rs + st_c16 / 2);
       vstage = *reinterpret_cast<const bf16x8v*>(
           vp + (long)min(k0 + 32 + st_row, S - 1) * sio.rs + st_c16 / 2);
     }
     const bool compute = live_wave && (!causal || k0 <= q0 + 31);
-    if (!compute) continue;
+    if (compute) {
     f32x16 s_acc, dp_acc;
 #pragma unroll
     for (int r = 0; r < 16; ++r) {
@@ -419,10 +428,8 @@ __global__ __launch_bounds__(256, 3) void attn_bwd_dq_kernel(
     for (int s = 0; s < 4; ++s) {
       const int foff = (j32 * 128 + (s * 16 + hi * 8) * 2) ^
                        ((j32 & 15) << 4);
-      bf16x8v kfr = *reinterpret_cast<const bf16x8v*>(
-          reinterpret_cast<char*>(k_ldsb) + foff);
-      bf16x8v vfr = *reinterpret_cast<const bf16x8v*>(
-          reinterpret_cast<char*>(v_ldsb) + foff);
+      bf16x8v kfr = *reinterpret_cast<const bf16x8v*>(k_ldsb + foff);
+      bf16x8v vfr = *reinterpret_cast<const bf16x8v*>(v_ldsb + foff);
       // S^T[key][qrow], dP^T[key][qrow]
       s_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kfr, qf[s], s_acc,
                                                       0, 0, 0);
@@ -452,13 +459,18 @@ __global__ __launch_bounds__(256, 3) void attn_bwd_dq_kernel(
           const int row = step * 16 + hi * 8 + e;
           const int boff = (row * 128 + (hh * 32 + j32) * 2) ^
                            ((row & 15) << 4);
-          kcol[e] = *reinterpret_cast<const bf16_t*>(
-              reinterpret_cast<const char*>(k_ldsb) + boff);
+          kcol[e] = *reinterpret_cast<const bf16_t*>(k_ldsb + boff);
         }
         dq_acc[hh] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
             *reinterpret_cast<const bf16x8v*>(&pa_ds[step][0]), kcol,
             dq_acc[hh], 0, 0, 0);
       }
+    }
+    }  // compute
+    if (have_next) {
+      char* nb = smem + (slot ^ 1) * 8192;
+      *reinterpret_cast<bf16x8v*>(nb + swz_dst) = kstage;
+      *reinterpret_cast<bf16x8v*>(nb + 4096 + swz_dst) = vstage;
     }
   }
   if (!live_wave) return;

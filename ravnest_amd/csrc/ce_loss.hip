@@ -45,13 +45,45 @@ __global__ void ce_fwd_kernel(const T* __restrict__ logits,
   const long tgt = targets[row];
 
   float m = -INFINITY, s = 0.f;
-  for (int i = threadIdx.x; i < V; i += blockDim.x) {
-    const float v = ld(lr + i);
-    if (v > m) {
-      s = s * __expf(m - v) + 1.f;
-      m = v;
-    } else {
-      s += __expf(v - m);
+  if constexpr (sizeof(T) == 2) {
+    // vectorized: 8 bf16 per thread per iteration (16B loads)
+    const int nv = V >> 3;
+    for (int g = threadIdx.x; g < nv; g += blockDim.x) {
+      const bf16x8 v8 = *reinterpret_cast<const bf16x8*>(lr + g * 8);
+      float vals[8], vmax = -INFINITY;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        vals[j] = us2f((unsigned short)v8[j]);
+        vmax = fmaxf(vmax, vals[j]);
+      }
+      float ls = 0.f;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) ls += __expf(vals[j] - vmax);
+      if (vmax > m) {
+        s = s * __expf(m - vmax) + ls;
+        m = vmax;
+      } else {
+        s += ls * __expf(vmax - m);
+      }
+    }
+    for (int i = (nv << 3) + threadIdx.x; i < V; i += blockDim.x) {
+      const float v = ld(lr + i);
+      if (v > m) {
+        s = s * __expf(m - v) + 1.f;
+        m = v;
+      } else {
+        s += __expf(v - m);
+      }
+    }
+  } else {
+    for (int i = threadIdx.x; i < V; i += blockDim.x) {
+      const float v = ld(lr + i);
+      if (v > m) {
+        s = s * __expf(m - v) + 1.f;
+        m = v;
+      } else {
+        s += __expf(v - m);
+      }
     }
   }
   // wave merge
@@ -123,13 +155,28 @@ __global__ void ce_bwd_kernel(const T* __restrict__ logits,
     }
     return;
   }
-  for (int i = threadIdx.x; i < V; i += blockDim.x) {
-    float p = __expf(ld(lr + i) - l);
-    float d = (p - (i == (int)tgt ? 1.f : 0.f)) * g;
-    if constexpr (sizeof(T) == 2)
-      dr[i] = f2bf(d);
-    else
-      dr[i] = d;
+  if constexpr (sizeof(T) == 2) {
+    const int nv = V >> 3;
+    for (int gi = threadIdx.x; gi < nv; gi += blockDim.x) {
+      const bf16x8 v8 = *reinterpret_cast<const bf16x8*>(lr + gi * 8);
+      bf16x8 o;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const int i = gi * 8 + j;
+        float p = __expf(us2f((unsigned short)v8[j]) - l);
+        o[j] = (short)f2us((p - (i == (int)tgt ? 1.f : 0.f)) * g);
+      }
+      *reinterpret_cast<bf16x8*>(dr + gi * 8) = o;
+    }
+    for (int i = (nv << 3) + threadIdx.x; i < V; i += blockDim.x) {
+      float p = __expf(ld(lr + i) - l);
+      dr[i] = f2bf((p - (i == (int)tgt ? 1.f : 0.f)) * g);
+    }
+  } else {
+    for (int i = threadIdx.x; i < V; i += blockDim.x) {
+      float p = __expf(ld(lr + i) - l);
+      dr[i] = (p - (i == (int)tgt ? 1.f : 0.f)) * g;
+    }
   }
 }
 

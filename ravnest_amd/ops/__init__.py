@@ -5,6 +5,7 @@ from .dropout import Dropout, dropout
 from .attention import (AttentionCore, AttentionCoreQKV, attention, attention_qkv)
 from .losses import CrossEntropyLoss, cross_entropy
 from .optim import FusedAdam, FusedSGD, FusedLAMB
+from .batchnorm import FusedBatchNorm2d
 
 __all__ = [
     "get_ext", "has_ext",
@@ -14,4 +15,5 @@ __all__ = [
     "AttentionCore", "AttentionCoreQKV", "attention", "attention_qkv",
     "CrossEntropyLoss", "cross_entropy",
     "FusedAdam", "FusedSGD", "FusedLAMB",
+    "FusedBatchNorm2d",
 ]

@@ -368,3 +368,32 @@ def test_add_layer_norm(dev):
     assert torch.allclose(a.grad.float(), a2.grad, atol=8e-2, rtol=8e-2)
     assert torch.allclose(b.grad.float(), b2.grad, atol=8e-2, rtol=8e-2)
     assert torch.allclose(w.grad, w2.grad, atol=1.0, rtol=3e-2)
+
+
+def test_fused_batchnorm(dev):
+    """Fused BN2d train fwd/bwd + running stats vs torch fp32."""
+    from ravnest_amd.ops import FusedBatchNorm2d
+    torch.manual_seed(0)
+    N, C, H, W = 16, 32, 14, 14
+    bn1 = FusedBatchNorm2d(C).to(dev)
+    bn2 = torch.nn.BatchNorm2d(C).to(dev)
+    x = torch.randn(N, C, H, W, device=dev)
+    x1 = x.clone().requires_grad_(True)
+    x2 = x.clone().requires_grad_(True)
+    y1 = bn1(x1)
+    y2 = bn2(x2)
+    dy = torch.randn_like(y1)
+    y1.backward(dy)
+    y2.backward(dy)
+    assert torch.allclose(y1, y2, atol=1e-4, rtol=1e-3)
+    assert torch.allclose(x1.grad, x2.grad, atol=1e-4, rtol=1e-3)
+    assert torch.allclose(bn1.weight.grad, bn2.weight.grad, atol=1e-3,
+                          rtol=1e-3)
+    assert torch.allclose(bn1.bias.grad, bn2.bias.grad, atol=1e-3, rtol=1e-3)
+    assert torch.allclose(bn1.running_mean, bn2.running_mean, atol=1e-4)
+    assert torch.allclose(bn1.running_var, bn2.running_var, atol=1e-4)
+    # eval path
+    bn1.eval()
+    bn2.eval()
+    with torch.no_grad():
+        assert torch.allclose(bn1(x), bn2(x), atol=1e-4, rtol=1e-3)

@@ -12,12 +12,14 @@ import torch
 import torch.nn as nn
 import torch.nn.functional as F
 
+from ..ops import FusedBatchNorm2d
+
 
 class BasicConv2d(nn.Module):
     def __init__(self, in_ch, out_ch, **kw):
         super().__init__()
         self.conv = nn.Conv2d(in_ch, out_ch, bias=False, **kw)
-        self.bn = nn.BatchNorm2d(out_ch, eps=0.001)
+        self.bn = FusedBatchNorm2d(out_ch, eps=0.001)
 
     def forward(self, x):
         return F.relu(self.bn(self.conv(x)), inplace=True)

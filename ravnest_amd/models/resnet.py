@@ -15,6 +15,8 @@ from __future__ import annotations
 import torch
 import torch.nn as nn
 
+from ..ops import FusedBatchNorm2d
+
 
 class Bottleneck(nn.Module):
     expansion = 4
@@ -22,12 +24,12 @@ class Bottleneck(nn.Module):
     def __init__(self, in_ch, ch, stride=1, downsample=None):
         super().__init__()
         self.conv1 = nn.Conv2d(in_ch, ch, 1, bias=False)
-        self.bn1 = nn.BatchNorm2d(ch)
+        self.bn1 = FusedBatchNorm2d(ch)
         self.conv2 = nn.Conv2d(ch, ch, 3, stride=stride, padding=1,
                                bias=False)
-        self.bn2 = nn.BatchNorm2d(ch)
+        self.bn2 = FusedBatchNorm2d(ch)
         self.conv3 = nn.Conv2d(ch, ch * self.expansion, 1, bias=False)
-        self.bn3 = nn.BatchNorm2d(ch * self.expansion)
+        self.bn3 = FusedBatchNorm2d(ch * self.expansion)
         self.relu = nn.ReLU(inplace=True)
         self.downsample = downsample
 
@@ -46,7 +48,7 @@ class ResNet(nn.Module):
         super().__init__()
         self.in_planes = 64
         self.conv1 = nn.Conv2d(in_ch, 64, 7, stride=2, padding=3, bias=False)
-        self.bn1 = nn.BatchNorm2d(64)
+        self.bn1 = FusedBatchNorm2d(64)
         self.relu = nn.ReLU(inplace=True)
         self.maxpool = nn.MaxPool2d(3, stride=2, padding=1)
         self.layer1 = self._make_layer(64, layers[0])
@@ -60,7 +62,7 @@ class ResNet(nn.Module):
             if isinstance(m, nn.Conv2d):
                 nn.init.kaiming_normal_(m.weight, mode="fan_out",
                                         nonlinearity="relu")
-            elif isinstance(m, nn.BatchNorm2d):
+            elif isinstance(m, (nn.BatchNorm2d, FusedBatchNorm2d)):
                 nn.init.ones_(m.weight)
                 nn.init.zeros_(m.bias)
 
@@ -70,7 +72,7 @@ class ResNet(nn.Module):
             downsample = nn.Sequential(
                 nn.Conv2d(self.in_planes, ch * Bottleneck.expansion, 1,
                           stride=stride, bias=False),
-                nn.BatchNorm2d(ch * Bottleneck.expansion))
+                FusedBatchNorm2d(ch * Bottleneck.expansion))
         layers = [Bottleneck(self.in_planes, ch, stride, downsample)]
         self.in_planes = ch * Bottleneck.expansion
         for _ in range(1, blocks):

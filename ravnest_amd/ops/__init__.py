@@ -6,6 +6,7 @@ from .attention import (AttentionCore, AttentionCoreQKV, attention, attention_qk
 from .losses import CrossEntropyLoss, cross_entropy
 from .optim import FusedAdam, FusedSGD, FusedLAMB
 from .batchnorm import FusedBatchNorm2d
+from .linear import Linear
 
 __all__ = [
     "get_ext", "has_ext",
@@ -15,5 +16,5 @@ __all__ = [
     "AttentionCore", "AttentionCoreQKV", "attention", "attention_qkv",
     "CrossEntropyLoss", "cross_entropy",
     "FusedAdam", "FusedSGD", "FusedLAMB",
-    "FusedBatchNorm2d",
+    "FusedBatchNorm2d", "Linear",
 ]

@@ -153,3 +153,52 @@ def test_pp_dp_hybrid(tmp_path):
         assert len(sums) == 2
         assert abs(sums[0] - sums[1]) < 1e-3, \
             f"stage {stage} replicas diverged: {sums}"
+
+
+def _pull_worker(rank, base_dir, port, out_dir, q):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    os.chdir(out_dir)
+    set_seed(42)
+    from ravnest_amd import Node
+    node = Node(name=f"node_{rank}", base_dir=base_dir,
+                optimizer=torch.optim.Adam,
+                device=torch.device("cpu"),
+                criterion=_loss_fn,
+                labels=_make_loader())
+    node.start()
+    if rank == 1:
+        # simulate a lagging/fresh replica: perturb, then pull the peer's
+        # latest snapshot over the ctrl channel (reference latest-weights
+        # pull, communication.py:279-330)
+        with torch.no_grad():
+            for p in node.model.parameters():
+                p.add_(1.0)
+        node.update_with_latest_weights(src_rank=0)
+    import time
+    time.sleep(1.0)  # rank 0 serves the request via its dispatch thread
+    flat = torch.cat([p.detach().reshape(-1) for p in node.model.parameters()])
+    q.put((rank, flat.sum().item()))
+    node.stop()
+
+
+def test_latest_weights_pull(tmp_path):
+    """Elastic-join weight pull between DP replicas over ctrl channels."""
+    set_seed(42)
+    model = CNN()
+    x = torch.randn(2, 1, 8, 8)
+    base = str(tmp_path / "node_data")
+    pool = [NodeSpec(name=f"n{i}", ram=100 * 2**20) for i in range(2)]
+    clusterize(model, (x,), node_pool=pool, max_clusters=2, base_dir=base)
+    port = 29750 + (os.getpid() % 90)
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_pull_worker,
+                         args=(r, base, port, str(tmp_path), q))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    res = dict(q.get(timeout=120) for _ in range(2))
+    for p in procs:
+        p.join(timeout=60)
+    assert abs(res[0] - res[1]) < 1e-4, f"pull failed: {res}"

@@ -89,8 +89,10 @@ __global__ __launch_bounds__(256, 4) void attn_bwd_dv_kernel(
   constexpr int D = 64;
   // block-shared tiles: dO (linear: 2B B-operand reads) and Q
   // (((row&15)<<4)-swizzled: conflict-free b128 A-fragment reads).
-  // After staging, a tile iteration does NO per-wave global reads.
-  __shared__ __attribute__((aligned(16))) char smem[16384];
+  // After staging, a tile iteration does NO per-wave global reads;
+  // lse staged once (same rationale as dk).
+  __shared__ __attribute__((aligned(16))) char smem[16384 + 8192];
+  float* lse_lds = reinterpret_cast<float*>(smem + 16384);
   // two 8KB slots: [slot][dO linear 4KB | Q swz 4KB]
   const int st_row = threadIdx.x >> 3;
   const int st_c16 = (threadIdx.x & 7) * 16;
@@ -107,6 +109,10 @@ __global__ __launch_bounds__(256, 4) void attn_bwd_dv_kernel(
   bf16_t* dvp = dv + b * sg.bs + h * sg.hs;
   const float* lsep = lse + bh * (long)S;
   const float* mp = has_mask ? (mask + b * (long)S) : nullptr;
+  const bool lse_in_lds = S <= 2048;
+  if (lse_in_lds) {
+    for (int i = threadIdx.x; i < S; i += blockDim.x) lse_lds[i] = lsep[i];
+  }
 
   const int key = k0 + j32;
   const bf16_t* krp = kp + (long)min(key, S - 1) * sio.rs;
@@ -162,7 +168,7 @@ __global__ __launch_bounds__(256, 4) void attn_bwd_dv_kernel(
     for (int r = 0; r < 16; ++r) {
       const int qrow = q0 + (r & 3) + 8 * (r >> 2) + 4 * hi;
       const int qclmp = min(qrow, S - 1);
-      const float l = lsep[qclmp];
+      const float l = lse_in_lds ? lse_lds[qclmp] : lsep[qclmp];
       float sv = s_acc[r] * scale + mask_val;
       bool dead = (qrow >= S) || (key >= S) || (causal && key > qrow) ||
                   !isfinite(l);
@@ -216,8 +222,12 @@ __global__ __launch_bounds__(256, 3) void attn_bwd_dk_kernel(
     StridesB sdo, StridesB sg) {
   constexpr int D = 64;
   // Q tile swizzled (b128 A-frags for S, XOR-adjusted u16 B-reads for
-  // dK) + dO tile swizzled (b128 A-frags for dP); double-buffered
-  __shared__ __attribute__((aligned(16))) char smem[16384];
+  // dK) + dO tile swizzled (b128 A-frags for dP); double-buffered.
+  // + lse/delta staged once (PMC: the per-reg L2 broadcast loads of
+  // lse/delta every tile kept 58% of wave cycles parked)
+  __shared__ __attribute__((aligned(16))) char smem[16384 + 16384];
+  float* lse_lds = reinterpret_cast<float*>(smem + 16384);
+  float* dlt_lds = lse_lds + 2048;
   const int st_row = threadIdx.x >> 3;
   const int st_c16 = (threadIdx.x & 7) * 16;
   const int lane = threadIdx.x & (WAVE - 1);
@@ -235,6 +245,13 @@ __global__ __launch_bounds__(256, 3) void attn_bwd_dk_kernel(
   const float* lsep = lse + bh * (long)S;
   const float* dltp = delta + bh * (long)S;
   const float* mp = has_mask ? (mask + b * (long)S) : nullptr;
+  const bool lse_in_lds = S <= 2048;
+  if (lse_in_lds) {
+    for (int i = threadIdx.x; i < S; i += blockDim.x) {
+      lse_lds[i] = lsep[i];
+      dlt_lds[i] = dltp[i];
+    }
+  }
 
   const int key = k0 + j32;
   const bf16_t* krp = kp + (long)min(key, S - 1) * sio.rs;
@@ -298,12 +315,13 @@ __global__ __launch_bounds__(256, 3) void attn_bwd_dk_kernel(
     for (int r = 0; r < 16; ++r) {
       const int qrow = q0 + (r & 3) + 8 * (r >> 2) + 4 * hi;
       const int qclmp = min(qrow, S - 1);
-      const float l = lsep[qclmp];
+      const float l = lse_in_lds ? lse_lds[qclmp] : lsep[qclmp];
+      const float dlt = lse_in_lds ? dlt_lds[qclmp] : dltp[qclmp];
       float sv = s_acc[r] * scale + mask_val;
       bool dead = (qrow >= S) || (key >= S) || (causal && key > qrow) ||
                   !isfinite(l);
       const float p = dead ? 0.f : __expf(sv - l);
-      dp_acc[r] = dead ? 0.f : p * (dp_acc[r] - dltp[qclmp]) * scale;
+      dp_acc[r] = dead ? 0.f : p * (dp_acc[r] - dlt) * scale;
     }
     unsigned int pa_ds[2][4];
     acc_to_afrag(dp_acc, pa_ds);

@@ -72,6 +72,8 @@ def add_layer_norm(a, b, weight, bias, eps: float = 1e-12):
     """y = LayerNorm(a + b) with the residual add fused into the LN read
     pass (one HBM round-trip instead of two)."""
     if a.is_cuda:
+        if b.dtype != a.dtype:  # mixed autocast inputs: follow the residual
+            b = b.to(a.dtype)
         return _AddLayerNormFn.apply(a.contiguous(), b.contiguous(),
                                      weight, bias, eps)
     return F.layer_norm(a + b, (a.shape[-1],), weight, bias, eps)

@@ -550,6 +550,57 @@ std::vector<at::Tensor> attn_fwd_qkv(at::Tensor qkv, at::Tensor mask,
   return {o, lse};
 }
 
+// ---- MX-scaled fp8 MFMA probe (gfx950 f8f6f4 path seed) -------------
+// v_mfma_scale_f32_32x32x64_f8f6f4: A 32x64 fp8 (8 VGPR/lane = 32 vals),
+// B 64x32 fp8, C/D f32x16 (same C layout as bf16 32x32). Assumed input
+// layout (extends the verified bf16 pattern): lane l holds
+// A[i=l&31][k=(l>>5)*32 + e], e=0..31 (bytes of 8 dwords), B mirrored.
+// Scales: e8m0 bytes, 127 = 1.0 (identity) — per-32-block scale operand.
+namespace {
+typedef int i32x8 __attribute__((ext_vector_type(8)));
+
+__global__ void mfma_mx_probe_kernel(const unsigned char* __restrict__ a,
+                                     const unsigned char* __restrict__ b,
+                                     float* __restrict__ d, int sa, int sb) {
+  if (threadIdx.x >= 64) return;
+  const int lane = threadIdx.x;
+  const int hi = lane >> 5;
+  i32x8 af, bf;
+  unsigned char* afb = reinterpret_cast<unsigned char*>(&af);
+  unsigned char* bfb = reinterpret_cast<unsigned char*>(&bf);
+#pragma unroll
+  for (int e = 0; e < 32; ++e) {
+    // A stored row-major 32x64; B stored 64x32 col-read: B[k][j]
+    afb[e] = a[(lane & 31) * 64 + hi * 32 + e];
+    bfb[e] = b[(hi * 32 + e) * 32 + (lane & 31)];
+  }
+  f32x16 acc;
+#pragma unroll
+  for (int r = 0; r < 16; ++r) acc[r] = 0.f;
+  acc = __builtin_amdgcn_mfma_scale_f32_32x32x64_f8f6f4(
+      af, bf, acc, 0, 0, 0, sa, 0, sb);
+#pragma unroll
+  for (int r = 0; r < 16; ++r) {
+    const int row = (r & 3) + 8 * (r >> 2) + 4 * hi;
+    d[row * 32 + (lane & 31)] = acc[r];
+  }
+}
+}  // namespace
+
+at::Tensor mfma_mx_probe(at::Tensor a, at::Tensor b, int64_t sa, int64_t sb) {
+  // a: (32,64) uint8 fp8-e4m3 bytes; b: (64,32) uint8
+  TORCH_CHECK(a.is_cuda() && a.scalar_type() == at::kByte);
+  TORCH_CHECK(a.sizes() == at::IntArrayRef({32, 64}));
+  TORCH_CHECK(b.sizes() == at::IntArrayRef({64, 32}));
+  auto d = at::zeros({32, 32}, a.options().dtype(at::kFloat));
+  auto stream = c10::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
+  hipLaunchKernelGGL(mfma_mx_probe_kernel, dim3(1), dim3(64), 0, stream,
+                     a.data_ptr<unsigned char>(), b.data_ptr<unsigned char>(),
+                     d.data_ptr<float>(), (int)sa, (int)sb);
+  HIP_CHECK_LAST();
+  return d;
+}
+
 at::Tensor mfma_probe(at::Tensor a, at::Tensor b) {
   TORCH_CHECK(a.is_cuda() && a.scalar_type() == at::kBFloat16);
   TORCH_CHECK(a.sizes() == at::IntArrayRef({32, 16}));

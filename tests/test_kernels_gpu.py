@@ -397,3 +397,22 @@ def test_fused_batchnorm(dev):
     bn2.eval()
     with torch.no_grad():
         assert torch.allclose(bn1(x), bn2(x), atol=1e-4, rtol=1e-3)
+
+
+def test_mfma_mx_fp8_probe(dev):
+    """Validate the MX-scaled fp8 (e4m3) 32x32x64 MFMA fragment layout:
+    the fp8 path runs at 2x the bf16 MFMA rate on gfx950 (guide section
+    3/4) and is the round-2 low-precision GEMM seed. Identity scales
+    (e8m0 127 = 1.0)."""
+    from ravnest_amd.ops import get_ext
+    ext = get_ext(True)
+    torch.manual_seed(0)
+    a32 = torch.randn(32, 64, device=dev)
+    b32 = torch.randn(64, 32, device=dev)
+    a8 = a32.to(torch.float8_e4m3fn)
+    b8 = b32.to(torch.float8_e4m3fn)
+    d = ext.mfma_mx_probe(a8.view(torch.uint8), b8.view(torch.uint8),
+                          0x7F7F7F7F, 0x7F7F7F7F)
+    ref = a8.float() @ b8.float()
+    err = (d - ref).abs().max().item()
+    assert err < 1e-2 * ref.abs().max().item() + 1e-3, f"max err {err}"

@@ -19,7 +19,6 @@ import torch
 import torch.distributed as dist
 
 from .p2p import Channel, Message
-from ..strings import ActionTypes
 
 
 @dataclass(frozen=True, order=True)

@@ -24,7 +24,7 @@ SURVEY.md section 2.4), re-built device-native for MI355X:
 from __future__ import annotations
 
 import threading
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 
 import torch
 

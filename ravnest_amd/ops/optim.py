@@ -9,8 +9,6 @@ path: plain torch math (used by the CPU pipeline tests).
 """
 from __future__ import annotations
 
-import math
-
 import torch
 from torch.optim import Optimizer
 

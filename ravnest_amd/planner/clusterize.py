@@ -29,7 +29,7 @@ from pathlib import Path
 
 import torch
 
-from .placement import NodeSpec, Cluster, form_clusters, load_node_pool, mi355x_pool
+from .placement import NodeSpec, form_clusters, load_node_pool, mi355x_pool
 from .splitter import split_model_by_proportions, SplitResult
 
 

@@ -362,6 +362,170 @@ __global__ __launch_bounds__(256, 3) void attn_bwd_dk_kernel(
 }
 
 // -------------------------------------------------------------------
+// merged dv+dk kernel (A/B LOSER, kept behind RAVNEST_ATTN_MERGED_DKV=1
+// for reference): one S-matmul + one staging feeds both accumulators,
+// but the combined register set lands at 210 VGPRs = 2 waves/SIMD and
+// measures 161 TF vs the split pair's 270 TF — the occupancy tier
+// outweighs the saved matmul pass on this latency-bound structure.
+// -------------------------------------------------------------------
+__global__ __launch_bounds__(256) void attn_bwd_dvdk_kernel(
+    const bf16_t* __restrict__ q, const bf16_t* __restrict__ k,
+    const bf16_t* __restrict__ v, const bf16_t* __restrict__ dout,
+    const float* __restrict__ lse, const float* __restrict__ delta,
+    const float* __restrict__ mask, bf16_t* __restrict__ dk,
+    bf16_t* __restrict__ dv, int S, int causal, float scale, int has_mask,
+    long H, StridesB sio, StridesB sdo, StridesB sg) {
+  constexpr int D = 64;
+  __shared__ __attribute__((aligned(16))) char smem[16384 + 16384];
+  float* lse_lds = reinterpret_cast<float*>(smem + 16384);
+  float* dlt_lds = lse_lds + 2048;
+  const int st_row = threadIdx.x >> 3;
+  const int st_c16 = (threadIdx.x & 7) * 16;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int hi = lane >> 5;
+  const int j32 = lane & 31;
+  const int k0 = (blockIdx.x * 4 + (int)(threadIdx.x / WAVE)) * 32;
+  const bool live_wave = k0 < S;
+  const long bh = blockIdx.y;
+  const long b = bh / H, h = bh % H;
+  const bf16_t* qp = q + b * sio.bs + h * sio.hs;
+  const bf16_t* kp = k + b * sio.bs + h * sio.hs;
+  const bf16_t* vp = v + b * sio.bs + h * sio.hs;
+  const bf16_t* dop = dout + b * sdo.bs + h * sdo.hs;
+  bf16_t* dkp = dk + b * sg.bs + h * sg.hs;
+  bf16_t* dvp = dv + b * sg.bs + h * sg.hs;
+  const float* lsep = lse + bh * (long)S;
+  const float* dltp = delta + bh * (long)S;
+  const float* mp = has_mask ? (mask + b * (long)S) : nullptr;
+  const bool lse_in_lds = S <= 2048;
+  if (lse_in_lds) {
+    for (int i = threadIdx.x; i < S; i += blockDim.x) {
+      lse_lds[i] = lsep[i];
+      dlt_lds[i] = dltp[i];
+    }
+  }
+
+  const int key = k0 + j32;
+  const bf16_t* krp = kp + (long)min(key, S - 1) * sio.rs;
+  const bf16_t* vrp = vp + (long)min(key, S - 1) * sio.rs;
+  bf16x8v kf[4], vf[4];
+  if (live_wave) {
+#pragma unroll
+    for (int s = 0; s < 4; ++s) {
+      kf[s] = *reinterpret_cast<const bf16x8v*>(krp + s * 16 + hi * 8);
+      vf[s] = *reinterpret_cast<const bf16x8v*>(vrp + s * 16 + hi * 8);
+    }
+  }
+  const float mask_val = (has_mask && key < S) ? mp[min(key, S - 1)] : 0.f;
+
+  f32x16 dv_acc[2], dk_acc[2];
+#pragma unroll
+  for (int hh = 0; hh < 2; ++hh)
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      dv_acc[hh][r] = 0.f;
+      dk_acc[hh][r] = 0.f;
+    }
+
+  const int q_start = causal ? blockIdx.x * 128 : 0;
+  const int swz_dst = (st_row * 128 + st_c16) ^ ((st_row & 15) << 4);
+  bf16x8v qstage = *reinterpret_cast<const bf16x8v*>(
+      qp + (long)min(q_start + st_row, S - 1) * sio.rs + st_c16 / 2);
+  bf16x8v dstage = *reinterpret_cast<const bf16x8v*>(
+      dop + (long)min(q_start + st_row, S - 1) * sdo.rs + st_c16 / 2);
+  *reinterpret_cast<bf16x8v*>(smem + swz_dst) = qstage;
+  *reinterpret_cast<bf16x8v*>(smem + 4096 + swz_dst) = dstage;
+  for (int q0 = q_start; q0 < S; q0 += 32) {
+    const int slot = ((q0 - q_start) >> 5) & 1;
+    const char* q_lds = smem + slot * 8192;
+    const char* do_lds = q_lds + 4096;
+    __syncthreads();
+    const bool have_next = q0 + 32 < S;
+    if (have_next) {
+      qstage = *reinterpret_cast<const bf16x8v*>(
+          qp + (long)min(q0 + 32 + st_row, S - 1) * sio.rs + st_c16 / 2);
+      dstage = *reinterpret_cast<const bf16x8v*>(
+          dop + (long)min(q0 + 32 + st_row, S - 1) * sdo.rs + st_c16 / 2);
+    }
+    const bool compute = live_wave && (!causal || q0 + 31 >= k0);
+    if (compute) {
+      f32x16 s_acc, dp_acc;
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        s_acc[r] = 0.f;
+        dp_acc[r] = 0.f;
+      }
+#pragma unroll
+      for (int s = 0; s < 4; ++s) {
+        const int foff = (j32 * 128 + (s * 16 + hi * 8) * 2) ^
+                         ((j32 & 15) << 4);
+        bf16x8v qf = *reinterpret_cast<const bf16x8v*>(q_lds + foff);
+        bf16x8v dof = *reinterpret_cast<const bf16x8v*>(do_lds + foff);
+        s_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qf, kf[s], s_acc,
+                                                        0, 0, 0);
+        dp_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dof, vf[s], dp_acc,
+                                                         0, 0, 0);
+      }
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int qrow = q0 + (r & 3) + 8 * (r >> 2) + 4 * hi;
+        const int qclmp = min(qrow, S - 1);
+        const float l = lse_in_lds ? lse_lds[qclmp] : lsep[qclmp];
+        const float dlt = lse_in_lds ? dlt_lds[qclmp] : dltp[qclmp];
+        float sv = s_acc[r] * scale + mask_val;
+        bool dead = (qrow >= S) || (key >= S) || (causal && key > qrow) ||
+                    !isfinite(l);
+        const float p = dead ? 0.f : __expf(sv - l);
+        s_acc[r] = p;                                          // P
+        dp_acc[r] = dead ? 0.f : p * (dp_acc[r] - dlt) * scale;  // dS
+      }
+      unsigned int pa_p[2][4], pa_ds[2][4];
+      acc_to_afrag(s_acc, pa_p);
+      acc_to_afrag(dp_acc, pa_ds);
+#pragma unroll
+      for (int hh = 0; hh < 2; ++hh) {
+#pragma unroll
+        for (int step = 0; step < 2; ++step) {
+          bf16x8v dof2, qf2;
+#pragma unroll
+          for (int e = 0; e < 8; ++e) {
+            const int row = step * 16 + hi * 8 + e;
+            const int boff = (row * 128 + (hh * 32 + j32) * 2) ^
+                             ((row & 15) << 4);
+            dof2[e] = *reinterpret_cast<const bf16_t*>(do_lds + boff);
+            qf2[e] = *reinterpret_cast<const bf16_t*>(q_lds + boff);
+          }
+          dv_acc[hh] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              *reinterpret_cast<const bf16x8v*>(&pa_p[step][0]), dof2,
+              dv_acc[hh], 0, 0, 0);
+          dk_acc[hh] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              *reinterpret_cast<const bf16x8v*>(&pa_ds[step][0]), qf2,
+              dk_acc[hh], 0, 0, 0);
+        }
+      }
+    }
+    if (have_next) {
+      char* nb = smem + (slot ^ 1) * 8192;
+      *reinterpret_cast<bf16x8v*>(nb + swz_dst) = qstage;
+      *reinterpret_cast<bf16x8v*>(nb + 4096 + swz_dst) = dstage;
+    }
+  }
+  if (!live_wave) return;
+#pragma unroll
+  for (int hh = 0; hh < 2; ++hh) {
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      const int krow = k0 + (r & 3) + 8 * (r >> 2) + 4 * hi;
+      if (krow < S) {
+        const long off = (long)krow * sg.rs + hh * 32 + j32;
+        dvp[off] = f2bf(dv_acc[hh][r]);
+        dkp[off] = f2bf(dk_acc[hh][r]);
+      }
+    }
+  }
+}
+
+// -------------------------------------------------------------------
 // dq kernel: wave owns q rows [q0, q0+32); loops kv tiles (forward
 // orientation: acc = [key-pattern][qrow=lane&31]).
 // -------------------------------------------------------------------
@@ -522,14 +686,29 @@ std::vector<at::Tensor> attn_bwd_impl(
                      delta.data_ptr<float>(), S, H, NR, sdo);
   dim3 block(256);
   dim3 gridk((S + 127) / 128, B * H);
-  hipLaunchKernelGGL(attn_bwd_dv_kernel, gridk, block, 0, stream, qb, kb,
-                     dob, lse.data_ptr<float>(), mask_ptr, dvb, S,
-                     causal ? 1 : 0, (float)scale, has_mask ? 1 : 0, H, sio,
-                     sdo, sg);
-  hipLaunchKernelGGL(attn_bwd_dk_kernel, gridk, block, 0, stream, qb, kb,
-                     vb, dob, lse.data_ptr<float>(), delta.data_ptr<float>(),
-                     mask_ptr, dkb, S, causal ? 1 : 0, (float)scale,
-                     has_mask ? 1 : 0, H, sio, sdo, sg);
+  // split is the measured default: the merged kernel lands at 210 VGPRs
+  // (2 waves/SIMD) and runs at 161 TF vs the split pair's 270 TF
+  static const bool split_dkv = [] {
+    const char* e = getenv("RAVNEST_ATTN_MERGED_DKV");
+    return !(e && e[0] == '1');
+  }();
+  if (split_dkv) {
+    hipLaunchKernelGGL(attn_bwd_dv_kernel, gridk, block, 0, stream, qb, kb,
+                       dob, lse.data_ptr<float>(), mask_ptr, dvb, S,
+                       causal ? 1 : 0, (float)scale, has_mask ? 1 : 0, H,
+                       sio, sdo, sg);
+    hipLaunchKernelGGL(attn_bwd_dk_kernel, gridk, block, 0, stream, qb, kb,
+                       vb, dob, lse.data_ptr<float>(),
+                       delta.data_ptr<float>(), mask_ptr, dkb, S,
+                       causal ? 1 : 0, (float)scale, has_mask ? 1 : 0, H,
+                       sio, sdo, sg);
+  } else {
+    hipLaunchKernelGGL(attn_bwd_dvdk_kernel, gridk, block, 0, stream, qb,
+                       kb, vb, dob, lse.data_ptr<float>(),
+                       delta.data_ptr<float>(), mask_ptr, dkb, dvb, S,
+                       causal ? 1 : 0, (float)scale, has_mask ? 1 : 0, H,
+                       sio, sdo, sg);
+  }
   hipLaunchKernelGGL(attn_bwd_dq_kernel, gridk, block, 0, stream, qb, kb,
                      vb, dob, lse.data_ptr<float>(), delta.data_ptr<float>(),
                      mask_ptr, dqb, S, causal ? 1 : 0, (float)scale,

@@ -31,11 +31,6 @@ def _loss_fn(preds, targets):
         preds.reshape(-1, preds.shape[-1]).float(), targets.reshape(-1))
 
 
-class _KwTrainerLoop:
-    """Feeds kwargs microbatches like the reference's BERT custom trainer
-    (examples/bert/bert_trainer.py:11-13)."""
-
-
 def _worker(rank, base_dir, port, out_dir):
     os.environ["MASTER_ADDR"] = "127.0.0.1"
     os.environ["MASTER_PORT"] = str(port)
@@ -96,3 +91,58 @@ def test_bert_pipeline_4stage(tmp_path):
     assert len(losses) == 3 * NBATCH
     # memorizing a 6-batch dataset: loss must drop substantially
     assert np.mean(losses[-NBATCH:]) < np.mean(losses[:NBATCH]) * 0.9, losses
+
+
+def _accum_worker(rank, base_dir, port, out_dir):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    os.chdir(out_dir)
+    set_seed(42)
+    from ravnest_amd import Node, Trainer
+    ids = _data()
+    labels = [ids[i * BATCH:(i + 1) * BATCH] for i in range(NBATCH)]
+    node = Node(name=f"node_{rank}", base_dir=base_dir,
+                optimizer=torch.optim.Adam,
+                device=torch.device("cpu"), criterion=_loss_fn,
+                labels=labels * 50,
+                update_frequency=2)  # BERT-style grad accumulation
+    node.start()
+    if node.node_type.value == "root":
+        mask = torch.ones(BATCH, SEQ, dtype=torch.int64)
+        for i in range(NBATCH):
+            node.forward_compute(input_ids=labels[i], attention_mask=mask)
+        node.wait_for_backwards(timeout=300)
+        node.stop_cluster()
+    else:
+        Trainer(node=node).prelim_checks()
+    node.stop()
+
+
+def test_pipeline_grad_accumulation(tmp_path):
+    """update_frequency=2 through the pipeline: the leaf logs loss only on
+    optimizer-step boundaries (parity: reference update_frequency
+    semantics, node.py:178/compute.py:292-301)."""
+    set_seed(42)
+    cfg = BertConfig(vocab_size=VOCAB, hidden=64, layers=2, heads=2,
+                     intermediate=128, max_seq=SEQ, dropout=0.0)
+    model = BertForMLM(cfg)
+    ids = _data()[:2]
+    mask = torch.ones(2, SEQ, dtype=torch.int64)
+    base = str(tmp_path / "node_data")
+    pool = [NodeSpec(name=f"n{i}", ram=10 * 2**20) for i in range(2)]
+    clusterize(model, (ids, mask), node_pool=pool, max_clusters=1,
+               base_dir=base)
+    port = 30500 + (os.getpid() % 90)
+    ctx = mp.get_context("spawn")
+    procs = [ctx.Process(target=_accum_worker,
+                         args=(r, base, port, str(tmp_path)))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(timeout=300)
+    for p in procs:
+        assert p.exitcode == 0
+    losses = [float(x) for x in (tmp_path / "losses.txt").read_text().split()]
+    # NBATCH microbatches, step every 2 -> NBATCH/2 logged losses
+    assert len(losses) == NBATCH // 2, losses

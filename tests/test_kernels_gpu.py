@@ -10,10 +10,6 @@ import torch
 
 pytestmark = pytest.mark.gpu
 
-requires_gpu = pytest.mark.skipif(not torch.cuda.is_available(),
-                                  reason="needs MI355X")
-
-
 @pytest.fixture(scope="module")
 def dev():
     if not torch.cuda.is_available():

@@ -56,8 +56,6 @@ def _worker(rank, world, base_dir, port, out_dir):
     if rank == 0:
         trainer.train()
         node.stop_cluster()
-        losses = np.loadtxt(os.path.join(out_dir, "leaf_losses.txt")) \
-            if os.path.exists(os.path.join(out_dir, "leaf_losses.txt")) else None
     else:
         trainer.prelim_checks()
     node.stop()

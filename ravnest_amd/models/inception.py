@@ -12,13 +12,17 @@ import torch
 import torch.nn as nn
 import torch.nn.functional as F
 
-from ..ops import FusedBatchNorm2d
+from ..ops import Conv1x1, FusedBatchNorm2d
 
 
 class BasicConv2d(nn.Module):
     def __init__(self, in_ch, out_ch, **kw):
         super().__init__()
-        self.conv = nn.Conv2d(in_ch, out_ch, bias=False, **kw)
+        if kw.get("kernel_size") == 1 and "stride" not in kw and \
+                "padding" not in kw:
+            self.conv = Conv1x1(in_ch, out_ch)
+        else:
+            self.conv = nn.Conv2d(in_ch, out_ch, bias=False, **kw)
         self.bn = FusedBatchNorm2d(out_ch, eps=0.001)
 
     def forward(self, x):

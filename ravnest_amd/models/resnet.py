@@ -15,7 +15,7 @@ from __future__ import annotations
 import torch
 import torch.nn as nn
 
-from ..ops import FusedBatchNorm2d
+from ..ops import Conv1x1, FusedBatchNorm2d
 
 
 class Bottleneck(nn.Module):
@@ -23,12 +23,12 @@ class Bottleneck(nn.Module):
 
     def __init__(self, in_ch, ch, stride=1, downsample=None):
         super().__init__()
-        self.conv1 = nn.Conv2d(in_ch, ch, 1, bias=False)
+        self.conv1 = Conv1x1(in_ch, ch)
         self.bn1 = FusedBatchNorm2d(ch)
         self.conv2 = nn.Conv2d(ch, ch, 3, stride=stride, padding=1,
                                bias=False)
         self.bn2 = FusedBatchNorm2d(ch)
-        self.conv3 = nn.Conv2d(ch, ch * self.expansion, 1, bias=False)
+        self.conv3 = Conv1x1(ch, ch * self.expansion)
         self.bn3 = FusedBatchNorm2d(ch * self.expansion)
         self.relu = nn.ReLU(inplace=True)
         self.downsample = downsample
@@ -70,8 +70,8 @@ class ResNet(nn.Module):
         downsample = None
         if stride != 1 or self.in_planes != ch * Bottleneck.expansion:
             downsample = nn.Sequential(
-                nn.Conv2d(self.in_planes, ch * Bottleneck.expansion, 1,
-                          stride=stride, bias=False),
+                Conv1x1(self.in_planes, ch * Bottleneck.expansion,
+                        stride=stride),
                 FusedBatchNorm2d(ch * Bottleneck.expansion))
         layers = [Bottleneck(self.in_planes, ch, stride, downsample)]
         self.in_planes = ch * Bottleneck.expansion

@@ -7,6 +7,7 @@ from .losses import CrossEntropyLoss, cross_entropy
 from .optim import FusedAdam, FusedSGD, FusedLAMB
 from .batchnorm import FusedBatchNorm2d
 from .linear import Linear
+from .conv import Conv1x1
 
 __all__ = [
     "get_ext", "has_ext",
@@ -16,5 +17,5 @@ __all__ = [
     "AttentionCore", "AttentionCoreQKV", "attention", "attention_qkv",
     "CrossEntropyLoss", "cross_entropy",
     "FusedAdam", "FusedSGD", "FusedLAMB",
-    "FusedBatchNorm2d", "Linear",
+    "FusedBatchNorm2d", "Linear", "Conv1x1",
 ]

@@ -53,6 +53,11 @@ class FusedBatchNorm2d(nn.Module):
 
     def forward(self, x):
         if x.is_cuda:
+            # stat buffers stay fp32 even under model.to(bf16): the kernel
+            # updates them in place at fp32 precision
+            if self.running_mean.dtype != torch.float32:
+                self.running_mean.data = self.running_mean.data.float()
+                self.running_var.data = self.running_var.data.float()
             if self.training:
                 self.num_batches_tracked += 1
             return _BatchNormFn.apply(x.contiguous(), self.weight, self.bias,

@@ -24,7 +24,17 @@
 #include "common.h"
 
 typedef __bf16 bf16x8v __attribute__((ext_vector_type(8)));
+typedef __bf16 bf16x4v __attribute__((ext_vector_type(4)));
 typedef float f32x16 __attribute__((ext_vector_type(16)));
+
+// hardware transpose read (guide T10; exact semantics verified on-device
+// by tr16_probe): every lane supplies its OWN 8-B-aligned address and
+// loads 4 contiguous bf16; per 16-lane group those 16 loads (in lane
+// order) form a [4][16] row-major block M, and lane q of the group
+// receives column q, i.e. M[0..3][q].
+#define DS_TR16(p) __builtin_amdgcn_ds_read_tr16_b64_v4bf16( \
+    (__attribute__((address_space(3))) bf16x4v*)( \
+        (__attribute__((address_space(3))) void*)(p)))
 
 namespace {
 
@@ -229,6 +239,8 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
 // G4/T2: the b128 16-lane groups read rows distinct mod 16), V stored
 // linear and consumed as 2B LDS reads (16 consecutive banks per group).
 // ---------------------------------------------------------------------
+template <bool TR16>  // V B-fragment reads: tr16 hardware-transpose
+                      // gather vs eight 2-B reads (A/B, env-selected)
 __global__ __launch_bounds__(256, 4) void attn_fwd_lds_kernel(
     const bf16_t* __restrict__ q, const bf16_t* __restrict__ k,
     const bf16_t* __restrict__ v, const float* __restrict__ mask,
@@ -236,8 +248,12 @@ __global__ __launch_bounds__(256, 4) void attn_fwd_lds_kernel(
     float scale, int has_mask, long H,
     Strides sq, Strides sk, Strides sv, Strides so) {
   constexpr int D = 64;
-  __shared__ __attribute__((aligned(16))) char smem[16384];
-  // two 8KB slots: [slot][K swz 4KB | V linear 4KB]
+  // per slot: K swizzled 4KB + V linear 4KB (row stride 128B) or, for
+  // the tr16 path, V padded to 144B rows (4608B) so the 4-lanes-per-row
+  // group gather hits distinct bank pairs.
+  constexpr int VSTRIDE = TR16 ? 144 : 128;
+  constexpr int SLOT = 4096 + (TR16 ? 4608 : 4096);
+  __shared__ __attribute__((aligned(16))) char smem[2 * SLOT];
   const int lane = threadIdx.x & (WAVE - 1);
   const int wid = threadIdx.x / WAVE;
   const int hi = lane >> 5;
@@ -273,7 +289,16 @@ __global__ __launch_bounds__(256, 4) void attn_fwd_lds_kernel(
   const int st_row = threadIdx.x >> 3;
   const int st_c16 = (threadIdx.x & 7) * 16;  // byte col
   const int k_dst = (st_row * 128 + st_c16) ^ ((st_row & 15) << 4);
-  const int v_dst = st_row * 128 + st_c16;
+  const int v_dst = st_row * VSTRIDE + st_c16;
+  // ds_read_b64_tr_b16 semantics (verified by tr16_probe on gfx950):
+  // each lane supplies its OWN 8-B address; per 16-lane group the 16
+  // loads (lane q -> 4 contiguous bf16) form a [4][16] row-major block
+  // M, and lane q receives column q (M[0..3][q]). For the PV B-fragment
+  // lane l needs V[kbase+e][hh*32 + (l&31)], e=0..7 -> lane q=l&15 loads
+  // row kbase+(q>>2), cols 4*(q&3) of the group's 16-col window.
+  const int q4 = lane & 15;
+  const int v_tr_base = (8 * hi + (q4 >> 2)) * VSTRIDE +
+                        (((lane >> 4) & 1) * 16 + 4 * (q4 & 3)) * 2;
 
   // every wave must loop over ALL tiles (barriers are block-wide); a
   // wave past the causal horizon just skips its compute.
@@ -292,7 +317,7 @@ __global__ __launch_bounds__(256, 4) void attn_fwd_lds_kernel(
   }
   for (int k0 = 0; k0 < kv_all; k0 += 32) {
     const int slot = (k0 >> 5) & 1;
-    char* k_lds = smem + slot * 8192;
+    char* k_lds = smem + slot * SLOT;
     char* v_lds = k_lds + 4096;
     __syncthreads();  // this slot staged; other slot's readers done
     const bool have_next = k0 + 32 < kv_all;
@@ -372,17 +397,29 @@ __global__ __launch_bounds__(256, 4) void attn_fwd_lds_kernel(
           pa[step][3] = r2[1];
         }
       }
-      // ---- PV: V B-frags from LDS (2B reads, 16-bank groups) ----
+      // ---- PV: V B-frags. TR16: two ds_read_b64_tr_b16 per fragment
+      // (T10) replace eight 2-B reads (16-bank groups) ----
 #pragma unroll
       for (int hh = 0; hh < 2; ++hh) {
 #pragma unroll
         for (int step = 0; step < 2; ++step) {
           bf16x8v vf;
+          if constexpr (TR16) {
+            char* vtp = v_lds + v_tr_base + step * 16 * VSTRIDE + hh * 64;
+            bf16x4v lo = DS_TR16(vtp);
+            bf16x4v hi4 = DS_TR16(vtp + 4 * VSTRIDE);
 #pragma unroll
-          for (int e = 0; e < 8; ++e) {
-            const int key_l = step * 16 + hi * 8 + e;
-            vf[e] = *reinterpret_cast<const bf16_t*>(
-                v_lds + key_l * 128 + (hh * 32 + j32) * 2);
+            for (int e = 0; e < 4; ++e) {
+              vf[e] = lo[e];
+              vf[e + 4] = hi4[e];
+            }
+          } else {
+#pragma unroll
+            for (int e = 0; e < 8; ++e) {
+              const int key_l = step * 16 + hi * 8 + e;
+              vf[e] = *reinterpret_cast<const bf16_t*>(
+                  v_lds + key_l * VSTRIDE + (hh * 32 + j32) * 2);
+            }
           }
           oacc[hh] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
               *reinterpret_cast<const bf16x8v*>(&pa[step][0]), vf,
@@ -392,7 +429,7 @@ __global__ __launch_bounds__(256, 4) void attn_fwd_lds_kernel(
     }
     if (have_next) {  // stage t+1 into the other slot (safe: its last
                       // readers passed this iteration's barrier)
-      char* nk = smem + (slot ^ 1) * 8192;
+      char* nk = smem + (slot ^ 1) * SLOT;
       *reinterpret_cast<bf16x8v*>(nk + k_dst) = kstage;
       *reinterpret_cast<bf16x8v*>(nk + 4096 + v_dst) = vstage;
     }
@@ -449,6 +486,11 @@ static const bool use_lds_fwd = [] {
   return !(e && e[0] == '1');  // default: the LDS kernel for D=64
 }();
 
+static const bool use_tr16 = [] {
+  const char* e = getenv("RAVNEST_ATTN_TR16");
+  return !(e && e[0] == '0');  // default: hardware-transpose V reads
+}();
+
 std::vector<at::Tensor> attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v,
                                  at::Tensor mask, bool causal, double scale) {
   TORCH_CHECK(q.is_cuda() && q.is_contiguous() && k.is_contiguous() &&
@@ -479,8 +521,15 @@ std::vector<at::Tensor> attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v,
 
 #define LAUNCH_ATTN_FWD(DD, QP, KP, VP, OP, SQ, SK, SV, SO)                 \
   do {                                                                      \
-    if (DD == 64 && use_lds_fwd)                                            \
-      hipLaunchKernelGGL(attn_fwd_lds_kernel, grid, block, 0, stream, QP,   \
+    if (DD == 64 && use_lds_fwd && use_tr16)                                \
+      hipLaunchKernelGGL((attn_fwd_lds_kernel<true>), grid, block, 0,       \
+                         stream, QP, KP, VP, mask_ptr,                      \
+                         OP, lse.data_ptr<float>(), S,                      \
+                         causal ? 1 : 0, (float)scale, has_mask ? 1 : 0,    \
+                         (long)H, SQ, SK, SV, SO);                          \
+    else if (DD == 64 && use_lds_fwd)                                       \
+      hipLaunchKernelGGL((attn_fwd_lds_kernel<false>), grid, block, 0,      \
+                         stream, QP,                                        \
                          KP, VP, mask_ptr, OP, lse.data_ptr<float>(), S,    \
                          causal ? 1 : 0, (float)scale, has_mask ? 1 : 0,    \
                          (long)H, SQ, SK, SV, SO);                          \
@@ -599,6 +648,38 @@ at::Tensor mfma_mx_probe(at::Tensor a, at::Tensor b, int64_t sa, int64_t sb) {
                      d.data_ptr<float>(), (int)sa, (int)sb);
   HIP_CHECK_LAST();
   return d;
+}
+
+namespace {
+// empirical mapping probe for ds_read_b64_tr_b16: LDS filled with
+// identity values; each lane passes addr = base + lane*8B; dump outputs.
+__global__ void tr16_probe_kernel(short* __restrict__ out) {
+  __shared__ __bf16 buf[512];
+  if (threadIdx.x < 64) {
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      // store the element INDEX as raw bf16 bits
+      reinterpret_cast<short*>(buf)[threadIdx.x * 8 + j] =
+          (short)(threadIdx.x * 8 + j);
+    }
+  }
+  __syncthreads();
+  if (threadIdx.x >= 64) return;
+  bf16x4v r = DS_TR16(reinterpret_cast<char*>(buf) + threadIdx.x * 8);
+#pragma unroll
+  for (int j = 0; j < 4; ++j)
+    out[threadIdx.x * 4 + j] = reinterpret_cast<short*>(&r)[j];
+}
+}  // namespace
+
+at::Tensor tr16_probe() {
+  auto out = at::zeros({64, 4},
+                       at::TensorOptions().dtype(at::kShort).device(at::kCUDA));
+  auto stream = c10::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
+  hipLaunchKernelGGL(tr16_probe_kernel, dim3(1), dim3(64), 0, stream,
+                     out.data_ptr<short>());
+  HIP_CHECK_LAST();
+  return out;
 }
 
 at::Tensor mfma_probe(at::Tensor a, at::Tensor b) {

@@ -52,6 +52,7 @@ at::Tensor attn_bwd_qkv(at::Tensor qkv, at::Tensor o, at::Tensor dout,
                         at::Tensor lse, at::Tensor mask, bool causal,
                         double scale);
 at::Tensor mfma_probe(at::Tensor a, at::Tensor b);
+at::Tensor tr16_probe();
 at::Tensor mfma_mx_probe(at::Tensor a, at::Tensor b, int64_t sa, int64_t sb);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -79,6 +80,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attn_bwd_qkv", &attn_bwd_qkv,
         "flash attention bwd, packed qkv -> dqkv");
   m.def("mfma_probe", &mfma_probe, "32x32x16 bf16 MFMA layout probe");
+  m.def("tr16_probe", &tr16_probe, "ds_read_b64_tr_b16 mapping probe");
   m.def("mfma_mx_probe", &mfma_mx_probe,
         "32x32x64 MX-scaled fp8 MFMA layout probe");
 }

@@ -20,7 +20,16 @@
 #include "common.h"
 
 typedef __bf16 bf16x8v __attribute__((ext_vector_type(8)));
+typedef __bf16 bf16x4v __attribute__((ext_vector_type(4)));
 typedef float f32x16 __attribute__((ext_vector_type(16)));
+
+// hardware transpose read (guide T10; semantics verified by tr16_probe in
+// attention.hip): each lane loads 4 contiguous bf16 at its OWN 8-B
+// address; per 16-lane group the loads form a [4][16] block M (lane
+// order) and lane q receives column q.
+#define DS_TR16(p) __builtin_amdgcn_ds_read_tr16_b64_v4bf16( \
+    (__attribute__((address_space(3))) bf16x4v*)( \
+        (__attribute__((address_space(3))) void*)(p)))
 
 namespace {
 
@@ -33,6 +42,21 @@ DEVINL unsigned int pack_bf2b(float lo, float hi) {
 
 // transform a [row-pattern][col=lane] f32 acc (16 regs) into the A-operand
 // fragments covering k=rows: pa[step] holds rows 16*step + (lane>>5)*8+e.
+// per-lane byte offset for a tr16 B-fragment read of
+// B[k = kbase + 0..3][j = hh*32 + (lane&31)] from a 128-B-row LDS image,
+// optionally XOR-swizzled ((row&15)<<4, 16-B granularity — an 8-B-aligned
+// read never crosses the flipped boundary). The row-dependent XOR also
+// de-conflicts the 4-lanes-per-row gather (rows r and r+2 land 32 B
+// apart), so the swizzled images need no extra padding.
+template <bool SWZ>
+DEVINL int tr16_b_off(int lane, int kbase, int hh) {
+  const int q4 = lane & 15;
+  const int row = kbase + (q4 >> 2);
+  int off = row * 128 + hh * 64 + ((lane >> 4) & 1) * 32 + 8 * (q4 & 3);
+  if (SWZ) off ^= (row & 15) << 4;
+  return off;
+}
+
 DEVINL void acc_to_afrag(const f32x16& acc, unsigned int pa[2][4]) {
 #pragma unroll
   for (int step = 0; step < 2; ++step) {
@@ -181,11 +205,13 @@ __global__ __launch_bounds__(256, 4) void attn_bwd_dv_kernel(
 #pragma unroll
       for (int step = 0; step < 2; ++step) {
         bf16x8v dof;
+        {
+          const int kb = step * 16 + 8 * hi;
+          bf16x4v lo = DS_TR16(do_lds + tr16_b_off<false>(lane, kb, hh));
+          bf16x4v h4 = DS_TR16(do_lds + tr16_b_off<false>(lane, kb + 4, hh));
 #pragma unroll
-        for (int e = 0; e < 8; ++e)
-          dof[e] = *reinterpret_cast<const bf16_t*>(
-              do_lds + (step * 16 + hi * 8 + e) * 128 +
-              (hh * 32 + j32) * 2);
+          for (int e = 0; e < 4; ++e) { dof[e] = lo[e]; dof[e + 4] = h4[e]; }
+        }
         dv_acc[hh] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
             *reinterpret_cast<const bf16x8v*>(&pa_p[step][0]), dof,
             dv_acc[hh], 0, 0, 0);
@@ -330,12 +356,12 @@ __global__ __launch_bounds__(256, 3) void attn_bwd_dk_kernel(
 #pragma unroll
       for (int step = 0; step < 2; ++step) {
         bf16x8v qf2;
+        {
+          const int kb = step * 16 + 8 * hi;
+          bf16x4v lo = DS_TR16(q_lds + tr16_b_off<true>(lane, kb, hh));
+          bf16x4v h4 = DS_TR16(q_lds + tr16_b_off<true>(lane, kb + 4, hh));
 #pragma unroll
-        for (int e = 0; e < 8; ++e) {
-          const int row = step * 16 + hi * 8 + e;
-          const int boff = (row * 128 + (hh * 32 + j32) * 2) ^
-                           ((row & 15) << 4);
-          qf2[e] = *reinterpret_cast<const bf16_t*>(q_lds + boff);
+          for (int e = 0; e < 4; ++e) { qf2[e] = lo[e]; qf2[e + 4] = h4[e]; }
         }
         dk_acc[hh] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
             *reinterpret_cast<const bf16x8v*>(&pa_ds[step][0]), qf2,
@@ -487,13 +513,17 @@ __global__ __launch_bounds__(256) void attn_bwd_dvdk_kernel(
 #pragma unroll
         for (int step = 0; step < 2; ++step) {
           bf16x8v dof2, qf2;
+          {
+            const int kb = step * 16 + 8 * hi;
+            const int o0 = tr16_b_off<true>(lane, kb, hh);
+            const int o1 = tr16_b_off<true>(lane, kb + 4, hh);
+            bf16x4v a0 = DS_TR16(do_lds + o0), a1 = DS_TR16(do_lds + o1);
+            bf16x4v b0 = DS_TR16(q_lds + o0), b1 = DS_TR16(q_lds + o1);
 #pragma unroll
-          for (int e = 0; e < 8; ++e) {
-            const int row = step * 16 + hi * 8 + e;
-            const int boff = (row * 128 + (hh * 32 + j32) * 2) ^
-                             ((row & 15) << 4);
-            dof2[e] = *reinterpret_cast<const bf16_t*>(do_lds + boff);
-            qf2[e] = *reinterpret_cast<const bf16_t*>(q_lds + boff);
+            for (int e = 0; e < 4; ++e) {
+              dof2[e] = a0[e]; dof2[e + 4] = a1[e];
+              qf2[e] = b0[e]; qf2[e + 4] = b1[e];
+            }
           }
           dv_acc[hh] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
               *reinterpret_cast<const bf16x8v*>(&pa_p[step][0]), dof2,
@@ -636,12 +666,12 @@ __global__ __launch_bounds__(256, 3) void attn_bwd_dq_kernel(
 #pragma unroll
       for (int step = 0; step < 2; ++step) {
         bf16x8v kcol;
+        {
+          const int kb = step * 16 + 8 * hi;
+          bf16x4v lo = DS_TR16(k_ldsb + tr16_b_off<true>(lane, kb, hh));
+          bf16x4v h4 = DS_TR16(k_ldsb + tr16_b_off<true>(lane, kb + 4, hh));
 #pragma unroll
-        for (int e = 0; e < 8; ++e) {
-          const int row = step * 16 + hi * 8 + e;
-          const int boff = (row * 128 + (hh * 32 + j32) * 2) ^
-                           ((row & 15) << 4);
-          kcol[e] = *reinterpret_cast<const bf16_t*>(k_ldsb + boff);
+          for (int e = 0; e < 4; ++e) { kcol[e] = lo[e]; kcol[e + 4] = h4[e]; }
         }
         dq_acc[hh] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
             *reinterpret_cast<const bf16x8v*>(&pa_ds[step][0]), kcol,

@@ -21,7 +21,8 @@ std::vector<at::Tensor> batchnorm_fwd(at::Tensor x, at::Tensor w,
 std::vector<at::Tensor> batchnorm_bwd(at::Tensor dy, at::Tensor x,
                                       at::Tensor w, at::Tensor mean,
                                       at::Tensor rstd);
-std::vector<at::Tensor> dropout_fwd(at::Tensor x, double p, int64_t seed);
+std::vector<at::Tensor> dropout_fwd(at::Tensor x, double p, int64_t seed,
+                                    c10::optional<at::Tensor> seed_buf);
 at::Tensor dropout_bwd(at::Tensor dy, at::Tensor mask, double p);
 std::vector<at::Tensor> ce_fwd(at::Tensor logits, at::Tensor targets,
                                int64_t ignore_index);
@@ -65,7 +66,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("colsum_bf16", &colsum_bf16, "column sum bf16 -> fp32");
   m.def("batchnorm_fwd", &batchnorm_fwd, "fused BatchNorm2d fwd (NCHW)");
   m.def("batchnorm_bwd", &batchnorm_bwd, "fused BatchNorm2d bwd (NCHW)");
-  m.def("dropout_fwd", &dropout_fwd, "philox dropout fwd (replayable)");
+  m.def("dropout_fwd", &dropout_fwd, "philox dropout fwd (replayable)",
+        py::arg("x"), py::arg("p"), py::arg("seed"),
+        py::arg("seed_buf") = py::none());
   m.def("dropout_bwd", &dropout_bwd, "dropout bwd");
   m.def("ce_fwd", &ce_fwd, "fused softmax cross-entropy fwd");
   m.def("ce_bwd", &ce_bwd, "fused softmax cross-entropy bwd");

@@ -1,10 +1,16 @@
-// Dropout with counter-based Philox RNG (replayable).
+// Dropout with counter-based Philox RNG (replayable + graph-safe).
 //
 // The (seed, offset) pair comes from the HOST (drawn from the torch CPU
 // generator by the python wrapper), so the versioned-recompute engine's
 // RNG-state capture/restore replays the identical mask — SURVEY.md
 // section 2.3 "Dropout (with replayable RNG)". Each thread generates 4
 // uniforms per philox call; mask stored as uint8.
+//
+// hipGraph support: the kernel additionally XORs in a value read from a
+// DEVICE counter buffer at run time. Outside graph mode the counter is 0
+// (identical behavior); inside a captured step the engine increments the
+// counter once per replay, so each replay of the frozen host seed draws
+// a fresh mask (engine/graphstep.py).
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
 
@@ -19,8 +25,9 @@ __global__ void dropout_fwd_kernel(const T* __restrict__ x,
                                    T* __restrict__ y,
                                    unsigned char* __restrict__ mask, long n,
                                    float p, float scale,
-                                   unsigned long long seed) {
-  const Philox4 ph(seed);
+                                   unsigned long long seed,
+                                   const long long* __restrict__ seed_buf) {
+  const Philox4 ph(seed_buf ? seed ^ (unsigned long long)*seed_buf : seed);
   const long i0 = (long)(blockIdx.x * blockDim.x + threadIdx.x);
   const long stride = (long)gridDim.x * blockDim.x;
   const float thresh = p;
@@ -126,8 +133,15 @@ static int grid_cap(long work, int block) {
   return (int)std::min<long>(g, 2048);
 }
 
-std::vector<at::Tensor> dropout_fwd(at::Tensor x, double p, int64_t seed) {
+std::vector<at::Tensor> dropout_fwd(at::Tensor x, double p, int64_t seed,
+                                    c10::optional<at::Tensor> seed_buf) {
   TORCH_CHECK(x.is_cuda() && x.is_contiguous());
+  const long long* sb = nullptr;
+  if (seed_buf.has_value()) {
+    TORCH_CHECK(seed_buf->is_cuda() &&
+                seed_buf->scalar_type() == at::kLong);
+    sb = reinterpret_cast<const long long*>(seed_buf->data_ptr<int64_t>());
+  }
   const long n = x.numel();
   auto y = at::empty_like(x);
   auto mask = at::empty({n}, x.options().dtype(at::kByte));
@@ -140,12 +154,12 @@ std::vector<at::Tensor> dropout_fwd(at::Tensor x, double p, int64_t seed) {
                        reinterpret_cast<const bf16_t*>(x.data_ptr()),
                        reinterpret_cast<bf16_t*>(y.data_ptr()),
                        mask.data_ptr<unsigned char>(), n, (float)p, scale,
-                       (unsigned long long)seed);
+                       (unsigned long long)seed, sb);
   } else {
     hipLaunchKernelGGL((dropout_fwd_kernel<float>), grid, block, 0, stream,
                        x.data_ptr<float>(), y.data_ptr<float>(),
                        mask.data_ptr<unsigned char>(), n, (float)p, scale,
-                       (unsigned long long)seed);
+                       (unsigned long long)seed, sb);
   }
   HIP_CHECK_LAST();
   return {y, mask};

@@ -23,6 +23,7 @@ SURVEY.md section 2.4), re-built device-native for MI355X:
 """
 from __future__ import annotations
 
+import os
 import threading
 from dataclasses import dataclass
 
@@ -60,6 +61,15 @@ class ComputeEngine:
         # versioning=False (single-fused-stage training) skips the
         # per-step parameter snapshot clones: no recompute ever happens
         self.versioning = versioning
+        # hipGraph capture of the fused step (fwd+loss+bwd single
+        # launch; see graphstep.py). Only safe when no recompute/grad
+        # accumulation semantics are in play.
+        self._graph_step = None
+        if (not versioning and device.type == "cuda"
+                and self.update_frequency == 1 and criterion is not None
+                and os.environ.get("RAVNEST_CUDA_GRAPH", "1") == "1"):
+            from .graphstep import GraphedTrainStep
+            self._graph_step = GraphedTrainStep(model, criterion, device)
 
         self.current_version = 0
         self.version_to_param: dict[int, list[torch.Tensor]] = {}
@@ -269,6 +279,16 @@ class ComputeEngine:
         self.join_recompute()
         if not self.model.training:
             self.model.train()
+        if self._graph_step is not None and not any(needs_grad) and \
+                self.amp_dtype is None:
+            loss_t = self._graph_step.run(args, targets)
+            if loss_t is not None:
+                return self._finish_graphed_step(args, loss_t)
+        if self._graph_step is not None and self._graph_step.graphs:
+            # eager step after a capture: drop any stale graph-pool grad
+            # buffers so this backward starts clean (the next replay
+            # re-attaches its references)
+            self.model.zero_grad(set_to_none=True)
         live_args = []
         for a, ng in zip(args, needs_grad):
             if ng and torch.is_tensor(a) and a.is_floating_point():
@@ -303,6 +323,25 @@ class ComputeEngine:
                     f.write(f"{round(self.file_loss, 4)}\n")
             self.file_loss = 0.0
         return input_grads, stepped, loss_val
+
+    def _finish_graphed_step(self, args, loss_t):
+        """Post-replay bookkeeping: optimizer step stays EAGER (Adam bias
+        correction is computed host-side per step and must not freeze at
+        capture time) and grads are NOT zeroed (the captured backward
+        records assignments into stable graph-pool buffers that
+        param.grad keeps referencing — see graphstep.py)."""
+        loss_val = float(loss_t)
+        self.file_loss += loss_val
+        self.n_backwards += 1
+        stepped = False
+        if self.optimizer is not None:
+            self.optimizer.step()
+            stepped = True
+            if self.loss_filename:
+                with open(self.loss_filename, "a") as f:
+                    f.write(f"{round(self.file_loss, 4)}\n")
+            self.file_loss = 0.0
+        return [None] * len(args), stepped, loss_val
 
     # ------------------------------------------------------------------
     # eval forwards

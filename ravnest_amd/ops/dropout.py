@@ -20,9 +20,13 @@ class _DropoutFn(torch.autograd.Function):
     def forward(ctx, x, p):
         ext = get_ext(required=True)
         # seed drawn from the torch CPU generator: the engine's RNG
-        # capture/restore (fork_rng) makes the recompute replay identical
+        # capture/restore (fork_rng) makes the recompute replay identical.
+        # The device counter (0 outside graph mode) keeps the seed live
+        # across hipGraph replays — see ops/rng.py.
+        from . import rng
         seed = int(torch.randint(0, 2**62, (1,)).item())
-        y, mask = ext.dropout_fwd(x, p, seed)
+        y, mask = ext.dropout_fwd(x, p, seed,
+                                  rng.device_seed_counter(x.device))
         ctx.save_for_backward(mask)
         ctx.p = p
         return y

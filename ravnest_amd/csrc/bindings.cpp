@@ -21,6 +21,12 @@ std::vector<at::Tensor> batchnorm_fwd(at::Tensor x, at::Tensor w,
 std::vector<at::Tensor> batchnorm_bwd(at::Tensor dy, at::Tensor x,
                                       at::Tensor w, at::Tensor mean,
                                       at::Tensor rstd);
+std::vector<at::Tensor> emb2_ln_fwd(at::Tensor ids, at::Tensor word,
+                                    at::Tensor pos, at::Tensor w,
+                                    at::Tensor b, double eps);
+at::Tensor emb2_add_fwd(at::Tensor ids, at::Tensor word, at::Tensor pos);
+std::vector<at::Tensor> emb2_bwd(at::Tensor dx, at::Tensor ids, long V,
+                                 long P);
 std::vector<at::Tensor> dropout_fwd(at::Tensor x, double p, int64_t seed,
                                     c10::optional<at::Tensor> seed_buf);
 at::Tensor dropout_bwd(at::Tensor dy, at::Tensor mask, double p);
@@ -70,6 +76,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("x"), py::arg("p"), py::arg("seed"),
         py::arg("seed_buf") = py::none());
   m.def("dropout_bwd", &dropout_bwd, "dropout bwd");
+  m.def("emb2_ln_fwd", &emb2_ln_fwd,
+        "fused word+pos embedding gather + LayerNorm fwd");
+  m.def("emb2_add_fwd", &emb2_add_fwd, "fused word+pos embedding gather");
+  m.def("emb2_bwd", &emb2_bwd,
+        "embedding bwd: dword scatter-add + dpos batch reduction");
   m.def("ce_fwd", &ce_fwd, "fused softmax cross-entropy fwd");
   m.def("ce_bwd", &ce_bwd, "fused softmax cross-entropy bwd");
   m.def("fused_adam", &fused_adam, "multi-tensor Adam");

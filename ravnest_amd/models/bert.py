@@ -14,6 +14,7 @@ from __future__ import annotations
 import torch
 import torch.nn as nn
 
+from ..ops import embedding_ln
 from ..ops import (AddLayerNorm, AttentionCoreQKV, Dropout, FusedLayerNorm,
                    GELU, LinearGelu)
 
@@ -59,6 +60,12 @@ class BertEmbeddings(nn.Module):
             persistent=False)
 
     def forward(self, input_ids):
+        if input_ids.is_cuda and self.word.weight.dtype == torch.bfloat16:
+            # single-pass gather+gather+LN kernel (ops/embedding.py)
+            y = embedding_ln(input_ids, self.word.weight,
+                             self.position.weight, self.ln.weight,
+                             self.ln.bias, self.ln.eps)
+            return self.drop(y)
         S = input_ids.size(1)
         x = self.word(input_ids) + self.position(self.pos_ids[:, :S])
         return self.drop(self.ln(x))

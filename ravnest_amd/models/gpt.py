@@ -13,6 +13,7 @@ import torch
 import torch.nn as nn
 
 from ..ops import AttentionCoreQKV, Dropout, FusedLayerNorm, LinearGelu
+from ..ops import embedding_add
 
 
 class GPTConfig:
@@ -53,6 +54,9 @@ class GPTEmbeddings(nn.Module):
                              persistent=False)
 
     def forward(self, idx):
+        if idx.is_cuda and self.tok_emb.weight.dtype == torch.bfloat16:
+            x = embedding_add(idx, self.tok_emb.weight, self.pos_emb.weight)
+            return self.drop(x)
         S = idx.size(1)
         x = self.tok_emb(idx) + self.pos_emb(self.pos_ids[:, :S])
         return self.drop(x)

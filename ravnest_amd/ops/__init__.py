@@ -8,6 +8,7 @@ from .optim import FusedAdam, FusedSGD, FusedLAMB
 from .batchnorm import FusedBatchNorm2d
 from .linear import Linear
 from .conv import Conv1x1
+from .embedding import embedding_ln, embedding_add
 
 __all__ = [
     "get_ext", "has_ext",
@@ -18,4 +19,5 @@ __all__ = [
     "CrossEntropyLoss", "cross_entropy",
     "FusedAdam", "FusedSGD", "FusedLAMB",
     "FusedBatchNorm2d", "Linear", "Conv1x1",
+    "embedding_ln", "embedding_add",
 ]

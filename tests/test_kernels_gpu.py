@@ -412,3 +412,58 @@ def test_mfma_mx_fp8_probe(dev):
     ref = a8.float() @ b8.float()
     err = (d - ref).abs().max().item()
     assert err < 1e-2 * ref.abs().max().item() + 1e-3, f"max err {err}"
+
+
+def test_fused_embedding_ln(dev):
+    """emb2_ln_fwd / emb2_bwd vs the plain-torch fp32 reference
+    (gather + gather + add + LayerNorm); repeated ids exercise the
+    scatter-add accumulation."""
+    from ravnest_amd.ops.embedding import embedding_ln
+    torch.manual_seed(3)
+    V, P, H, B, S = 97, 40, 128, 4, 24
+    ids = torch.randint(0, V, (B, S), device=dev)
+    ids[0, :4] = 7  # forced collisions
+    word = torch.randn(V, H, device=dev).to(torch.bfloat16).requires_grad_()
+    pos = torch.randn(P, H, device=dev).to(torch.bfloat16).requires_grad_()
+    w = torch.randn(H, device=dev).to(torch.bfloat16).requires_grad_()
+    b = torch.randn(H, device=dev).to(torch.bfloat16).requires_grad_()
+    y = embedding_ln(ids, word, pos, w, b, 1e-12)
+    dy = torch.randn_like(y)
+    y.backward(dy)
+
+    word2 = word.detach().float().requires_grad_()
+    pos2 = pos.detach().float().requires_grad_()
+    w2 = w.detach().float().requires_grad_()
+    b2 = b.detach().float().requires_grad_()
+    x2 = torch.nn.functional.embedding(ids, word2) + pos2[:S]
+    y2 = torch.nn.functional.layer_norm(x2, (H,), w2, b2, 1e-12)
+    y2.backward(dy.float())
+
+    assert torch.allclose(y.float(), y2, atol=5e-2, rtol=5e-2)
+    assert torch.allclose(word.grad.float(), word2.grad, atol=8e-2, rtol=8e-2)
+    assert torch.allclose(pos.grad.float()[:S], pos2.grad[:S], atol=8e-2,
+                          rtol=8e-2)
+    assert pos.grad.float()[S:].abs().max() == 0
+    assert torch.allclose(w.grad.float(), w2.grad, atol=2e-1, rtol=5e-2)
+    assert torch.allclose(b.grad.float(), b2.grad, atol=2e-1, rtol=5e-2)
+
+
+def test_fused_embedding_add(dev):
+    """emb2_add_fwd (GPT stem, no LN) fwd+bwd vs plain torch."""
+    from ravnest_amd.ops.embedding import embedding_add
+    torch.manual_seed(4)
+    V, P, H, B, S = 64, 32, 256, 2, 16
+    ids = torch.randint(0, V, (B, S), device=dev)
+    word = torch.randn(V, H, device=dev).to(torch.bfloat16).requires_grad_()
+    pos = torch.randn(P, H, device=dev).to(torch.bfloat16).requires_grad_()
+    y = embedding_add(ids, word, pos)
+    dy = torch.randn_like(y)
+    y.backward(dy)
+    word2 = word.detach().float().requires_grad_()
+    pos2 = pos.detach().float().requires_grad_()
+    y2 = torch.nn.functional.embedding(ids, word2) + pos2[:S]
+    y2.backward(dy.float())
+    assert torch.allclose(y.float(), y2, atol=2e-2, rtol=2e-2)
+    assert torch.allclose(word.grad.float(), word2.grad, atol=5e-2, rtol=5e-2)
+    assert torch.allclose(pos.grad.float()[:S], pos2.grad[:S], atol=5e-2,
+                          rtol=5e-2)

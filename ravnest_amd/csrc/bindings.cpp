@@ -21,6 +21,15 @@ std::vector<at::Tensor> batchnorm_fwd(at::Tensor x, at::Tensor w,
 std::vector<at::Tensor> batchnorm_bwd(at::Tensor dy, at::Tensor x,
                                       at::Tensor w, at::Tensor mean,
                                       at::Tensor rstd);
+std::vector<at::Tensor> maxpool2d_fwd(at::Tensor x, long K, long S, long P);
+at::Tensor maxpool2d_bwd(at::Tensor dy, at::Tensor idx, long H, long W,
+                         long K, long S, long P);
+at::Tensor avgpool2d_fwd(at::Tensor x, long K, long S, long P,
+                         bool include_pad);
+at::Tensor avgpool2d_bwd(at::Tensor dy, long H, long W, long K, long S,
+                         long P, bool include_pad);
+at::Tensor global_avgpool_fwd(at::Tensor x);
+at::Tensor global_avgpool_bwd(at::Tensor dy, long H, long W);
 std::vector<at::Tensor> emb2_ln_fwd(at::Tensor ids, at::Tensor word,
                                     at::Tensor pos, at::Tensor w,
                                     at::Tensor b, double eps);
@@ -76,6 +85,13 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("x"), py::arg("p"), py::arg("seed"),
         py::arg("seed_buf") = py::none());
   m.def("dropout_bwd", &dropout_bwd, "dropout bwd");
+  m.def("maxpool2d_fwd", &maxpool2d_fwd, "MaxPool2d fwd (+argmax idx)");
+  m.def("maxpool2d_bwd", &maxpool2d_bwd, "MaxPool2d bwd (gather)");
+  m.def("avgpool2d_fwd", &avgpool2d_fwd, "AvgPool2d fwd");
+  m.def("avgpool2d_bwd", &avgpool2d_bwd, "AvgPool2d bwd");
+  m.def("global_avgpool_fwd", &global_avgpool_fwd,
+        "global (adaptive 1x1) avg pool fwd");
+  m.def("global_avgpool_bwd", &global_avgpool_bwd, "global avg pool bwd");
   m.def("emb2_ln_fwd", &emb2_ln_fwd,
         "fused word+pos embedding gather + LayerNorm fwd");
   m.def("emb2_add_fwd", &emb2_add_fwd, "fused word+pos embedding gather");

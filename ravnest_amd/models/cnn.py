@@ -8,6 +8,8 @@ walkthrough and the CPU plumbing config of BASELINE.json.
 import torch
 import torch.nn as nn
 
+from ..ops import MaxPool2d as KMaxPool2d
+
 
 class CNN(nn.Module):
     def __init__(self, in_channels: int = 1, n_classes: int = 10,
@@ -19,7 +21,7 @@ class CNN(nn.Module):
         self.bn1 = nn.BatchNorm2d(16)
         self.conv2 = nn.Conv2d(16, 32, 3, stride=1, padding=1)
         self.act2 = nn.ReLU()
-        self.pool2 = nn.MaxPool2d(2)
+        self.pool2 = KMaxPool2d(2)
         self.drop2 = nn.Dropout2d(dropout)
         self.bn2 = nn.BatchNorm2d(32)
         self.flatten = nn.Flatten()

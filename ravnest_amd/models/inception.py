@@ -13,6 +13,8 @@ import torch.nn as nn
 import torch.nn.functional as F
 
 from ..ops import Conv1x1, FusedBatchNorm2d
+from ..ops import (MaxPool2d as KMaxPool2d, AvgPool2d as KAvgPool2d,
+                   AdaptiveAvgPool2d as KAdaptiveAvgPool2d)
 
 
 class BasicConv2d(nn.Module):
@@ -39,7 +41,7 @@ class InceptionA(nn.Module):
         self.branch3x3dbl_2 = BasicConv2d(64, 96, kernel_size=3, padding=1)
         self.branch3x3dbl_3 = BasicConv2d(96, 96, kernel_size=3, padding=1)
         self.branch_pool = BasicConv2d(in_ch, pool_features, kernel_size=1)
-        self.pool = nn.AvgPool2d(3, stride=1, padding=1)
+        self.pool = KAvgPool2d(3, stride=1, padding=1)
 
     def forward(self, x):
         b1 = self.branch1x1(x)
@@ -56,7 +58,7 @@ class InceptionB(nn.Module):
         self.branch3x3dbl_1 = BasicConv2d(in_ch, 64, kernel_size=1)
         self.branch3x3dbl_2 = BasicConv2d(64, 96, kernel_size=3, padding=1)
         self.branch3x3dbl_3 = BasicConv2d(96, 96, kernel_size=3, stride=2)
-        self.pool = nn.MaxPool2d(3, stride=2)
+        self.pool = KMaxPool2d(3, stride=2)
 
     def forward(self, x):
         return torch.cat([self.branch3x3(x),
@@ -84,7 +86,7 @@ class InceptionC(nn.Module):
         self.branch7x7dbl_5 = BasicConv2d(ch7, 192, kernel_size=(1, 7),
                                           padding=(0, 3))
         self.branch_pool = BasicConv2d(in_ch, 192, kernel_size=1)
-        self.pool = nn.AvgPool2d(3, stride=1, padding=1)
+        self.pool = KAvgPool2d(3, stride=1, padding=1)
 
     def forward(self, x):
         b1 = self.branch1x1(x)
@@ -106,7 +108,7 @@ class InceptionD(nn.Module):
         self.branch7x7x3_3 = BasicConv2d(192, 192, kernel_size=(7, 1),
                                          padding=(3, 0))
         self.branch7x7x3_4 = BasicConv2d(192, 192, kernel_size=3, stride=2)
-        self.pool = nn.MaxPool2d(3, stride=2)
+        self.pool = KMaxPool2d(3, stride=2)
 
     def forward(self, x):
         return torch.cat([
@@ -132,7 +134,7 @@ class InceptionE(nn.Module):
         self.branch3x3dbl_3b = BasicConv2d(384, 384, kernel_size=(3, 1),
                                            padding=(1, 0))
         self.branch_pool = BasicConv2d(in_ch, 192, kernel_size=1)
-        self.pool = nn.AvgPool2d(3, stride=1, padding=1)
+        self.pool = KAvgPool2d(3, stride=1, padding=1)
 
     def forward(self, x):
         b1 = self.branch1x1(x)
@@ -165,7 +167,7 @@ class Inception3(nn.Module):
         self.mixed7a = InceptionD(768)
         self.mixed7b = InceptionE(1280)
         self.mixed7c = InceptionE(2048)
-        self.avgpool = nn.AdaptiveAvgPool2d((1, 1))
+        self.avgpool = KAdaptiveAvgPool2d((1, 1))
         self.flatten = nn.Flatten()
         self.dropout = nn.Dropout(0.5)
         self.fc = nn.Linear(2048, num_classes)

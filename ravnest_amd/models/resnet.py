@@ -15,7 +15,8 @@ from __future__ import annotations
 import torch
 import torch.nn as nn
 
-from ..ops import Conv1x1, FusedBatchNorm2d
+from ..ops import (AdaptiveAvgPool2d, Conv1x1, FusedBatchNorm2d,
+                   MaxPool2d)
 
 
 class Bottleneck(nn.Module):
@@ -50,12 +51,12 @@ class ResNet(nn.Module):
         self.conv1 = nn.Conv2d(in_ch, 64, 7, stride=2, padding=3, bias=False)
         self.bn1 = FusedBatchNorm2d(64)
         self.relu = nn.ReLU(inplace=True)
-        self.maxpool = nn.MaxPool2d(3, stride=2, padding=1)
+        self.maxpool = MaxPool2d(3, stride=2, padding=1)
         self.layer1 = self._make_layer(64, layers[0])
         self.layer2 = self._make_layer(128, layers[1], stride=2)
         self.layer3 = self._make_layer(256, layers[2], stride=2)
         self.layer4 = self._make_layer(512, layers[3], stride=2)
-        self.avgpool = nn.AdaptiveAvgPool2d((1, 1))
+        self.avgpool = AdaptiveAvgPool2d((1, 1))
         self.flatten = nn.Flatten()
         self.fc = nn.Linear(512 * Bottleneck.expansion, num_classes)
         for m in self.modules():

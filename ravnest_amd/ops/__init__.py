@@ -9,6 +9,7 @@ from .batchnorm import FusedBatchNorm2d
 from .linear import Linear
 from .conv import Conv1x1
 from .embedding import embedding_ln, embedding_add
+from .pool import MaxPool2d, AvgPool2d, AdaptiveAvgPool2d
 
 __all__ = [
     "get_ext", "has_ext",
@@ -20,4 +21,5 @@ __all__ = [
     "FusedAdam", "FusedSGD", "FusedLAMB",
     "FusedBatchNorm2d", "Linear", "Conv1x1",
     "embedding_ln", "embedding_add",
+    "MaxPool2d", "AvgPool2d", "AdaptiveAvgPool2d",
 ]

@@ -467,3 +467,50 @@ def test_fused_embedding_add(dev):
     assert torch.allclose(word.grad.float(), word2.grad, atol=5e-2, rtol=5e-2)
     assert torch.allclose(pos.grad.float()[:S], pos2.grad[:S], atol=5e-2,
                           rtol=5e-2)
+
+
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+@pytest.mark.parametrize("k,s,p", [(3, 2, 1), (3, 2, 0), (2, 2, 0)])
+def test_maxpool2d_kernel(dev, dtype, k, s, p):
+    from ravnest_amd.ops.pool import _MaxPoolFn
+    torch.manual_seed(5)
+    x = torch.randn(3, 5, 17, 17, device=dev).to(dtype).requires_grad_()
+    y = _MaxPoolFn.apply(x, k, s, p)
+    dy = torch.randn_like(y)
+    y.backward(dy)
+    x2 = x.detach().float().requires_grad_()
+    y2 = torch.nn.functional.max_pool2d(x2, k, s, p)
+    y2.backward(dy.float())
+    tol = 1e-5 if dtype == torch.float32 else 2e-2
+    assert torch.allclose(y.float(), y2, atol=tol, rtol=tol)
+    assert torch.allclose(x.grad.float(), x2.grad, atol=tol, rtol=tol)
+
+
+@pytest.mark.parametrize("inc", [True, False])
+def test_avgpool2d_kernel(dev, inc):
+    from ravnest_amd.ops.pool import _AvgPoolFn
+    torch.manual_seed(6)
+    x = torch.randn(2, 4, 15, 15, device=dev).requires_grad_()
+    y = _AvgPoolFn.apply(x, 3, 1, 1, inc)
+    dy = torch.randn_like(y)
+    y.backward(dy)
+    x2 = x.detach().requires_grad_()
+    y2 = torch.nn.functional.avg_pool2d(x2, 3, 1, 1,
+                                        count_include_pad=inc)
+    y2.backward(dy)
+    assert torch.allclose(y, y2, atol=1e-5, rtol=1e-5)
+    assert torch.allclose(x.grad, x2.grad, atol=1e-5, rtol=1e-5)
+
+
+def test_global_avgpool_kernel(dev):
+    from ravnest_amd.ops.pool import _GlobalAvgPoolFn
+    torch.manual_seed(7)
+    x = torch.randn(4, 8, 9, 9, device=dev).requires_grad_()
+    y = _GlobalAvgPoolFn.apply(x)
+    dy = torch.randn_like(y)
+    y.backward(dy)
+    x2 = x.detach().requires_grad_()
+    y2 = torch.nn.functional.adaptive_avg_pool2d(x2, (1, 1))
+    y2.backward(dy)
+    assert torch.allclose(y, y2, atol=1e-5, rtol=1e-5)
+    assert torch.allclose(x.grad, x2.grad, atol=1e-5, rtol=1e-5)

@@ -60,6 +60,8 @@ class _GlobalAvgPoolFn(torch.autograd.Function):
 
 
 class MaxPool2d(nn.Module):
+    _is_leaf_module = True  # fx: device-dependent dispatch
+
     def __init__(self, kernel_size, stride=None, padding=0):
         super().__init__()
         self.k = kernel_size
@@ -73,6 +75,8 @@ class MaxPool2d(nn.Module):
 
 
 class AvgPool2d(nn.Module):
+    _is_leaf_module = True  # fx: device-dependent dispatch
+
     def __init__(self, kernel_size, stride=None, padding=0,
                  count_include_pad=True):
         super().__init__()
@@ -91,6 +95,8 @@ class AvgPool2d(nn.Module):
 
 class AdaptiveAvgPool2d(nn.Module):
     """Only the (1,1) target the reference workloads use."""
+
+    _is_leaf_module = True  # fx: device-dependent dispatch
 
     def __init__(self, output_size=(1, 1)):
         super().__init__()

@@ -21,6 +21,8 @@ std::vector<at::Tensor> batchnorm_fwd(at::Tensor x, at::Tensor w,
 std::vector<at::Tensor> batchnorm_bwd(at::Tensor dy, at::Tensor x,
                                       at::Tensor w, at::Tensor mean,
                                       at::Tensor rstd);
+at::Tensor softmax_fwd(at::Tensor x);
+at::Tensor softmax_bwd(at::Tensor dy, at::Tensor y);
 std::vector<at::Tensor> maxpool2d_fwd(at::Tensor x, long K, long S, long P);
 at::Tensor maxpool2d_bwd(at::Tensor dy, at::Tensor idx, long H, long W,
                          long K, long S, long P);
@@ -85,6 +87,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("x"), py::arg("p"), py::arg("seed"),
         py::arg("seed_buf") = py::none());
   m.def("dropout_bwd", &dropout_bwd, "dropout bwd");
+  m.def("softmax_fwd", &softmax_fwd, "standalone softmax fwd (last dim)");
+  m.def("softmax_bwd", &softmax_bwd, "standalone softmax bwd");
   m.def("maxpool2d_fwd", &maxpool2d_fwd, "MaxPool2d fwd (+argmax idx)");
   m.def("maxpool2d_bwd", &maxpool2d_bwd, "MaxPool2d bwd (gather)");
   m.def("avgpool2d_fwd", &avgpool2d_fwd, "AvgPool2d fwd");

@@ -9,6 +9,7 @@ import torch
 import torch.nn as nn
 
 from ..ops import MaxPool2d as KMaxPool2d
+from ..ops import Softmax as KSoftmax
 
 
 class CNN(nn.Module):
@@ -30,10 +31,11 @@ class CNN(nn.Module):
         self.drop3 = nn.Dropout(dropout * 2)
         self.bn3 = nn.BatchNorm1d(256)
         self.fc2 = nn.Linear(256, n_classes)
+        self.softmax = KSoftmax(-1)
 
     def forward(self, x):
         x = self.bn1(self.drop1(self.act1(self.conv1(x))))
         x = self.bn2(self.drop2(self.pool2(self.act2(self.conv2(x)))))
         x = self.flatten(x)
         x = self.bn3(self.drop3(self.act3(self.fc1(x))))
-        return torch.softmax(self.fc2(x), dim=-1)
+        return self.softmax(self.fc2(x))

@@ -1,6 +1,7 @@
 from ._ext import get_ext, has_ext
 from .layernorm import (FusedLayerNorm, AddLayerNorm, layer_norm, add_layer_norm)
-from .activation import GELU, LinearGelu, gelu, bias_gelu
+from .activation import (GELU, LinearGelu, Softmax, gelu, bias_gelu,
+                         softmax)
 from .dropout import Dropout, dropout
 from .attention import (AttentionCore, AttentionCoreQKV, attention, attention_qkv)
 from .losses import CrossEntropyLoss, cross_entropy
@@ -14,7 +15,7 @@ from .pool import MaxPool2d, AvgPool2d, AdaptiveAvgPool2d
 __all__ = [
     "get_ext", "has_ext",
     "FusedLayerNorm", "AddLayerNorm", "layer_norm", "add_layer_norm",
-    "GELU", "LinearGelu", "gelu", "bias_gelu",
+    "GELU", "LinearGelu", "Softmax", "gelu", "bias_gelu", "softmax",
     "Dropout", "dropout",
     "AttentionCore", "AttentionCoreQKV", "attention", "attention_qkv",
     "CrossEntropyLoss", "cross_entropy",

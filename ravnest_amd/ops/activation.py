@@ -90,3 +90,37 @@ class LinearGelu(nn.Module):
     def forward(self, x):
         h = torch.nn.functional.linear(x, self.weight)
         return bias_gelu(h, self.bias)
+
+
+class _SoftmaxFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x):
+        ext = get_ext(required=True)
+        y = ext.softmax_fwd(x.contiguous())
+        ctx.save_for_backward(y)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        (y,) = ctx.saved_tensors
+        ext = get_ext(required=True)
+        return ext.softmax_bwd(dy.contiguous(), y)
+
+
+def softmax(x: torch.Tensor, dim: int = -1) -> torch.Tensor:
+    """Standalone softmax (csrc/softmax.hip) over the last dim; torch
+    fallback on CPU or non-last dims. SURVEY.md section 2.3 softmax row."""
+    if x.is_cuda and (dim == -1 or dim == x.dim() - 1):
+        return _SoftmaxFn.apply(x)
+    return torch.softmax(x, dim)
+
+
+class Softmax(nn.Module):
+    _is_leaf_module = True  # fx: device-dependent dispatch
+
+    def __init__(self, dim: int = -1):
+        super().__init__()
+        self.dim = dim
+
+    def forward(self, x):
+        return softmax(x, self.dim)

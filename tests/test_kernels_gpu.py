@@ -514,3 +514,20 @@ def test_global_avgpool_kernel(dev):
     y2.backward(dy)
     assert torch.allclose(y, y2, atol=1e-5, rtol=1e-5)
     assert torch.allclose(x.grad, x2.grad, atol=1e-5, rtol=1e-5)
+
+
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+@pytest.mark.parametrize("D", [10, 64, 1000])
+def test_softmax_kernel(dev, dtype, D):
+    from ravnest_amd.ops import softmax
+    torch.manual_seed(8)
+    x = (torch.randn(33, D, device=dev) * 4).to(dtype).requires_grad_()
+    y = softmax(x, -1)
+    dy = torch.randn_like(y)
+    y.backward(dy)
+    x2 = x.detach().float().requires_grad_()
+    y2 = torch.softmax(x2, -1)
+    y2.backward(dy.float())
+    tol = 1e-5 if dtype == torch.float32 else 1e-2
+    assert torch.allclose(y.float(), y2, atol=tol, rtol=1e-2)
+    assert torch.allclose(x.grad.float(), x2.grad, atol=tol * 5, rtol=2e-2)

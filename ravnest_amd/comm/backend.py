@@ -91,6 +91,14 @@ def dp_groups_from_plan(base_dir: str) -> list[list[int]]:
     return [[c["stage_ranks"][s] for c in clusters] for s in range(n_stages)]
 
 
+def dp_segments_from_plan(base_dir: str) -> list[dict]:
+    """Param-range averaging segments for unequal-cluster DP (parity:
+    reference param-range rings, operations/utils.py:463-516)."""
+    with open(Path(base_dir) / "plan.json") as f:
+        plan = json.load(f)
+    return plan.get("dp_segments", [])
+
+
 class CommBackend:
     """Owns the distributed init, all edge channels and DP groups for one
     rank."""
@@ -98,6 +106,7 @@ class CommBackend:
     def __init__(self, rank: int, world_size: int,
                  edges: list[Edge] | None = None,
                  dp_groups: list[list[int]] | None = None,
+                 dp_segments: list[dict] | None = None,
                  base_dir: str | None = None,
                  device: torch.device = torch.device("cpu"),
                  backend: str | None = None,
@@ -124,6 +133,8 @@ class CommBackend:
             edges = edges_from_plan(base_dir)
         if dp_groups is None and base_dir is not None:
             dp_groups = dp_groups_from_plan(base_dir)
+        if dp_segments is None and base_dir is not None:
+            dp_segments = dp_segments_from_plan(base_dir)
         self.edges = edges or []
         self.dp_group_ranks = dp_groups or []
 
@@ -144,6 +155,22 @@ class CommBackend:
             if rank in ranks:
                 self.my_dp_group = g
                 self.my_dp_ranks = ranks
+
+        # param-range segments (unequal clusters): one group per distinct
+        # owner-rank tuple, created by EVERY rank in the same order
+        self.dp_segments = []          # (ranks, group, param_names)
+        self.my_dp_segments = []
+        seg_groups: dict[tuple, object] = {}
+        for seg in (dp_segments or []):
+            t = tuple(seg["ranks"])
+            if len(set(t)) <= 1:
+                continue
+            if t not in seg_groups:
+                seg_groups[t] = dist.new_group(sorted(set(t)))
+            g = seg_groups[t]
+            self.dp_segments.append((list(t), g, seg["params"]))
+            if rank in t:
+                self.my_dp_segments.append((list(t), g, seg["params"]))
 
         self._warmed = False
 

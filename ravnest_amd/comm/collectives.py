@@ -78,3 +78,31 @@ def average_optimizer_state(optimizer: torch.optim.Optimizer, group,
             if isinstance(v, torch.Tensor) and v.is_floating_point() and v.numel() > 0:
                 tensors.append(v)
     average_tensors(tensors, group, bucket_bytes)
+
+
+def average_parameter_segments(model: torch.nn.Module, segments,
+                               optimizer=None,
+                               bucket_bytes: int = 64 * 2**20) -> None:
+    """Unequal-cluster DP: average each param-range segment on its own
+    group (parity: reference param-range rings,
+    operations/utils.py:463-516). `segments` = [(ranks, group, names)]
+    restricted to this rank; optionally averages the matching optimizer
+    state (moments / momentum / fp32 masters)."""
+    named = dict(model.named_parameters())
+    for _, group, names in segments:
+        params = [named[n] for n in names if n in named]
+        if not params:
+            continue
+        average_tensors(params, group, bucket_bytes)
+        if optimizer is not None:
+            sts = []
+            for p in params:
+                st = optimizer.state.get(p)
+                if not st:
+                    continue
+                for v in st.values():
+                    if isinstance(v, torch.Tensor) and \
+                            v.is_floating_point() and v.numel() > 0:
+                        sts.append(v)
+            if sts:
+                average_tensors(sts, group, bucket_bytes)

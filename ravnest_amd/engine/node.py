@@ -30,7 +30,9 @@ from pathlib import Path
 import torch
 
 from ..comm import CommBackend, Message
-from ..comm.collectives import average_parameters, average_optimizer_state
+from ..comm.collectives import (average_parameters,
+                                average_optimizer_state,
+                                average_parameter_segments)
 from ..strings import ActionTypes, NodeTypes
 from ..utils import load_node_json_configs
 from .compute import ComputeEngine
@@ -616,8 +618,16 @@ class Node:
     def _maybe_reduce(self):
         if not self.reduce_threshold or self.comm is None:
             return
-        if self.engine.n_backwards % self.reduce_threshold == 0 and \
-                len(self.dp_ranks) > 1:
+        if self.engine.n_backwards % self.reduce_threshold != 0:
+            return
+        if getattr(self.comm, "my_dp_segments", None):
+            # unequal-cluster DP: per-param-range groups
+            average_parameter_segments(
+                self.model, self.comm.my_dp_segments,
+                self.optimizer if self.average_optim else None)
+            self.engine.bump_version()
+            return
+        if len(self.dp_ranks) > 1:
             average_parameters(self.model, self.comm.my_dp_group)
             if self.average_optim and self.optimizer is not None:
                 average_optimizer_state(self.optimizer, self.comm.my_dp_group)
@@ -797,7 +807,15 @@ class _CommSessionFacade:
 
     def parallel_ring_reduce(self):
         n = self.node
-        if n.comm is None or len(n.dp_ranks) <= 1:
+        if n.comm is None:
+            return
+        if getattr(n.comm, "my_dp_segments", None):
+            average_parameter_segments(
+                n.model, n.comm.my_dp_segments,
+                n.optimizer if n.average_optim else None)
+            n.engine.bump_version()
+            return
+        if len(n.dp_ranks) <= 1:
             return
         average_parameters(n.model, n.comm.my_dp_group)
         if n.average_optim and n.optimizer is not None:

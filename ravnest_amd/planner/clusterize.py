@@ -29,7 +29,8 @@ from pathlib import Path
 
 import torch
 
-from .placement import NodeSpec, form_clusters, load_node_pool, mi355x_pool
+from .placement import (Cluster, NodeSpec, form_clusters,
+                        load_node_pool, mi355x_pool)
 from .splitter import split_model_by_proportions, SplitResult
 
 
@@ -53,6 +54,7 @@ def clusterize(model: torch.nn.Module,
                node_config_path: str | None = None,
                n_gpus: int | None = None,
                max_clusters: int = 5,
+               cluster_assignment: list[list[int]] | None = None,
                base_dir: str = "node_data",
                concrete_args: dict | None = None,
                pass_data=None,
@@ -87,8 +89,14 @@ def clusterize(model: torch.nn.Module,
                                   bandwidth=1.0) for i in range(3)]
 
     model_bytes = estimate_model_bytes(model)
-    clusters = form_clusters(node_pool, model_bytes,
-                             max_clusters=max_clusters, seed=seed)
+    if cluster_assignment is not None:
+        # explicit placement (e.g. unequal-cluster DP): lists of pool
+        # indices, one list per cluster
+        clusters = [Cluster(cid=i, nodes=[node_pool[j] for j in idxs])
+                    for i, idxs in enumerate(cluster_assignment)]
+    else:
+        clusters = form_clusters(node_pool, model_bytes,
+                                 max_clusters=max_clusters, seed=seed)
     clusters.sort(key=lambda c: -len(c.nodes))  # largest first (ring owner)
 
     # ---- split the model once per distinct cluster size/proportions ----
@@ -120,17 +128,43 @@ def clusterize(model: torch.nn.Module,
 
     world_size = rank
 
-    # DP groups: ranks holding the same stage index across clusters.
-    # (Equal-size clusters on the homogeneous MI355X node; for unequal
-    # clusters the reference's param-range rings apply — tracked as a gap.)
+    # DP groups: ranks holding the same stage index across clusters
+    # (equal-size clusters). For UNEQUAL clusters the reference forms
+    # param-range rings (operations/utils.py:463-516): every parameter
+    # belongs to exactly one stage per cluster, so bucketing parameters
+    # by their (owner-rank per cluster) tuple yields one averaging group
+    # per bucket — emitted as dp_segments and realized as RCCL groups.
     sizes = {len(c.nodes) for _, c, _, _ in cluster_infos}
     uniform = len(sizes) == 1
     n_stages_max = max(sizes)
     dp_groups: list[list[int]] = []
+    dp_segments: list[dict] = []
     if uniform:
         for s in range(n_stages_max):
             group = [info[3][s] for info in cluster_infos]
             dp_groups.append(group)
+    elif len(cluster_infos) > 1:
+        owners = []  # per cluster: param name -> owning rank
+        for _, _, split, stage_ranks in cluster_infos:
+            o = {}
+            for s, stage_gm in enumerate(split.stages):
+                for name, _ in stage_gm.named_parameters():
+                    o[name] = stage_ranks[s]
+            owners.append(o)
+        buckets: dict[tuple, list[str]] = {}
+        order: list[tuple] = []
+        for s, stage_gm in enumerate(cluster_infos[0][2].stages):
+            for name, _ in stage_gm.named_parameters():
+                t = tuple(o.get(name) for o in owners)
+                if None in t:
+                    continue  # dropped by a cluster's split: not averaged
+                if t not in buckets:
+                    buckets[t] = []
+                    order.append(t)
+                buckets[t].append(name)
+        dp_segments = [{"ranks": list(t), "params": buckets[t]}
+                       for t in order]
+    meta["dp_segments"] = dp_segments
 
     # ---- emit per-cluster stage artifacts & per-rank node JSON ----
     for c_idx, cluster, split, stage_ranks in cluster_infos:
@@ -169,6 +203,7 @@ def clusterize(model: torch.nn.Module,
                 "dp_ranks": dp_ranks,
                 "ring_size": len(dp_ranks),
                 "dp_groups": dp_groups,
+                "dp_segments": dp_segments,
                 "all_stage_ranks": [info[3] for info in cluster_infos],
             }
             with open(base / "nodes" / f"node_{r}.json", "w") as f:

@@ -330,13 +330,17 @@ class ComputeEngine:
         capture time) and grads are NOT zeroed (the captured backward
         records assignments into stable graph-pool buffers that
         param.grad keeps referencing — see graphstep.py)."""
-        loss_val = float(loss_t)
-        self.file_loss += loss_val
+        # ENQUEUE the optimizer before reading the loss: float(loss_t)
+        # synchronizes on the replay, and the optimizer's host-side chunk
+        # building would otherwise run on an idle GPU
         self.n_backwards += 1
         stepped = False
         if self.optimizer is not None:
             self.optimizer.step()
             stepped = True
+        loss_val = float(loss_t)
+        self.file_loss += loss_val
+        if stepped:
             if self.loss_filename:
                 with open(self.loss_filename, "a") as f:
                     f.write(f"{round(self.file_loss, 4)}\n")

@@ -470,7 +470,15 @@ class Node:
                 if src.get("pyscalar") and torch.is_tensor(t):
                     args.append(int(t.item()))
                 elif torch.is_tensor(t):
-                    args.append(t.to(self.device))
+                    t = t.to(self.device)
+                    # decompress wire-cast activations to the template
+                    # dtype (parity: reference compute.py:160-163)
+                    want = src.get("dtype")
+                    if want and t.is_floating_point():
+                        wd = _dtype_from_str(want)
+                        if wd.is_floating_point and t.dtype != wd:
+                            t = t.to(wd)
+                    args.append(t)
                 else:
                     args.append(t)
             elif kind == "model_input" and model_inputs is not None:
@@ -510,6 +518,12 @@ class Node:
             src_rank, j, k = self._pos_to_producer[pos]
             if src_rank == self.rank:
                 continue
+            # grads ride the wire at wire_dtype too (parity: reference
+            # compresses both directions, utils.py:184-194; the engine
+            # casts back to the output dtype at the consumer,
+            # compute.py backward g.to(out.dtype))
+            if self.wire_dtype is not None and g.is_floating_point():
+                g = g.to(self.wire_dtype)
             by_rank.setdefault(src_rank, []).append((j * _MAX_OUTS + k, g))
         for r, tensors in by_rank.items():
             self.comm.send(r, "bwd", Message(action=ActionTypes.BACKWARD,

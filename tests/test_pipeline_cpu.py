@@ -91,3 +91,55 @@ def test_three_stage_pipeline_cpu(tmp_path):
     last_epoch = sum(losses[-n_batches:])
     assert last_epoch < first_epoch, \
         f"loss did not decrease: {first_epoch} -> {last_epoch}"
+
+
+def _comp_worker(rank, base_dir, port, out_dir):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    os.environ["GLOO_SOCKET_IFNAME"] = os.environ.get("GLOO_SOCKET_IFNAME", "lo")
+    os.chdir(out_dir)
+    set_seed(42)
+    from ravnest_amd import Node, Trainer
+    node = Node(name=f"node_{rank}", base_dir=base_dir,
+                optimizer=torch.optim.Adam,
+                device=torch.device("cpu"),
+                criterion=_loss_fn,
+                labels=_make_loaders(),
+                update_frequency=1,
+                compression=True)  # bf16 wire for activations AND grads
+    node.start()
+    trainer = Trainer(node=node, train_loader=_make_loaders(),
+                      epochs=3, batch_size=32, inputs_dtype=torch.float32)
+    if rank == 0:
+        trainer.train()
+        node.stop_cluster()
+    else:
+        trainer.prelim_checks()
+    node.stop()
+
+
+def test_pipeline_wire_compression(tmp_path):
+    """bf16 on-the-wire compression for both directions (parity:
+    reference fp16 compression, utils.py:184-194; grads cast back to the
+    output dtype at the consumer)."""
+    set_seed(42)
+    model = CNN()
+    x = torch.randn(2, 1, 8, 8)
+    base = str(tmp_path / "node_data")
+    pool = [NodeSpec(name=f"n{i}", ram=10 * 2**20) for i in range(2)]
+    clusterize(model, (x,), node_pool=pool, max_clusters=1, base_dir=base)
+    port = 29900 + (os.getpid() % 90)
+    ctx = mp.get_context("spawn")
+    procs = [ctx.Process(target=_comp_worker,
+                         args=(r, base, port, str(tmp_path)))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(timeout=180)
+    for p in procs:
+        assert p.exitcode == 0, f"worker exited {p.exitcode}"
+    losses = [float(l) for l in (tmp_path / "losses.txt").read_text().split()]
+    n_batches = 256 // 32
+    assert len(losses) == 3 * n_batches
+    assert sum(losses[-n_batches:]) < sum(losses[:n_batches])

@@ -49,7 +49,10 @@ __global__ void softmax_fwd_kernel(const T* __restrict__ x,
     const float mo = __shfl_xor(m, off, WAVE);
     const float so = __shfl_xor(s, off, WAVE);
     const float mn = fmaxf(m, mo);
-    s = s * __expf(m - mn) + so * __expf(mo - mn);
+    // lanes past a short row carry m = -inf: exp(-inf - -inf) is NaN,
+    // so zero those contributions explicitly
+    s = (m == -INFINITY ? 0.f : s * __expf(m - mn)) +
+        (mo == -INFINITY ? 0.f : so * __expf(mo - mn));
     m = mn;
   }
   const float inv = 1.f / s;

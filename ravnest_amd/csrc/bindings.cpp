@@ -21,6 +21,11 @@ std::vector<at::Tensor> batchnorm_fwd(at::Tensor x, at::Tensor w,
 std::vector<at::Tensor> batchnorm_bwd(at::Tensor dy, at::Tensor x,
                                       at::Tensor w, at::Tensor mean,
                                       at::Tensor rstd);
+std::vector<at::Tensor> mx_quant(at::Tensor x);
+at::Tensor mx_gemm(at::Tensor x, at::Tensor xs, at::Tensor w,
+                   at::Tensor ws);
+at::Tensor mx_scale_probe(at::Tensor a, at::Tensor b, at::Tensor sa,
+                          at::Tensor sb);
 at::Tensor softmax_fwd(at::Tensor x);
 at::Tensor softmax_bwd(at::Tensor dy, at::Tensor y);
 std::vector<at::Tensor> maxpool2d_fwd(at::Tensor x, long K, long S, long P);
@@ -87,6 +92,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("x"), py::arg("p"), py::arg("seed"),
         py::arg("seed_buf") = py::none());
   m.def("dropout_bwd", &dropout_bwd, "dropout bwd");
+  m.def("mx_quant", &mx_quant,
+        "bf16 -> MX fp8 (e4m3 + per-32 e8m0 scales)");
+  m.def("mx_gemm", &mx_gemm, "MX fp8 GEMM: x @ W^T, 32x32x64 scaled MFMA");
+  m.def("mx_scale_probe", &mx_scale_probe,
+        "per-lane scale-byte semantics probe");
   m.def("softmax_fwd", &softmax_fwd, "standalone softmax fwd (last dim)");
   m.def("softmax_bwd", &softmax_bwd, "standalone softmax bwd");
   m.def("maxpool2d_fwd", &maxpool2d_fwd, "MaxPool2d fwd (+argmax idx)");

@@ -11,6 +11,7 @@ from .linear import Linear
 from .conv import Conv1x1
 from .embedding import embedding_ln, embedding_add
 from .pool import MaxPool2d, AvgPool2d, AdaptiveAvgPool2d
+from .mx import MXLinear, mx_linear
 
 __all__ = [
     "get_ext", "has_ext",
@@ -23,4 +24,5 @@ __all__ = [
     "FusedBatchNorm2d", "Linear", "Conv1x1",
     "embedding_ln", "embedding_add",
     "MaxPool2d", "AvgPool2d", "AdaptiveAvgPool2d",
+    "MXLinear", "mx_linear",
 ]

@@ -93,57 +93,93 @@ __global__ void mx_quant_kernel(const bf16_t* __restrict__ x,
 }
 
 // y[M,N] (bf16) = x[M,K]e4m3 @ W[N,K]e4m3^T, both with (.., K/32) e8m0
-// scales. One wave per 32x32 output tile, K streamed from L2 in 64-elem
-// steps (v1 structure: correctness + L2-resident operands; the LDS/
-// 8-phase ladder is the round-2 upgrade).
-__global__ __launch_bounds__(256) void mx_gemm_kernel(
+// scales. One wave per 64x64 output tile (2x2 register tiling of the
+// 32x32 MFMA): 2 A + 2 B fragment loads feed 4 MFMAs per k-step, 2x the
+// arithmetic intensity of a 32x32 tile. K streamed from L2; the LDS/
+// 8-phase ladder is the round-2 upgrade.
+typedef int i32x4v __attribute__((ext_vector_type(4)));
+
+DEVINL i32x8v load_frag_ilv(const unsigned char* p, int hi) {
+  // k-interleaved fragment halves (see header comment): positions
+  // e<16 <- block0 cols [hi*16,+16), e>=16 <- block1 cols [hi*16,+16)
+  i32x8v f;
+  *reinterpret_cast<i32x4v*>(&f) =
+      *reinterpret_cast<const i32x4v*>(p + hi * 16);
+  *(reinterpret_cast<i32x4v*>(&f) + 1) =
+      *reinterpret_cast<const i32x4v*>(p + 32 + hi * 16);
+  return f;
+}
+
+__global__ __launch_bounds__(256, 1) void mx_gemm_kernel(
     const unsigned char* __restrict__ x, const unsigned char* __restrict__ xs,
     const unsigned char* __restrict__ w, const unsigned char* __restrict__ ws,
     bf16_t* __restrict__ y, int M, int N, int K) {
+  constexpr int MI = 4, NJ = 2;  // 128x64 per wave: 6 loads, 8 MFMAs
   const int lane = threadIdx.x & (WAVE - 1);
   const int wid = threadIdx.x / WAVE;
   const int hi = lane >> 5;
   const int j32 = lane & 31;
-  const int tiles_n = (N + 31) / 32;
+  const int tiles_n = (N + NJ * 32 - 1) / (NJ * 32);
   const long tile = (long)blockIdx.x * (blockDim.x / WAVE) + wid;
-  const int m0 = (int)(tile / tiles_n) * 32;
-  const int n0 = (int)(tile % tiles_n) * 32;
+  const int m0 = (int)(tile / tiles_n) * (MI * 32);
+  const int n0 = (int)(tile % tiles_n) * (NJ * 32);
   if (m0 >= M) return;
-  const int kb = K / 32;  // scale blocks per row
-  const int row_a = min(m0 + j32, M - 1);
-  const int row_b = min(n0 + j32, N - 1);
-  const unsigned char* xp = x + (long)row_a * K;
-  const unsigned char* wp = w + (long)row_b * K;
-  const unsigned char* xsp = xs + (long)row_a * kb;
-  const unsigned char* wsp = ws + (long)row_b * kb;
+  const int kb = K / 32;
+  const unsigned char* xp[MI];
+  const unsigned char* wp[NJ];
+  const unsigned char* xsp[MI];
+  const unsigned char* wsp[NJ];
+#pragma unroll
+  for (int t = 0; t < MI; ++t) {
+    const int ra = min(m0 + t * 32 + j32, M - 1);
+    xp[t] = x + (long)ra * K;
+    xsp[t] = xs + (long)ra * kb;
+  }
+#pragma unroll
+  for (int t = 0; t < NJ; ++t) {
+    const int rb = min(n0 + t * 32 + j32, N - 1);
+    wp[t] = w + (long)rb * K;
+    wsp[t] = ws + (long)rb * kb;
+  }
 
-  f32x16v acc;
+  f32x16v acc[MI][NJ];
 #pragma unroll
-  for (int r = 0; r < 16; ++r) acc[r] = 0.f;
-  typedef int i32x4v __attribute__((ext_vector_type(4)));
+  for (int i = 0; i < MI; ++i)
+#pragma unroll
+    for (int j = 0; j < NJ; ++j)
+#pragma unroll
+      for (int r = 0; r < 16; ++r) acc[i][j][r] = 0.f;
+
   for (int k0 = 0; k0 < K; k0 += 64) {
-    // k-interleaved fragment halves (see header comment): positions
-    // e<16 <- block0 cols [hi*16,+16), e>=16 <- block1 cols [hi*16,+16)
-    i32x8v af, bf;
-    *reinterpret_cast<i32x4v*>(&af) =
-        *reinterpret_cast<const i32x4v*>(xp + k0 + hi * 16);
-    *(reinterpret_cast<i32x4v*>(&af) + 1) =
-        *reinterpret_cast<const i32x4v*>(xp + k0 + 32 + hi * 16);
-    *reinterpret_cast<i32x4v*>(&bf) =
-        *reinterpret_cast<const i32x4v*>(wp + k0 + hi * 16);
-    *(reinterpret_cast<i32x4v*>(&bf) + 1) =
-        *reinterpret_cast<const i32x4v*>(wp + k0 + 32 + hi * 16);
-    const int sa = xsp[k0 / 32 + hi];
-    const int sb = wsp[k0 / 32 + hi];
-    acc = __builtin_amdgcn_mfma_scale_f32_32x32x64_f8f6f4(
-        af, bf, acc, 0, 0, 0, sa, 0, sb);
+    i32x8v af[MI], bf[NJ];
+    int sa[MI], sb[NJ];
+#pragma unroll
+    for (int t = 0; t < MI; ++t) {
+      af[t] = load_frag_ilv(xp[t] + k0, hi);
+      sa[t] = xsp[t][k0 / 32 + hi];
+    }
+#pragma unroll
+    for (int t = 0; t < NJ; ++t) {
+      bf[t] = load_frag_ilv(wp[t] + k0, hi);
+      sb[t] = wsp[t][k0 / 32 + hi];
+    }
+#pragma unroll
+    for (int i = 0; i < MI; ++i)
+#pragma unroll
+      for (int j = 0; j < NJ; ++j)
+        acc[i][j] = __builtin_amdgcn_mfma_scale_f32_32x32x64_f8f6f4(
+            af[i], bf[j], acc[i][j], 0, 0, 0, sa[i], 0, sb[j]);
   }
 #pragma unroll
-  for (int r = 0; r < 16; ++r) {
-    const int row = m0 + (r & 3) + 8 * (r >> 2) + 4 * hi;
-    const int col = n0 + j32;
-    if (row < M && col < N) y[(long)row * N + col] = f2bf(acc[r]);
-  }
+  for (int i = 0; i < MI; ++i)
+#pragma unroll
+    for (int j = 0; j < NJ; ++j)
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int row = m0 + i * 32 + (r & 3) + 8 * (r >> 2) + 4 * hi;
+        const int col = n0 + j * 32 + j32;
+        if (row < M && col < N) y[(long)row * N + col] = f2bf(acc[i][j][r]);
+      }
 }
 
 // scale-semantics probe: per-lane scale operands
@@ -208,7 +244,7 @@ at::Tensor mx_gemm(at::Tensor x, at::Tensor xs, at::Tensor w,
               "mx_gemm: K must match and be a multiple of 64");
   auto y = at::empty({(long)M, (long)N},
                      x.options().dtype(at::kBFloat16));
-  const long tiles = (long)((M + 31) / 32) * ((N + 31) / 32);
+  const long tiles = (long)((M + 127) / 128) * ((N + 63) / 64);
   auto stream = c10::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
   dim3 block(256);
   dim3 grid((unsigned)((tiles + 3) / 4));

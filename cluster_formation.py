@@ -20,6 +20,8 @@ if __name__ == "__main__":
     ap.add_argument("--node-config", default="node_data/node_configs.json")
     ap.add_argument("--base-dir", default="node_data")
     ap.add_argument("--max-clusters", type=int, default=5)
+    ap.add_argument("--fp8", action="store_true",
+                    help="sorter: MX fp8 block projections")
     args = ap.parse_args()
 
     if args.model == "cnn":
@@ -34,8 +36,15 @@ if __name__ == "__main__":
     elif args.model == "sorter":
         from examples.sorter.dataset import SortDataset
         ds = SortDataset("train")
-        model = GPT(GPTConfig.nano(vocab_size=ds.vocab_size,
-                                   block_size=ds.block_size))
+        # --fp8 plans the sorter with MX fp8 block projections
+        # (head_dim-64 scale so the attention kernel + MXLinear apply)
+        cfg = (GPTConfig.nano64(vocab_size=ds.vocab_size,
+                                block_size=ds.block_size)
+               if args.fp8 else
+               GPTConfig.nano(vocab_size=ds.vocab_size,
+                              block_size=ds.block_size))
+        cfg.fp8 = args.fp8
+        model = GPT(cfg)
         ex = (ds[0][0].unsqueeze(0),)
     else:  # bert
         cfg = BertConfig.base()

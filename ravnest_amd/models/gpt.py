@@ -12,19 +12,25 @@ from __future__ import annotations
 import torch
 import torch.nn as nn
 
-from ..ops import AttentionCoreQKV, Dropout, FusedLayerNorm, LinearGelu
+from ..ops import (AttentionCoreQKV, Dropout, FusedLayerNorm,
+                   LinearGelu, MXLinear)
 from ..ops import embedding_add
 
 
 class GPTConfig:
     def __init__(self, vocab_size, block_size, n_layer, n_head, n_embd,
-                 dropout=0.1):
+                 dropout=0.1, fp8=False):
         self.vocab_size = vocab_size
         self.block_size = block_size
         self.n_layer = n_layer
         self.n_head = n_head
         self.n_embd = n_embd
         self.dropout = dropout
+        # fp8=True runs the block projections on the MX-scaled fp8 MFMA
+        # path (ops.MXLinear, state-dict compatible with nn.Linear) —
+        # BASELINE.json "GPT-Sorter fp8 CDNA4 MFMA path". Requires
+        # n_embd % 64 == 0.
+        self.fp8 = fp8
 
     @classmethod
     def nano(cls, vocab_size=16, block_size=16):
@@ -67,14 +73,18 @@ class GPTBlock(nn.Module):
         super().__init__()
         self.n_head = cfg.n_head
         self.head_dim = cfg.n_embd // cfg.n_head
+        fp8 = getattr(cfg, "fp8", False)
+        if fp8 and cfg.n_embd % 64 != 0:
+            raise ValueError("fp8 GPT needs n_embd % 64 == 0")
+        lin = MXLinear if fp8 else nn.Linear
         self.ln1 = FusedLayerNorm(cfg.n_embd)
-        self.qkv = nn.Linear(cfg.n_embd, 3 * cfg.n_embd)
+        self.qkv = lin(cfg.n_embd, 3 * cfg.n_embd)
         self.core = AttentionCoreQKV(causal=True)
-        self.proj = nn.Linear(cfg.n_embd, cfg.n_embd)
+        self.proj = lin(cfg.n_embd, cfg.n_embd)
         self.attn_drop = Dropout(cfg.dropout)
         self.ln2 = FusedLayerNorm(cfg.n_embd)
         self.mlp_in = LinearGelu(cfg.n_embd, 4 * cfg.n_embd)
-        self.mlp_out = nn.Linear(4 * cfg.n_embd, cfg.n_embd)
+        self.mlp_out = lin(4 * cfg.n_embd, cfg.n_embd)
         self.mlp_drop = Dropout(cfg.dropout)
 
     def forward(self, x):

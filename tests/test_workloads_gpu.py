@@ -17,9 +17,12 @@ def dev():
     return torch.device("cuda", 0)
 
 
-def test_sorter_trains_and_sorts(dev):
+@pytest.mark.parametrize("fp8", [False, True])
+def test_sorter_trains_and_sorts(dev, fp8):
     """The reference's end-to-end oracle (sorter_inference.py check): a
-    small GPT trained on the sort task must actually sort."""
+    small GPT trained on the sort task must actually sort. fp8=True runs
+    the block projections on the MX-scaled fp8 MFMA path (BASELINE.json
+    "GPT-Sorter fp8 CDNA4 MFMA path")."""
     sys.path.insert(0, ".")
     from examples.sorter.dataset import SortDataset
     from ravnest_amd import set_seed
@@ -30,6 +33,7 @@ def test_sorter_trains_and_sorts(dev):
     cfg = GPTConfig.nano64(vocab_size=ds.vocab_size,
                            block_size=ds.block_size)
     cfg.dropout = 0.0
+    cfg.fp8 = fp8
     model = GPT(cfg).to(dev).to(torch.bfloat16)
     opt = FusedAdam(model.parameters(), lr=5e-4)
     X = torch.stack([ds[i][0] for i in range(len(ds))]).to(dev)

@@ -78,25 +78,31 @@ DEVINL void acc_to_afrag(const f32x16& acc, unsigned int pa[2][4]) {
   }
 }
 
-// Delta[row] = sum_d dO[row][d] * O[row][d]; one wave per row (D<=128)
+// Delta[row] = sum_d dO[row][d] * O[row][d]. D/8 lanes per row, bf16x8
+// loads (the one-wave-per-row 2-B-scalar version measured 7x off the
+// bandwidth bound at 2.5% of the BERT step — rocprofv3 r01 final).
 template <int D>
 __global__ void attn_bwd_delta_kernel(const bf16_t* __restrict__ dout,
                                       const bf16_t* __restrict__ o,
                                       float* __restrict__ delta, int S,
                                       long H, long NR, StridesB sdo) {
+  constexpr int LPR = D / 8;  // lanes per row
   const int lane = threadIdx.x & (WAVE - 1);
-  const long id = (long)blockIdx.x * (blockDim.x / WAVE) +
-                  threadIdx.x / WAVE;
+  const long t = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long id = t / LPR;
   if (id >= NR) return;
   const long bh = id / S;
   const long row = id % S;
-  const long off = (bh / H) * sdo.bs + (bh % H) * sdo.hs + row * sdo.rs;
+  const long off = (bh / H) * sdo.bs + (bh % H) * sdo.hs + row * sdo.rs +
+                   (t % LPR) * 8;
+  bf16x8v a = *reinterpret_cast<const bf16x8v*>(dout + off);
+  bf16x8v b = *reinterpret_cast<const bf16x8v*>(o + off);
   float s = 0.f;
 #pragma unroll
-  for (int d = lane; d < D; d += WAVE)
-    s += bf2f(dout[off + d]) * bf2f(o[off + d]);
-  s = wave_sum(s);
-  if (lane == 0) delta[id] = s;
+  for (int j = 0; j < 8; ++j) s += (float)a[j] * (float)b[j];
+#pragma unroll
+  for (int m = 1; m < LPR; m <<= 1) s += __shfl_xor(s, m, WAVE);
+  if ((lane & (LPR - 1)) == 0) delta[id] = s;
 }
 
 // -------------------------------------------------------------------
@@ -712,8 +718,8 @@ std::vector<at::Tensor> attn_bwd_impl(
   auto stream = c10::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
   const long NR = B * H * (long)S;
   hipLaunchKernelGGL((attn_bwd_delta_kernel<64>),
-                     dim3((NR + 3) / 4), dim3(256), 0, stream, dob, ob,
-                     delta.data_ptr<float>(), S, H, NR, sdo);
+                     dim3((NR * 8 + 255) / 256), dim3(256), 0, stream, dob,
+                     ob, delta.data_ptr<float>(), S, H, NR, sdo);
   dim3 block(256);
   dim3 gridk((S + 127) / 128, B * H);
   // split is the measured default: the merged kernel lands at 210 VGPRs

@@ -50,6 +50,13 @@ std::vector<at::Tensor> ce_fwd(at::Tensor logits, at::Tensor targets,
                                int64_t ignore_index);
 at::Tensor ce_bwd(at::Tensor logits, at::Tensor targets, at::Tensor lse,
                   at::Tensor gscale, int64_t ignore_index);
+std::tuple<at::Tensor, long, long> adam_build_table(
+    std::vector<at::Tensor> ps, std::vector<at::Tensor> gs,
+    std::vector<at::Tensor> ms, std::vector<at::Tensor> vs,
+    std::vector<at::Tensor> masters);
+void fused_adam_graph(at::Tensor table, long nchunks, long esize,
+                      at::Tensor lr_buf, double b1, double b2, double eps,
+                      double wd, at::Tensor step_buf);
 void fused_adam(std::vector<at::Tensor> ps, std::vector<at::Tensor> gs,
                 std::vector<at::Tensor> ms, std::vector<at::Tensor> vs,
                 std::vector<at::Tensor> masters, double lr, double b1,
@@ -114,6 +121,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("ce_fwd", &ce_fwd, "fused softmax cross-entropy fwd");
   m.def("ce_bwd", &ce_bwd, "fused softmax cross-entropy bwd");
   m.def("fused_adam", &fused_adam, "multi-tensor Adam");
+  m.def("adam_build_table", &adam_build_table,
+        "prebuild the device chunk table (graph-capturable Adam)");
+  m.def("fused_adam_graph", &fused_adam_graph,
+        "multi-tensor Adam with device lr/step buffers (hipGraph mode)");
   m.def("fused_sgd", &fused_sgd, "multi-tensor SGD+momentum");
   m.def("fused_lamb", &fused_lamb, "multi-tensor LAMB");
   m.def("fused_copy", &fused_copy, "multi-tensor device copy");

@@ -63,7 +63,18 @@ DEVINL void adam_one(float& p, float g, float& m, float& v, float lr,
 template <typename T>
 __global__ void adam_mt_kernel(const Chunk* __restrict__ chunks, float lr,
                                float b1, float b2, float eps, float wd,
-                               float bc1, float bc2) {
+                               float bc1, float bc2,
+                               const float* __restrict__ lr_buf,
+                               const long long* __restrict__ step_buf) {
+  // hipGraph-capturable mode: lr and the bias-correction step come from
+  // DEVICE buffers so replays see live values (host scalars would
+  // freeze at capture — engine/graphstep.py)
+  if (lr_buf) {
+    lr = *lr_buf;
+    const float st = (float)*step_buf;
+    bc1 = 1.f - __powf(b1, st);
+    bc2 = 1.f - __powf(b2, st);
+  }
   const Chunk c = chunks[blockIdx.x];
   const int n4 = c.n >> 2;
   float4* m4 = reinterpret_cast<float4*>(c.m);
@@ -278,13 +289,52 @@ void fused_adam(std::vector<at::Tensor> ps, std::vector<at::Tensor> gs,
                        stream,
                        reinterpret_cast<const Chunk*>(dev_chunks.data_ptr()),
                        (float)lr, (float)b1, (float)b2, (float)eps, (float)wd,
-                       (float)bc1, (float)bc2);
+                       (float)bc1, (float)bc2, nullptr, nullptr);
   else
     hipLaunchKernelGGL((adam_mt_kernel<float>), dim3(nchunks), dim3(256), 0,
                        stream,
                        reinterpret_cast<const Chunk*>(dev_chunks.data_ptr()),
                        (float)lr, (float)b1, (float)b2, (float)eps, (float)wd,
-                       (float)bc1, (float)bc2);
+                       (float)bc1, (float)bc2, nullptr, nullptr);
+  HIP_CHECK_LAST();
+}
+
+// graph-capturable Adam: the chunk table is prebuilt ONCE (pointers are
+// stable inside a captured step: params/state outside the graph pool,
+// grads at fixed pool addresses) and lr/step live in device buffers.
+std::tuple<at::Tensor, long, long> adam_build_table(
+    std::vector<at::Tensor> ps, std::vector<at::Tensor> gs,
+    std::vector<at::Tensor> ms, std::vector<at::Tensor> vs,
+    std::vector<at::Tensor> masters) {
+  TORCH_CHECK(!ps.empty());
+  int nchunks = 0, esize = 4;
+  auto table = build_chunks(ps, gs, ms, vs, masters, nchunks, esize);
+  return {table, (long)nchunks, (long)esize};
+}
+
+void fused_adam_graph(at::Tensor table, long nchunks, long esize,
+                      at::Tensor lr_buf, double b1, double b2, double eps,
+                      double wd, at::Tensor step_buf) {
+  TORCH_CHECK(table.is_cuda() && lr_buf.is_cuda() && step_buf.is_cuda());
+  TORCH_CHECK(lr_buf.scalar_type() == at::kFloat &&
+              step_buf.scalar_type() == at::kLong);
+  auto stream = c10::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
+  if (esize == 2)
+    hipLaunchKernelGGL((adam_mt_kernel<bf16_t>), dim3((int)nchunks),
+                       dim3(256), 0, stream,
+                       reinterpret_cast<const Chunk*>(table.data_ptr()), 0.f,
+                       (float)b1, (float)b2, (float)eps, (float)wd, 1.f, 1.f,
+                       lr_buf.data_ptr<float>(),
+                       reinterpret_cast<const long long*>(
+                           step_buf.data_ptr<int64_t>()));
+  else
+    hipLaunchKernelGGL((adam_mt_kernel<float>), dim3((int)nchunks),
+                       dim3(256), 0, stream,
+                       reinterpret_cast<const Chunk*>(table.data_ptr()), 0.f,
+                       (float)b1, (float)b2, (float)eps, (float)wd, 1.f, 1.f,
+                       lr_buf.data_ptr<float>(),
+                       reinterpret_cast<const long long*>(
+                           step_buf.data_ptr<int64_t>()));
   HIP_CHECK_LAST();
 }
 

@@ -69,7 +69,8 @@ class ComputeEngine:
                 and self.update_frequency == 1 and criterion is not None
                 and os.environ.get("RAVNEST_CUDA_GRAPH", "1") == "1"):
             from .graphstep import GraphedTrainStep
-            self._graph_step = GraphedTrainStep(model, criterion, device)
+            self._graph_step = GraphedTrainStep(model, criterion, device,
+                                                optimizer=optimizer)
 
         self.current_version = 0
         self.version_to_param: dict[int, list[torch.Tensor]] = {}
@@ -332,10 +333,15 @@ class ComputeEngine:
         param.grad keeps referencing — see graphstep.py)."""
         # ENQUEUE the optimizer before reading the loss: float(loss_t)
         # synchronizes on the replay, and the optimizer's host-side chunk
-        # building would otherwise run on an idle GPU
+        # building would otherwise run on an idle GPU. When the optimizer
+        # was captured INTO the graph the replay already applied it; only
+        # the host-side step mirror needs advancing (checkpoints).
         self.n_backwards += 1
         stepped = False
-        if self.optimizer is not None:
+        if self._graph_step.opt_captured:
+            self.optimizer.bump_host_steps()
+            stepped = True
+        elif self.optimizer is not None:
             self.optimizer.step()
             stepped = True
         loss_val = float(loss_t)

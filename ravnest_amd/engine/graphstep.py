@@ -40,10 +40,23 @@ class GraphedTrainStep:
 
     WARMUP = 3
 
-    def __init__(self, model, criterion, device):
+    def __init__(self, model, criterion, device, optimizer=None):
         self.model = model
         self.criterion = criterion
         self.device = device
+        # optimizer captured INTO the graph when it supports device-side
+        # lr/step buffers (FusedAdam.graph_step). OFF by default: the
+        # capture needs pre-existing grad buffers, which flips autograd
+        # from assign-into-pool to accumulate (temp + add per weight
+        # grad) — measured 1457 vs 1561 samples/s on BERT-base, so the
+        # eager one-kernel optimizer wins. RAVNEST_GRAPH_OPT=1 enables
+        # full-step capture (single launch incl. update) for launch-
+        # latency-dominated regimes (small models / many tiny steps).
+        import os
+        use_opt = os.environ.get("RAVNEST_GRAPH_OPT", "0") == "1"
+        self.optimizer = (optimizer if use_opt and
+                          hasattr(optimizer, "graph_step") else None)
+        self.opt_captured = False
         self.graphs = {}
         self.pool = None
         self.failed = False
@@ -75,6 +88,8 @@ class GraphedTrainStep:
                 return None
             self.graphs[key] = entry
         graph, s_args, s_tgts, s_loss, grad_refs = entry
+        if self.opt_captured:
+            self.optimizer.sync_lr()  # outside the graph: live schedules
         for dst, src in zip(s_args, arg_ts):
             dst.copy_(src, non_blocking=True)
         for dst, src in zip(s_tgts, tgt_ts):
@@ -94,11 +109,20 @@ class GraphedTrainStep:
         self.model.train()
         rng.device_seed_counter(self.device)  # materialize pre-capture
 
-        def step():
+        if self.optimizer is not None and not self.opt_captured:
+            self.opt_captured = self.optimizer.enable_graph_capture(
+                self.device)
+
+        def step(with_opt=False, stable_grads=None):
             rng.bump_seed_counter(self.device)
+            if with_opt:
+                # stable (non-pool) grad buffers: zero, accumulate, apply
+                torch._foreach_zero_(stable_grads)
             out = self.model(*s_args)
             loss = self.criterion(out, s_tgts[0])
             loss.backward()
+            if with_opt:
+                self.optimizer.graph_step()
             return loss
 
         # warmup on a side stream (per the torch.cuda.graph recipe):
@@ -109,18 +133,26 @@ class GraphedTrainStep:
             for _ in range(self.WARMUP):
                 step()
         torch.cuda.current_stream(self.device).wait_stream(side)
-        # capture must record grad ASSIGNMENT (not accumulation): replays
-        # then overwrite the same pool buffers every step
+        # opt-captured mode: stable pre-allocated grad buffers (zeroed
+        # and accumulated INSIDE the graph); else grads stay None so
+        # capture records plain assignments into pool buffers
         self.model.zero_grad(set_to_none=True)
+        stable_grads = None
+        if self.opt_captured:
+            # built ONCE: a second shape-signature capture must reuse the
+            # same stable buffers (earlier graphs recorded their pointers)
+            stable_grads = self.optimizer._graph.get("grads")
+            if stable_grads is None:
+                stable_grads = self.optimizer.build_graph_table()
 
         graph = torch.cuda.CUDAGraph()
         if self.pool is None:
             with torch.cuda.graph(graph):
-                s_loss = step()
+                s_loss = step(self.opt_captured, stable_grads)
             self.pool = graph.pool()
         else:
             with torch.cuda.graph(graph, pool=self.pool):
-                s_loss = step()
+                s_loss = step(self.opt_captured, stable_grads)
         grad_refs = [(p, p.grad) for p in self.model.parameters()
                      if p.grad is not None]
         return graph, s_args, s_tgts, s_loss, grad_refs

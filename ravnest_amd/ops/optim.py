@@ -25,6 +25,91 @@ class FusedAdam(Optimizer):
         defaults = dict(lr=lr, betas=betas, eps=eps,
                         weight_decay=weight_decay)
         super().__init__(params, defaults)
+        self._graph = None  # hipGraph-capture buffers (see graph_step)
+
+    # ---- hipGraph capture support (engine/graphstep.py) --------------
+    # lr and the bias-correction step live in DEVICE buffers so graph
+    # replays see live values; the chunk table is built once at capture
+    # (param/grad/state pointers are stable across replays).
+    def enable_graph_capture(self, device) -> bool:
+        if len(self.param_groups) != 1:
+            return False
+        self._ensure_state()
+        start = 0
+        for st in self.state.values():
+            if "step" in st:
+                start = max(start, st["step"])
+        self._graph = {
+            "lr": torch.zeros(1, dtype=torch.float32, device=device),
+            "step": torch.full((1,), start, dtype=torch.int64,
+                               device=device),
+            "table": None,
+        }
+        self.sync_lr()
+        return True
+
+    def _ensure_state(self):
+        for group in self.param_groups:
+            for p in group["params"]:
+                st = self.state[p]
+                if len(st) == 0:
+                    st["step"] = 0
+                    st["m"] = torch.zeros_like(p, dtype=torch.float32)
+                    st["v"] = torch.zeros_like(p, dtype=torch.float32)
+                    if p.dtype == torch.bfloat16 and p.is_cuda:
+                        st["master"] = p.detach().float().clone()
+
+    def sync_lr(self):
+        """Host -> device lr refresh; call OUTSIDE the graph, before each
+        replay (lets lr schedules work under capture)."""
+        self._graph["lr"].fill_(float(self.param_groups[0]["lr"]))
+
+    def bump_host_steps(self):
+        """Mirror the device step counter into the python state (kept
+        consistent for checkpoints); call once per replayed step."""
+        for st in self.state.values():
+            if "step" in st:
+                st["step"] += 1
+
+    @torch.no_grad()
+    def build_graph_table(self):
+        """Allocate stable grad buffers and prebuild the device chunk
+        table. Must run OUTSIDE capture (the table upload is a blocking
+        H2D copy); the captured step then zeroes these buffers, lets
+        backward ACCUMULATE into them, and applies the one-kernel update.
+        Returns the stable grad list (for the capture's zero pass)."""
+        ext = get_ext(required=True)
+        group = self.param_groups[0]
+        grads = []
+        ps = []
+        for p in group["params"]:
+            if not p.requires_grad:
+                continue
+            p.grad = torch.zeros_like(p)
+            ps.append(p)
+            grads.append(p.grad)
+        ms = [self.state[p]["m"] for p in ps]
+        vs = [self.state[p]["v"] for p in ps]
+        masters = ([self.state[p]["master"] for p in ps]
+                   if ps and ps[0].dtype == torch.bfloat16 else [])
+        self._graph["table"] = ext.adam_build_table(ps, grads, ms, vs,
+                                                    masters)
+        self._graph["grads"] = grads
+        return grads
+
+    @torch.no_grad()
+    def graph_step(self):
+        """The capturable step body: device step bump + ONE kernel.
+        Must run inside the captured region, after backward."""
+        ext = get_ext(required=True)
+        g = self._graph
+        g["step"].add_(1)
+        group = self.param_groups[0]
+        b1, b2 = group["betas"]
+        table, nchunks, esize = g["table"]
+        ext.fused_adam_graph(table, nchunks, esize, g["lr"], b1, b2,
+                             group["eps"], group["weight_decay"],
+                             g["step"])
 
     @torch.no_grad()
     def step(self, closure=None):

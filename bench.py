@@ -65,6 +65,106 @@ class _ListLoader:
         return iter(self.items)
 
 
+def run_pp(args, rank, world, device, on_gpu, family, model, batches,
+           labels):
+    """Async pipeline-parallel bench: ONE model split into `world` stages
+    (1 cluster x N stages — the reference's headline topology). Rank 0
+    plans with clusterize() and drives microbatch injection; every stage
+    reports the barrier-bracketed timed window (MAX over ranks)."""
+    import time as _time
+    from ravnest_amd import clusterize
+    from ravnest_amd.planner import NodeSpec
+
+    base = os.path.abspath(f"bench_pp_plan_w{world}")
+    ready = os.path.join(base, ".ready")
+    if rank == 0:
+        pool = [NodeSpec(name=f"n{i}", ram=8 * 2**30) for i in range(world)]
+        plan_model = build_model(args.model, args.seq)[0]  # fp32 for fx
+        ex = batches[0]
+        ex_args = tuple(v.cpu() for v in ex.values())
+        clusterize(plan_model, ex_args, node_pool=pool, max_clusters=1,
+                   base_dir=base)
+        with open(ready, "w") as f:
+            f.write("ok")
+    else:
+        deadline = time.time() + 600
+        me = os.path.join(base, "nodes", f"node_{rank}.json")
+        while not (os.path.exists(ready) and os.path.exists(me)):
+            if time.time() > deadline:
+                raise TimeoutError("pp plan did not appear")
+            time.sleep(0.5)
+
+    transform = ((lambda m: m.to(torch.bfloat16))
+                 if args.dtype == "bf16" and on_gpu else None)
+    from ravnest_amd.ops import CrossEntropyLoss, FusedAdam
+    node = Node(name=f"node_{rank}", base_dir=base,
+                optimizer=FusedAdam if on_gpu else torch.optim.Adam,
+                optimizer_params={"lr": 1e-4},
+                device=device,
+                criterion=CrossEntropyLoss(-100),
+                labels=_ListLoader([labels[i % len(labels)] for i in
+                                    range(args.warmup + args.steps +
+                                          len(labels))]),
+                update_frequency=1,
+                model_transform=transform,
+                loss_filename="bench_pp_losses.txt")
+    node.start()
+    import torch.distributed as dist
+
+    def sync():
+        if on_gpu:
+            torch.cuda.synchronize(device)
+
+    is_root = node.node_type.value == "root"
+    dist.barrier()
+    if is_root:
+        for i in range(args.warmup):
+            node.forward_compute(**batches[i % len(batches)])
+        node.wait_for_backwards()
+    sync()
+    dist.barrier()
+    sync()
+    t0 = _time.perf_counter()
+    if is_root:
+        for i in range(args.steps):
+            node.forward_compute(**batches[i % len(batches)])
+        node.wait_for_backwards()
+    sync()
+    dist.barrier()
+    elapsed = _time.perf_counter() - t0
+    t = torch.tensor([elapsed], dtype=torch.float64,
+                     device=device if on_gpu else "cpu")
+    dist.all_reduce(t, op=dist.ReduceOp.MAX)
+    elapsed = float(t.item())
+
+    if rank == 0:
+        samples_per_sec = args.micro_batch * args.steps / elapsed
+        print(json.dumps({
+            "metric": "samples/sec (whole node), "
+                      f"{args.model} async pipeline",
+            "value": round(samples_per_sec, 2),
+            "unit": "samples/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(elapsed / args.steps * 1000.0, 2),
+            "higher_is_better": True,
+            "scaling": "strong",
+            "vs_baseline": None,
+            "dtype": args.dtype if on_gpu else "fp32",
+            "data": "synthetic",
+            "config": {"model": args.model,
+                       "global_batch": args.micro_batch,
+                       "seq_len": args.seq,
+                       "parallelism": f"pp{world}",
+                       "optimizer": "fused_adam" if on_gpu else "adam"},
+        }), flush=True)
+        node.stop_cluster()
+    node.stop()
+    sys.stdout.flush()
+    os._exit(0)
+
+
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
@@ -75,6 +175,10 @@ def main():
     ap.add_argument("--seq", type=int, default=512)
     ap.add_argument("--reduce-factor", type=int, default=4,
                     help="average DP replicas every N steps")
+    ap.add_argument("--parallelism", default="dp", choices=["dp", "pp"],
+                    help="dp: one fused replica per GPU (driver default); "
+                         "pp: ONE async pipeline of N stages across the "
+                         "N ranks (the reference's headline topology)")
     ap.add_argument("--dtype", default=None, choices=["bf16", "fp32"])
     ap.add_argument("--cpu", action="store_true",
                     help="CPU sanity mode (tiny model)")
@@ -149,6 +253,11 @@ def main():
             y = torch.randint(0, 200, (args.micro_batch,), device=device)
             batches.append({"x": X})
             labels.append(y)
+
+    if args.parallelism == "pp" and world > 1:
+        run_pp(args, rank, world, device, on_gpu, family, model,
+               batches, labels)
+        return
 
     comm = None
     if world > 1:

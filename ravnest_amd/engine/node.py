@@ -91,7 +91,8 @@ class Node:
                  comm: CommBackend | None = None,
                  compression: bool = False,
                  wire_dtype: torch.dtype | None = None,
-                 amp_dtype: torch.dtype | None = None):
+                 amp_dtype: torch.dtype | None = None,
+                 model_transform=None):
         # parity: the reference's fp16 wire compression (utils.py:184-194)
         # becomes an optional on-the-wire cast; bf16 is the natural MI355X
         # wire dtype (no clamping needed)
@@ -129,6 +130,10 @@ class Node:
         if model is None:
             model = torch.load(Path(self.template_path) / "submod.pt",
                                map_location="cpu", weights_only=False)
+        if model_transform is not None:
+            # applied BEFORE optimizer/engine init (e.g. .to(bfloat16)
+            # for bf16-native pipeline stages — bench.py --parallelism pp)
+            model = model_transform(model)
         self.model = model.to(self.device)
         if input_template is None and self.template_path:
             with open(Path(self.template_path) / "inputs.json") as f:

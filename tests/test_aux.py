@@ -2,7 +2,6 @@
 injection, latest-weights pull."""
 import os
 
-import numpy as np
 import torch
 import torch.multiprocessing as mp
 

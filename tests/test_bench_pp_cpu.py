@@ -7,8 +7,6 @@ import os
 import subprocess
 import sys
 
-import pytest
-
 
 def test_bench_pp_two_stage(tmp_path):
     env = dict(os.environ)

@@ -6,7 +6,6 @@ routing), CE leaf loss. The hard-parts checklist of SURVEY.md section 7
 import os
 
 import numpy as np
-import pytest
 import torch
 import torch.multiprocessing as mp
 

@@ -8,7 +8,6 @@ tuples (dp_segments), every segment group spans both clusters, and
 after training + a final reduce both replicas hold identical parameters
 while the loss decreased.
 """
-import json
 import os
 
 import numpy as np

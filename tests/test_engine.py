@@ -2,7 +2,6 @@
 accumulation, version GC (the reference's signature mechanism,
 SURVEY.md section 2.4)."""
 import torch
-import pytest
 
 from ravnest_amd import set_seed
 from ravnest_amd.engine.compute import ComputeEngine

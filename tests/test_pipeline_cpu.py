@@ -5,11 +5,9 @@ This is the reference's canonical walkthrough workload (CNN on synthetic
 the MI355X framework's comm stack with the gloo backend standing in for
 RCCL. Asserts: training runs, loss decreases, shutdown is clean.
 """
-import json
 import os
 
 import numpy as np
-import pytest
 import torch
 import torch.multiprocessing as mp
 

@@ -5,7 +5,6 @@ evaluate/val_accuracy path on the CNN walkthrough."""
 import os
 
 import numpy as np
-import pytest
 import torch
 import torch.multiprocessing as mp
 

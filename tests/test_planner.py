@@ -3,7 +3,6 @@
 Mirrors the parity surface of reference operations/ (SURVEY.md section 3.1).
 """
 import json
-from pathlib import Path
 
 import pytest
 import torch

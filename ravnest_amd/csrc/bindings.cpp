@@ -14,6 +14,8 @@ std::vector<at::Tensor> layernorm_add_fwd(at::Tensor a, at::Tensor b,
 at::Tensor bias_gelu_fwd(at::Tensor x, at::Tensor bias);
 at::Tensor bias_gelu_bwd(at::Tensor dy, at::Tensor x, at::Tensor bias);
 at::Tensor colsum_bf16(at::Tensor x);
+std::vector<at::Tensor> bias_gelu_bwd_db(at::Tensor dy, at::Tensor x,
+                                         at::Tensor bias);
 std::vector<at::Tensor> batchnorm_fwd(at::Tensor x, at::Tensor w,
                                       at::Tensor b, at::Tensor running_mean,
                                       at::Tensor running_var, bool training,
@@ -93,6 +95,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bias_gelu_fwd", &bias_gelu_fwd, "fused bias+GELU fwd");
   m.def("bias_gelu_bwd", &bias_gelu_bwd, "fused bias+GELU bwd");
   m.def("colsum_bf16", &colsum_bf16, "column sum bf16 -> fp32");
+  m.def("bias_gelu_bwd_db", &bias_gelu_bwd_db,
+        "fused bias+GELU bwd with in-pass bias-grad column reduction");
   m.def("batchnorm_fwd", &batchnorm_fwd, "fused BatchNorm2d fwd (NCHW)");
   m.def("batchnorm_bwd", &batchnorm_bwd, "fused BatchNorm2d bwd (NCHW)");
   m.def("dropout_fwd", &dropout_fwd, "philox dropout fwd (replayable)",

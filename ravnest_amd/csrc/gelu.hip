@@ -121,6 +121,48 @@ __global__ void colsum_kernel(const bf16_t* __restrict__ x,
   atomicAdd(&out[col], s);
 }
 
+// fused bias_gelu backward + bias-grad column reduction for bf16:
+// dx = dy * gelu'(x+b) written AND column-summed in the same pass (the
+// separate colsum re-read of dx measured 1.5% of the BERT step).
+// Column-walk layout (colsum recipe): thread owns a column, 8 rows of
+// both streams in flight, one atomicAdd per (column, chunk).
+template <typename PT>
+__global__ void bias_gelu_bwd_db_kernel(const bf16_t* __restrict__ dy,
+                                        const bf16_t* __restrict__ x,
+                                        const PT* __restrict__ bias,
+                                        bf16_t* __restrict__ dx,
+                                        float* __restrict__ db, int H,
+                                        long N, long rows_per_chunk) {
+  const int col = blockIdx.x * blockDim.x + threadIdx.x;
+  const int chunk = blockIdx.y;
+  if (col >= H) return;
+  const float b = load_pt(bias + col);
+  const long r0 = (long)chunk * rows_per_chunk;
+  const long r1 = min(N, r0 + rows_per_chunk);
+  float s = 0.f;
+  long r = r0;
+  for (; r + 8 <= r1; r += 8) {
+    float xv[8], gv[8];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      xv[j] = bf2f(x[(r + j) * H + col]);
+      gv[j] = bf2f(dy[(r + j) * H + col]);
+    }
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const float d = gv[j] * gelu_grad_f(xv[j] + b);
+      dx[(r + j) * H + col] = f2bf(d);
+      s += d;
+    }
+  }
+  for (; r < r1; ++r) {
+    const float d = bf2f(dy[r * H + col]) * gelu_grad_f(bf2f(x[r * H + col]) + b);
+    dx[r * H + col] = f2bf(d);
+    s += d;
+  }
+  atomicAdd(&db[col], s);
+}
+
 }  // namespace
 
 static int grid_for(long work, int block) {
@@ -185,6 +227,42 @@ at::Tensor bias_gelu_bwd(at::Tensor dy, at::Tensor x, at::Tensor bias) {
   return dx;
 }
 
+
+std::vector<at::Tensor> bias_gelu_bwd_db(at::Tensor dy, at::Tensor x,
+                                         at::Tensor bias) {
+  TORCH_CHECK(dy.is_cuda() && dy.is_contiguous() && x.is_contiguous());
+  TORCH_CHECK(dy.scalar_type() == at::kBFloat16 &&
+              x.scalar_type() == at::kBFloat16);
+  const int H = (int)bias.numel();
+  TORCH_CHECK(x.size(-1) == H);
+  const long N = x.numel() / H;
+  auto dx = at::empty_like(dy);
+  auto db = at::zeros({H}, x.options().dtype(at::kFloat));
+  auto stream = c10::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
+  const int col_blocks = (H + 255) / 256;
+  int nchunks = (int)std::min<long>((N + 63) / 64,
+                                    std::max(1L, (long)(768 / col_blocks)));
+  const long rows_per_chunk = (N + nchunks - 1) / nchunks;
+  dim3 grid(col_blocks, nchunks);
+  if (bias.scalar_type() == at::kFloat)
+    hipLaunchKernelGGL((bias_gelu_bwd_db_kernel<float>), grid, dim3(256), 0,
+                       stream,
+                       reinterpret_cast<const bf16_t*>(dy.data_ptr()),
+                       reinterpret_cast<const bf16_t*>(x.data_ptr()),
+                       bias.data_ptr<float>(),
+                       reinterpret_cast<bf16_t*>(dx.data_ptr()),
+                       db.data_ptr<float>(), H, N, rows_per_chunk);
+  else
+    hipLaunchKernelGGL((bias_gelu_bwd_db_kernel<bf16_t>), grid, dim3(256), 0,
+                       stream,
+                       reinterpret_cast<const bf16_t*>(dy.data_ptr()),
+                       reinterpret_cast<const bf16_t*>(x.data_ptr()),
+                       reinterpret_cast<const bf16_t*>(bias.data_ptr()),
+                       reinterpret_cast<bf16_t*>(dx.data_ptr()),
+                       db.data_ptr<float>(), H, N, rows_per_chunk);
+  HIP_CHECK_LAST();
+  return {dx, db};
+}
 
 at::Tensor colsum_bf16(at::Tensor x) {
   TORCH_CHECK(x.is_cuda() && x.is_contiguous());

@@ -33,12 +33,12 @@ class _BiasGeluFn(torch.autograd.Function):
     def backward(ctx, dy):
         x, bias = ctx.saved_tensors
         ext = get_ext(required=True)
+        if x.dtype == torch.bfloat16:
+            # single pass: dx and the bias-grad column sum together
+            dx, db = ext.bias_gelu_bwd_db(dy.contiguous(), x, bias)
+            return dx, db.to(bias.dtype)
         dx = ext.bias_gelu_bwd(dy.contiguous(), x, bias)
-        # bias grad: fused bf16 column sum (no fp32 materialization)
-        if dx.dtype == torch.bfloat16:
-            db = ext.colsum_bf16(dx.reshape(-1, dx.shape[-1])).to(bias.dtype)
-        else:
-            db = dx.float().reshape(-1, dx.shape[-1]).sum(0).to(bias.dtype)
+        db = dx.float().reshape(-1, dx.shape[-1]).sum(0).to(bias.dtype)
         return dx, db
 
 

@@ -23,6 +23,13 @@ std::vector<at::Tensor> batchnorm_fwd(at::Tensor x, at::Tensor w,
 std::vector<at::Tensor> batchnorm_bwd(at::Tensor dy, at::Tensor x,
                                       at::Tensor w, at::Tensor mean,
                                       at::Tensor rstd);
+at::Tensor conv2d_fwd(at::Tensor x, at::Tensor w,
+                      c10::optional<at::Tensor> bias, long st, long pad,
+                      bool relu);
+at::Tensor conv2d_dgrad(at::Tensor dy, at::Tensor w, long N, long H,
+                        long W, long st, long pad);
+at::Tensor conv2d_wgrad(at::Tensor dy, at::Tensor x, long R, long S,
+                        long st, long pad);
 std::vector<at::Tensor> mx_quant(at::Tensor x);
 at::Tensor mx_gemm(at::Tensor x, at::Tensor xs, at::Tensor w,
                    at::Tensor ws);
@@ -103,6 +110,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("x"), py::arg("p"), py::arg("seed"),
         py::arg("seed_buf") = py::none());
   m.def("dropout_bwd", &dropout_bwd, "dropout bwd");
+  m.def("conv2d_fwd", &conv2d_fwd,
+        "implicit-GEMM MFMA conv fwd (bf16 NCHW)", py::arg("x"),
+        py::arg("w"), py::arg("bias") = py::none(), py::arg("st") = 1,
+        py::arg("pad") = 0, py::arg("relu") = false);
+  m.def("conv2d_dgrad", &conv2d_dgrad, "implicit-GEMM MFMA conv dgrad");
+  m.def("conv2d_wgrad", &conv2d_wgrad, "implicit-GEMM MFMA conv wgrad");
   m.def("mx_quant", &mx_quant,
         "bf16 -> MX fp8 (e4m3 + per-32 e8m0 scales)");
   m.def("mx_gemm", &mx_gemm, "MX fp8 GEMM: x @ W^T, 32x32x64 scaled MFMA");

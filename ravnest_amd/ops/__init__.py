@@ -8,7 +8,7 @@ from .losses import CrossEntropyLoss, cross_entropy
 from .optim import FusedAdam, FusedSGD, FusedLAMB
 from .batchnorm import FusedBatchNorm2d
 from .linear import Linear
-from .conv import Conv1x1
+from .conv import Conv1x1, conv2d_mfma
 from .embedding import embedding_ln, embedding_add
 from .pool import MaxPool2d, AvgPool2d, AdaptiveAvgPool2d
 from .mx import MXLinear, mx_linear
@@ -21,7 +21,7 @@ __all__ = [
     "AttentionCore", "AttentionCoreQKV", "attention", "attention_qkv",
     "CrossEntropyLoss", "cross_entropy",
     "FusedAdam", "FusedSGD", "FusedLAMB",
-    "FusedBatchNorm2d", "Linear", "Conv1x1",
+    "FusedBatchNorm2d", "Linear", "Conv1x1", "conv2d_mfma",
     "embedding_ln", "embedding_add",
     "MaxPool2d", "AvgPool2d", "AdaptiveAvgPool2d",
     "MXLinear", "mx_linear",

@@ -612,3 +612,41 @@ def test_mx_linear_trains(dev):
     # fp8 forward noise bounds the floor; memorization must still cut
     # the loss several-fold
     assert loss.item() < first * 0.3, (first, loss.item())
+
+
+@pytest.mark.parametrize("geom", [
+    # (Ci, Co, H, W, R, stride, pad)
+    (8, 16, 14, 14, 3, 1, 1),
+    (8, 16, 15, 15, 3, 2, 1),
+    (4, 8, 9, 9, 5, 1, 2),
+    (8, 16, 8, 8, 1, 1, 0),
+])
+def test_conv2d_mfma(dev, geom):
+    """Implicit-GEMM MFMA conv fwd/dgrad/wgrad vs fp32 torch conv."""
+    from ravnest_amd.ops import conv2d_mfma
+    Ci, Co, H, W, R, st, pad = geom
+    torch.manual_seed(12)
+    N = 3
+    x = torch.randn(N, Ci, H, W, device=dev).to(torch.bfloat16)
+    w = (torch.randn(Co, Ci, R, R, device=dev) / (Ci * R)).to(torch.bfloat16)
+    b = torch.randn(Co, device=dev).to(torch.bfloat16)
+    x1 = x.clone().requires_grad_()
+    w1 = w.clone().requires_grad_()
+    b1 = b.clone().requires_grad_()
+    y = conv2d_mfma(x1, w1, b1, stride=st, padding=pad)
+    dy = torch.randn_like(y)
+    y.backward(dy)
+
+    x2 = x.detach().float().requires_grad_()
+    w2 = w.detach().float().requires_grad_()
+    b2 = b.detach().float().requires_grad_()
+    y2 = torch.nn.functional.conv2d(x2, w2, b2, st, pad)
+    y2.backward(dy.float())
+
+    assert torch.allclose(y.float(), y2, atol=8e-2, rtol=5e-2), \
+        (y.float() - y2).abs().max()
+    assert torch.allclose(x1.grad.float(), x2.grad, atol=1e-1, rtol=8e-2), \
+        (x1.grad.float() - x2.grad).abs().max()
+    assert torch.allclose(w1.grad.float(), w2.grad, atol=2e-1, rtol=8e-2), \
+        (w1.grad.float() - w2.grad).abs().max()
+    assert torch.allclose(b1.grad.float(), b2.grad, atol=2e-1, rtol=5e-2)

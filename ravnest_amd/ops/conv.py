@@ -4,13 +4,16 @@ A 1x1 stride-1 convolution IS a GEMM over (C_in -> C_out) per pixel —
 MIOpen still runs it through per-image im2col + GEMM (profile r01:
 thousands of Im2d2Col launches per ResNet step). Conv1x1 routes it
 straight through torch.matmul (hipBLASLt), state-dict compatible with
-nn.Conv2d(k=1). Spatial convolutions stay on MIOpen (the library path;
-hand-written implicit-GEMM conv kernels are round-2 work).
+nn.Conv2d(k=1). Spatial convolutions: fp32 stays on MIOpen (measured
+faster); bf16 has the hand-written implicit-GEMM MFMA path
+(conv2d_mfma, csrc/conv.hip).
 """
 from __future__ import annotations
 
 import torch
 import torch.nn as nn
+
+from ._ext import get_ext
 
 
 class Conv1x1(nn.Module):

@@ -124,3 +124,25 @@ def test_health_monitor_pipeline(tmp_path):
         p.join(timeout=60)
     # middle node pings both neighbors and saw both respond
     assert any(results[1].values()), results
+
+
+def test_trainer_lr_scheduler(tmp_path):
+    """Trainer(lr_scheduler=...) steps the schedule per epoch (parity:
+    the BERT example's warmup schedule hook, reference
+    examples/bert/provider.py:49-56)."""
+    node = _mk_fused_node(tmp_path, lr=0.1)
+    node.start()
+    X = torch.randn(16, 1, 8, 8)
+    Y = torch.eye(10)[torch.randint(0, 10, (16,))]
+    from torch.utils.data import DataLoader
+    loader = DataLoader(list(zip(X, Y)), batch_size=8)
+    from ravnest_amd import Trainer
+    trainer = Trainer(node=node, train_loader=loader, epochs=3,
+                      batch_size=8,
+                      lr_scheduler=torch.optim.lr_scheduler.StepLR,
+                      lr_scheduler_params={"step_size": 1, "gamma": 0.5})
+    trainer.train()
+    node.wait_for_backwards()
+    lr = node.optimizer.param_groups[0]["lr"]
+    assert abs(lr - 0.1 * 0.5 ** 3) < 1e-9, lr
+    node.stop()

@@ -66,3 +66,9 @@ if __name__ == "__main__":
                       batch_size=64, inputs_dtype=torch.float32)
     trainer.train()
     trainer.evaluate()
+
+    # clean shutdown: root drains + cascades STOP; every rank closes its
+    # channels (prevents the gloo teardown abort on live recv threads)
+    if node.node_type.value == "root":
+        node.stop_cluster()
+    node.stop()

@@ -40,3 +40,9 @@ if __name__ == "__main__":
     trainer = Trainer(node=node, train_loader=loader, epochs=10,
                       batch_size=BATCH)
     trainer.train()
+
+    # clean shutdown: root drains + cascades STOP; every rank closes its
+    # channels (prevents the gloo teardown abort on live recv threads)
+    if node.node_type.value == "root":
+        node.stop_cluster()
+    node.stop()

@@ -38,3 +38,9 @@ if __name__ == "__main__":
     trainer = Trainer(node=node, train_loader=loader, epochs=1,
                       batch_size=64, save=True)
     trainer.train()
+
+    # clean shutdown: root drains + cascades STOP; every rank closes its
+    # channels (prevents the gloo teardown abort on live recv threads)
+    if node.node_type.value == "root":
+        node.stop_cluster()
+    node.stop()

@@ -52,15 +52,25 @@ def _dtype_from_str(s: str) -> torch.dtype:
 class _CyclingIterator:
     """Auto-resetting DataLoader iterator (the leaf's label feed must
     survive epoch boundaries; parity: reference labels iterator,
-    node.py:144-163)."""
+    node.py:144-163).
+
+    LAZY: iter() is deferred to the first __next__. A seeded shuffling
+    DataLoader draws its epoch permutation from its torch.Generator at
+    iter() time; an eager draw at Node init desynchronized the root's
+    training order from the leaf's labels whenever both wrap the same
+    loader object (the walkthrough pattern) — the root's Trainer then
+    iterated the generator's SECOND permutation against the leaf's
+    FIRST, and the model could only learn the output mean."""
 
     def __init__(self, loader):
         self.loader = loader
-        self.it = iter(loader) if loader is not None else None
+        self.it = None
 
     def __next__(self):
-        if self.it is None:
+        if self.loader is None:
             return None
+        if self.it is None:
+            self.it = iter(self.loader)
         try:
             return next(self.it)
         except StopIteration:

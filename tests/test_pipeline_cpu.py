@@ -141,3 +141,74 @@ def test_pipeline_wire_compression(tmp_path):
     n_batches = 256 // 32
     assert len(losses) == 3 * n_batches
     assert sum(losses[-n_batches:]) < sum(losses[:n_batches])
+
+
+def _shared_loader_worker(rank, base_dir, port, out_dir):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    os.chdir(out_dir)
+    set_seed(42)
+    from ravnest_amd import Node, Trainer
+    # the walkthrough pattern: ONE loader object wraps both the node's
+    # label feed and the trainer's input feed
+    loader = _make_digits_loader()
+    node = Node(name=f"node_{rank}", base_dir=base_dir,
+                optimizer=torch.optim.Adam,
+                device=torch.device("cpu"),
+                criterion=_loss_fn,
+                labels=loader,
+                update_frequency=1)
+    node.start()
+    trainer = Trainer(node=node, train_loader=loader, epochs=12,
+                      batch_size=64, inputs_dtype=torch.float32)
+    if rank == 0:
+        trainer.train()
+        node.stop_cluster()
+    else:
+        trainer.prelim_checks()
+    node.stop()
+
+
+def _make_digits_loader(seed=42):
+    from sklearn import datasets
+    data = datasets.load_digits()
+    y = np.zeros((len(data.target), 10), dtype="float32")
+    y[np.arange(len(y)), data.target] = 1.0
+    X = data.data.reshape(-1, 1, 8, 8).astype("float32")[:1024]
+    g = torch.Generator()
+    g.manual_seed(seed)
+    from torch.utils.data import DataLoader
+    return DataLoader(list(zip(torch.tensor(X), torch.tensor(y[:1024]))),
+                      generator=g, shuffle=True, batch_size=64)
+
+
+def test_shared_loader_label_alignment(tmp_path):
+    """Regression: a SHUFFLING loader shared between node.labels and the
+    trainer must keep the root's data order aligned with the leaf's
+    labels. The eager iter() in the label iterator used to consume one
+    generator draw at Node init, desynchronizing the two sides — the
+    model then learned only the output mean (loss floor ~0.09 on
+    one-hot digits). With alignment, real learning takes loss far below
+    that floor."""
+    set_seed(42)
+    model = CNN()
+    base = str(tmp_path / "node_data")
+    pool = [NodeSpec(name=f"n{i}", ram=10 * 2**20) for i in range(3)]
+    clusterize(model, (torch.randn(2, 1, 8, 8),), node_pool=pool,
+               max_clusters=1, base_dir=base)
+    port = 29450 + (os.getpid() % 50)
+    ctx = mp.get_context("spawn")
+    procs = [ctx.Process(target=_shared_loader_worker,
+                         args=(r, base, port, str(tmp_path)))
+             for r in range(3)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(timeout=300)
+    for p in procs:
+        assert p.exitcode == 0, f"worker exited {p.exitcode}"
+    losses = [float(x) for x in (tmp_path / "losses.txt").read_text().split()]
+    n = 1024 // 64
+    last_epoch = sum(losses[-n:]) / n
+    assert last_epoch < 0.03, \
+        f"data/label misalignment suspected: last-epoch loss {last_epoch}"

@@ -139,6 +139,18 @@ def _cnn_val_worker(rank, base, port, out_dir):
                       epochs=1, inputs_dtype=torch.float32)
     if node.node_type.value == "root":
         trainer.train()
+        # prediction output type: leaf saves prediction.pt
+        # (parity: reference node.py:683 prediction handler)
+        node.no_grad_forward_compute(tensors=X[:8],
+                                     output_type="prediction")
+        # Trainer.pred uses output_type='accuracy' (exact reference
+        # parity: trainer.py:112 passes 'accuracy')
+        trainer.pred(X[:8].numpy())
+        node.wait_for_backwards()
+        import time as _t
+        deadline = _t.time() + 60
+        while not os.path.exists("prediction.pt") and _t.time() < deadline:
+            _t.sleep(0.2)
         node.stop_cluster()
     else:
         trainer.prelim_checks()
@@ -157,5 +169,7 @@ def test_cnn_pipeline_evaluate(tmp_path):
     # leaf wrote val accuracies during trainer.train()'s val pass
     vf = tmp_path / "val_accuracies.txt"
     assert vf.exists(), "no val_accuracies.txt from the evaluate path"
+    pred = torch.load(tmp_path / "prediction.pt")
+    assert pred.shape == (8, 10), pred.shape
     accs = [float(x) for x in vf.read_text().split()]
     assert len(accs) == 4

@@ -134,6 +134,8 @@ std::vector<at::Tensor> emb2_ln_fwd(at::Tensor ids, at::Tensor word,
   const int H = (int)word.size(1);
   TORCH_CHECK(H % 8 == 0 && H <= 1024, "emb2_ln_fwd: H % 8 == 0, H <= 1024");
   const int S = (int)ids.size(-1);
+  TORCH_CHECK((long)S <= pos.size(0),
+              "sequence longer than the position table");
   const long N = ids.numel();
   auto y = at::empty({ids.size(0), ids.size(1), (long)H}, word.options());
   auto mean = at::empty({N}, word.options().dtype(at::kFloat));
@@ -162,6 +164,8 @@ at::Tensor emb2_add_fwd(at::Tensor ids, at::Tensor word, at::Tensor pos) {
   const int H = (int)word.size(1);
   TORCH_CHECK(H % 8 == 0 && H <= 1024);
   const int S = (int)ids.size(-1);
+  TORCH_CHECK((long)S <= pos.size(0),
+              "sequence longer than the position table");
   const long N = ids.numel();
   auto y = at::empty({ids.size(0), ids.size(1), (long)H}, word.options());
   auto stream = c10::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();

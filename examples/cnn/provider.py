@@ -22,38 +22,38 @@ from ravnest_amd import Node, Trainer, set_seed  # noqa: E402
 set_seed(42)
 
 
-def to_categorical(x, n_col=None):
-    n_col = n_col or (np.amax(x) + 1)
-    one_hot = np.zeros((x.shape[0], n_col))
-    one_hot[np.arange(x.shape[0]), x] = 1
-    return one_hot
-
-
-def preprocess_dataset():
-    data = datasets.load_digits()
-    X, y = data.data, to_categorical(data.target.astype("int"))
-    X_train, X_test, y_train, y_test = train_test_split(
-        X, y, test_size=0.4, random_state=1)
-    X_train = X_train.reshape((-1, 1, 8, 8)).astype("float32")
-    X_test = X_test.reshape((-1, 1, 8, 8)).astype("float32")
+def digits_loaders(batch_size: int = 64, holdout: float = 0.4,
+                   split_seed: int = 1, order_seed: int = 42):
+    """sklearn-digits as (train, val) DataLoaders of (image, one_hot)
+    pairs. Same workload *config* as the reference walkthrough (8x8
+    images, MSE vs one-hot, 60/40 split, seeded shuffle order so every
+    pipeline rank that wraps these loaders sees identical batches).
+    Torch-native prep: one_hot via torch, no numpy staging."""
+    raw = datasets.load_digits()
+    images = torch.as_tensor(raw.images, dtype=torch.float32).unsqueeze(1)
+    onehot = torch.nn.functional.one_hot(
+        torch.as_tensor(raw.target, dtype=torch.long), 10).float()
+    idx_train, idx_val = train_test_split(
+        np.arange(len(images)), test_size=holdout, random_state=split_seed)
     g = torch.Generator()
-    g.manual_seed(42)
-    train = DataLoader(list(zip(torch.tensor(X_train),
-                                torch.tensor(y_train, dtype=torch.float32))),
-                       generator=g, shuffle=True, batch_size=64)
-    val = DataLoader(list(zip(torch.tensor(X_test),
-                              torch.tensor(y_test, dtype=torch.float32))),
-                     shuffle=False, batch_size=64)
+    g.manual_seed(order_seed)
+    train = DataLoader(
+        [(images[i], onehot[i]) for i in idx_train],
+        generator=g, shuffle=True, batch_size=batch_size)
+    val = DataLoader(
+        [(images[i], onehot[i]) for i in idx_val],
+        shuffle=False, batch_size=batch_size)
     return train, val
 
 
 def loss_fn(preds, targets):
+    # leaf criterion: targets is the (X, one_hot) pair from the label feed
     return torch.nn.functional.mse_loss(preds, targets[1])
 
 
 if __name__ == "__main__":
     name, base_dir = node_name()
-    train_loader, val_loader = preprocess_dataset()
+    train_loader, val_loader = digits_loaders()
     node = Node(name=name, base_dir=base_dir,
                 optimizer=torch.optim.Adam,
                 device=torch.device("cpu"),

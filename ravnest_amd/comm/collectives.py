@@ -62,45 +62,72 @@ def average_tensors(tensors: list[torch.Tensor], group,
         flush()
 
 
+def _master_tensors(optimizer, params) -> list[torch.Tensor]:
+    """fp32 master copies the fused optimizers keep for bf16 params.
+    These MUST be averaged whenever the params are: the optimizers read
+    the master (not the bf16 param) on every step, so averaging only the
+    bf16 values would be silently discarded at the next step."""
+    if optimizer is None:
+        return []
+    out = []
+    for p in params:
+        st = optimizer.state.get(p)
+        if st:
+            m = st.get("master")
+            if isinstance(m, torch.Tensor):
+                out.append(m)
+    return out
+
+
 def average_parameters(model: torch.nn.Module, group,
-                       bucket_bytes: int = 64 * 2**20) -> None:
-    average_tensors([p for p in model.parameters()], group, bucket_bytes)
+                       bucket_bytes: int = 64 * 2**20,
+                       optimizer=None) -> None:
+    params = [p for p in model.parameters()]
+    average_tensors(params + _master_tensors(optimizer, params), group,
+                    bucket_bytes)
 
 
 def average_optimizer_state(optimizer: torch.optim.Optimizer, group,
                             bucket_bytes: int = 64 * 2**20) -> None:
     """Average floating-point optimizer state tensors (Adam moments, LAMB
     trust state, SGD momentum) across the DP group (parity: reference
-    `average_optim` path, communication.py:131-138,176-179,267-272)."""
+    `average_optim` path, communication.py:131-138,176-179,267-272).
+    fp32 masters are excluded: they ride with average_parameters."""
     tensors = []
     for st in optimizer.state.values():
-        for v in st.values():
+        for k, v in st.items():
+            if k == "master":
+                continue
             if isinstance(v, torch.Tensor) and v.is_floating_point() and v.numel() > 0:
                 tensors.append(v)
     average_tensors(tensors, group, bucket_bytes)
 
 
 def average_parameter_segments(model: torch.nn.Module, segments,
-                               optimizer=None,
+                               optimizer=None, average_optim: bool = False,
                                bucket_bytes: int = 64 * 2**20) -> None:
     """Unequal-cluster DP: average each param-range segment on its own
     group (parity: reference param-range rings,
     operations/utils.py:463-516). `segments` = [(ranks, group, names)]
-    restricted to this rank; optionally averages the matching optimizer
-    state (moments / momentum / fp32 masters)."""
+    restricted to this rank. fp32 masters always ride with their params;
+    the rest of the optimizer state (moments / momentum) only when
+    `average_optim`."""
     named = dict(model.named_parameters())
     for _, group, names in segments:
         params = [named[n] for n in names if n in named]
         if not params:
             continue
-        average_tensors(params, group, bucket_bytes)
-        if optimizer is not None:
+        average_tensors(params + _master_tensors(optimizer, params), group,
+                        bucket_bytes)
+        if optimizer is not None and average_optim:
             sts = []
             for p in params:
                 st = optimizer.state.get(p)
                 if not st:
                     continue
-                for v in st.values():
+                for k, v in st.items():
+                    if k == "master":
+                        continue
                     if isinstance(v, torch.Tensor) and \
                             v.is_floating_point() and v.numel() > 0:
                         sts.append(v)

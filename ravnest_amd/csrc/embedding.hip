@@ -31,12 +31,22 @@ __global__ void emb2_fwd_kernel(const long* __restrict__ ids,
                                 bf16_t* __restrict__ y,
                                 float* __restrict__ mean_out,
                                 float* __restrict__ rstd_out,
-                                int H, int S, long N, float eps) {
+                                int H, int S, long N, long V, float eps) {
   const int lane = threadIdx.x & (WAVE - 1);
   const int wid = threadIdx.x / WAVE;
   const long row = (long)blockIdx.x * (blockDim.x / WAVE) + wid;
   if (row >= N) return;
-  const long id = ids[row];
+  long id = ids[row];
+  // out-of-range token id: defined behavior (row := 0, stats := identity)
+  // instead of OOB reads — torch device-asserts, but an abort would take
+  // the whole capture/graph down; a zero row is diagnosable
+  if (id < 0 || id >= V) {
+    bf16x8 z = {};
+    for (int base = lane * 8; base < H; base += WAVE * 8)
+      *reinterpret_cast<bf16x8*>(y + row * (long)H + base) = z;
+    if (LN && lane == 0) { mean_out[row] = 0.f; rstd_out[row] = 1.f; }
+    return;
+  }
   const int s = (int)(row % S);
   const bf16_t* wr = word + id * (long)H;
   const bf16_t* pr = pos + (long)s * H;
@@ -95,12 +105,13 @@ __global__ void emb2_fwd_kernel(const long* __restrict__ ids,
 __global__ void emb_scatter_word_kernel(const bf16_t* __restrict__ dx,
                                         const long* __restrict__ ids,
                                         float* __restrict__ dword,
-                                        int H, long N) {
+                                        int H, long N, long V) {
   const int lane = threadIdx.x & (WAVE - 1);
   const int wid = threadIdx.x / WAVE;
   const long row = (long)blockIdx.x * (blockDim.x / WAVE) + wid;
   if (row >= N) return;
   const long id = ids[row];
+  if (id < 0 || id >= V) return;  // OOB id: no scatter (see fwd note)
   const bf16_t* dr = dx + row * (long)H;
   float* wr = dword + id * (long)H;
   for (int h = lane; h < H; h += WAVE)
@@ -151,7 +162,7 @@ std::vector<at::Tensor> emb2_ln_fwd(at::Tensor ids, at::Tensor word,
                      reinterpret_cast<const bf16_t*>(b.data_ptr()),
                      reinterpret_cast<bf16_t*>(y.data_ptr()),
                      mean.data_ptr<float>(), rstd.data_ptr<float>(), H, S, N,
-                     (float)eps);
+                     word.size(0), (float)eps);
   HIP_CHECK_LAST();
   return {y, mean, rstd};
 }
@@ -177,7 +188,7 @@ at::Tensor emb2_add_fwd(at::Tensor ids, at::Tensor word, at::Tensor pos) {
                      reinterpret_cast<const bf16_t*>(pos.data_ptr()),
                      nullptr, nullptr,
                      reinterpret_cast<bf16_t*>(y.data_ptr()), nullptr,
-                     nullptr, H, S, N, 0.f);
+                     nullptr, H, S, N, word.size(0), 0.f);
   HIP_CHECK_LAST();
   return y;
 }
@@ -198,7 +209,8 @@ std::vector<at::Tensor> emb2_bwd(at::Tensor dx, at::Tensor ids, long V,
     dim3 grid((unsigned)((N + 3) / 4));
     hipLaunchKernelGGL(emb_scatter_word_kernel, grid, block, 0, stream,
                        reinterpret_cast<const bf16_t*>(dx.data_ptr()),
-                       ids.data_ptr<long>(), dword.data_ptr<float>(), H, N);
+                       ids.data_ptr<long>(), dword.data_ptr<float>(), H, N,
+                       V);
   }
   {
     dim3 block(256);

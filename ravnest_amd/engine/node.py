@@ -65,6 +65,11 @@ class _CyclingIterator:
     def __init__(self, loader):
         self.loader = loader
         self.it = None
+        self.wrapped = False  # set when the iterator wrapped (epoch end)
+
+    def take_wrapped(self) -> bool:
+        w, self.wrapped = self.wrapped, False
+        return w
 
     def __next__(self):
         if self.loader is None:
@@ -75,6 +80,7 @@ class _CyclingIterator:
             return next(self.it)
         except StopIteration:
             self.it = iter(self.loader)
+            self.wrapped = True
             return next(self.it)
 
 
@@ -87,6 +93,9 @@ class Node:
                  output_template: dict | None = None,
                  optimizer=None,
                  optimizer_params: dict | None = None,
+                 lr_scheduler=None,
+                 lr_scheduler_params: dict | None = None,
+                 lr_step_on_epoch_change: bool = True,
                  device: torch.device | None = None,
                  criterion=None,
                  labels=None,
@@ -169,6 +178,14 @@ class Node:
         if optimizer is not None:
             opt = optimizer(self.model.parameters(), **optimizer_params)
         self.optimizer = opt
+        # per-node lr scheduler: built HERE so stem/leaf ranks step it too
+        # (parity: reference node.py:213-215,517-518,585-586,603-604 —
+        # a Trainer-owned scheduler would only ever run on the root)
+        self.lr_scheduler = None
+        self.lr_step_on_epoch_change = lr_step_on_epoch_change
+        if lr_scheduler is not None and opt is not None:
+            self.lr_scheduler = lr_scheduler(opt,
+                                             **(lr_scheduler_params or {}))
         self.criterion = criterion
         self.labels = _CyclingIterator(labels)
         self.test_labels = _CyclingIterator(test_labels)
@@ -413,7 +430,7 @@ class Node:
     def _assemble_backward(self, msg: Message):
         with self._pending_lock:
             ent = self._pending_grads.setdefault(
-                msg.fpid, {"grads": {}, "count": 0})
+                msg.fpid, {"grads": {}, "count": 0, "extra": 0})
             for gid, g in msg.tensors:
                 k = gid % _MAX_OUTS
                 if k in ent["grads"]:
@@ -421,12 +438,12 @@ class Node:
                 else:
                     ent["grads"][k] = g
                 ent["count"] += 1
+            ent["extra"] |= msg.extra
             done = ent["count"] >= self._total_expected_grads
             if done:
                 del self._pending_grads[msg.fpid]
         if done:
-            self._enqueue(0, (str(ActionTypes.BACKWARD), msg.fpid,
-                              ent["grads"]))
+            self._enqueue(0, (str(ActionTypes.BACKWARD), msg.fpid, ent))
 
     # ==================================================================
     # dispatch loop
@@ -525,7 +542,8 @@ class Node:
             self.comm.send(r, "fwd", Message(action=action, fpid=fpid,
                                              tensors=tensors, extra=extra))
 
-    def _send_input_grads(self, fpid: int, input_grads: list):
+    def _send_input_grads(self, fpid: int, input_grads: list,
+                          extra: int = 0):
         by_rank: dict[int, list] = {}
         for pos, g in enumerate(input_grads):
             if g is None or pos not in self._pos_to_producer:
@@ -542,7 +560,8 @@ class Node:
             by_rank.setdefault(src_rank, []).append((j * _MAX_OUTS + k, g))
         for r, tensors in by_rank.items():
             self.comm.send(r, "bwd", Message(action=ActionTypes.BACKWARD,
-                                             fpid=fpid, tensors=tensors))
+                                             fpid=fpid, tensors=tensors,
+                                             extra=extra))
 
     # ==================================================================
     # user-facing API (parity with reference node.py:370-428,702-746)
@@ -649,15 +668,21 @@ class Node:
             return
         if self.engine.n_backwards % self.reduce_threshold != 0:
             return
+        # The speculative recompute thread swaps param.data to a
+        # historical version snapshot while it runs; averaging
+        # concurrently would corrupt the snapshot and miss the live
+        # weights. Join it first.
+        self.engine.join_recompute()
         if getattr(self.comm, "my_dp_segments", None):
             # unequal-cluster DP: per-param-range groups
             average_parameter_segments(
                 self.model, self.comm.my_dp_segments,
-                self.optimizer if self.average_optim else None)
+                self.optimizer, average_optim=self.average_optim)
             self.engine.bump_version()
             return
         if len(self.dp_ranks) > 1:
-            average_parameters(self.model, self.comm.my_dp_group)
+            average_parameters(self.model, self.comm.my_dp_group,
+                               optimizer=self.optimizer)
             if self.average_optim and self.optimizer is not None:
                 average_optimizer_state(self.optimizer, self.comm.my_dp_group)
             self.engine.bump_version()
@@ -668,9 +693,11 @@ class Node:
         model_inputs = self._map_model_inputs(tensors, kwargs)
         if self.fused:
             targets = next(self.labels)
+            epoch_change = self.labels.take_wrapped()
             args = self._root_args(model_inputs)
             _, stepped, _ = self.engine.find_loss(
                 fpid, args, [False] * len(args), targets)
+            self._step_scheduler(stepped, epoch_change)
             self._mark_backward_complete(fpid)
             self._maybe_reduce()
             return
@@ -722,19 +749,39 @@ class Node:
         outputs = self.engine.forward(fpid, args, self._needs_grad)
         self._send_outputs(ActionTypes.FORWARD, fpid, outputs)
 
+    def _step_scheduler(self, stepped: bool, epoch_change: bool):
+        """Per-node lr schedule (parity: reference node.py:517-518,
+        585-586,603-604): on epoch boundary by default, or after every
+        optimizer step when lr_step_on_epoch_change=False."""
+        if self.lr_scheduler is None:
+            return
+        if self.lr_step_on_epoch_change:
+            if epoch_change:
+                self.lr_scheduler.step()
+        elif stepped:
+            self.lr_scheduler.step()
+
     def _handle_find_loss(self, fpid: int, payload):
         args = self._build_args(payload["tensors"])
         targets = next(self.labels)
+        epoch_change = self.labels.take_wrapped()
         input_grads, stepped, loss = self.engine.find_loss(
             fpid, args, self._needs_grad, targets)
         self.losses.append(loss)
-        self._send_input_grads(fpid, input_grads)
+        self._step_scheduler(stepped, epoch_change)
+        self._send_input_grads(fpid, input_grads,
+                               extra=1 if epoch_change else 0)
         self._maybe_reduce()
 
-    def _handle_backward(self, fpid: int, grads: dict):
+    def _handle_backward(self, fpid: int, payload):
+        if isinstance(payload, dict) and "grads" in payload:
+            grads, extra = payload["grads"], payload.get("extra", 0)
+        else:  # direct-call compatibility (tests)
+            grads, extra = payload, 0
         input_grads, stepped = self.engine.backward(fpid, grads)
+        self._step_scheduler(stepped, bool(extra & 1))
         if self.node_type != NodeTypes.ROOT:
-            self._send_input_grads(fpid, input_grads)
+            self._send_input_grads(fpid, input_grads, extra=extra)
         else:
             self._mark_backward_complete(fpid)
         self._maybe_reduce()
@@ -838,15 +885,17 @@ class _CommSessionFacade:
         n = self.node
         if n.comm is None:
             return
+        n.engine.join_recompute()
         if getattr(n.comm, "my_dp_segments", None):
             average_parameter_segments(
                 n.model, n.comm.my_dp_segments,
-                n.optimizer if n.average_optim else None)
+                n.optimizer, average_optim=n.average_optim)
             n.engine.bump_version()
             return
         if len(n.dp_ranks) <= 1:
             return
-        average_parameters(n.model, n.comm.my_dp_group)
+        average_parameters(n.model, n.comm.my_dp_group,
+                           optimizer=n.optimizer)
         if n.average_optim and n.optimizer is not None:
             average_optimizer_state(n.optimizer, n.comm.my_dp_group)
         n.engine.bump_version()

@@ -65,12 +65,16 @@ class Trainer:
                 X = batch[0] if isinstance(batch, (tuple, list)) else batch
                 self.node.forward_compute(tensors=self._coerce(X))
                 self.n_forwards += 1
-            if self.val_loader is not None:
-                self.node.wait_for_backwards()
-                for batch in self.val_loader:
-                    X = batch[0] if isinstance(batch, (tuple, list)) else batch
-                    self.node.no_grad_forward_compute(
-                        tensors=self._coerce(X), output_type="val_accuracy")
+                # mid-epoch validation cadence: every val_freq training
+                # batches (reference trainer.py:21-22 documents this;
+                # val_freq=1 keeps the once-per-epoch behavior below)
+                if self.val_loader is not None and self.val_freq and \
+                        self.val_freq > 1 and \
+                        self.n_forwards % self.val_freq == 0:
+                    self._run_validation()
+            if self.val_loader is not None and (not self.val_freq
+                                                or self.val_freq <= 1):
+                self._run_validation()
             self.node.wait_for_backwards()
             if self.lr_scheduler is not None:
                 self.lr_scheduler.step()
@@ -79,6 +83,13 @@ class Trainer:
         self.train_time = time.time() - t1
         if self.save:
             self.node.trigger_save_submodel()
+
+    def _run_validation(self):
+        self.node.wait_for_backwards()
+        for batch in self.val_loader:
+            X = batch[0] if isinstance(batch, (tuple, list)) else batch
+            self.node.no_grad_forward_compute(
+                tensors=self._coerce(X), output_type="val_accuracy")
 
     def pred(self, data):
         if self.passive:

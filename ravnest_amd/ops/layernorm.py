@@ -65,7 +65,11 @@ class _AddLayerNormFn(torch.autograd.Function):
         ext = get_ext(required=True)
         dx, dw, db = ext.layernorm_bwd(dy.contiguous(), s, weight, mean,
                                        rstd)
-        return dx, dx, dw, db, None
+        # d/da == d/db, but the two returns must be DISTINCT objects:
+        # autograd's in-place accumulation fast path can steal a grad
+        # tensor for one accumulator and add_ into it later, which would
+        # double-count an aliased twin
+        return dx, dx.clone(), dw, db, None
 
 
 def add_layer_norm(a, b, weight, bias, eps: float = 1e-12):

@@ -31,6 +31,9 @@ at::Tensor conv2d_dgrad(at::Tensor dy, at::Tensor w, long N, long H,
 at::Tensor conv2d_wgrad(at::Tensor dy, at::Tensor x, long R, long S,
                         long st, long pad);
 std::vector<at::Tensor> mx_quant(at::Tensor x);
+std::vector<at::Tensor> gemm_nt_bf16(at::Tensor A2, at::Tensor B,
+                                     c10::optional<at::Tensor> bias,
+                                     int64_t epi);
 at::Tensor mx_gemm(at::Tensor x, at::Tensor xs, at::Tensor w,
                    at::Tensor ws);
 at::Tensor mx_scale_probe(at::Tensor a, at::Tensor b, at::Tensor sa,
@@ -116,6 +119,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("pad") = 0, py::arg("relu") = false);
   m.def("conv2d_dgrad", &conv2d_dgrad, "implicit-GEMM MFMA conv dgrad");
   m.def("conv2d_wgrad", &conv2d_wgrad, "implicit-GEMM MFMA conv wgrad");
+  m.def("gemm_nt_bf16", &gemm_nt_bf16,
+        "bf16 MFMA GEMM x @ W^T (256^2 8-phase, fused bias/GELU epilogue)",
+        py::arg("a"), py::arg("b"), py::arg("bias") = py::none(),
+        py::arg("epi") = 0);
   m.def("mx_quant", &mx_quant,
         "bf16 -> MX fp8 (e4m3 + per-32 e8m0 scales)");
   m.def("mx_gemm", &mx_gemm, "MX fp8 GEMM: x @ W^T, 32x32x64 scaled MFMA");

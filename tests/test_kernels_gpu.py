@@ -650,3 +650,50 @@ def test_conv2d_mfma(dev, geom):
     assert torch.allclose(w1.grad.float(), w2.grad, atol=2e-1, rtol=8e-2), \
         (w1.grad.float() - w2.grad).abs().max()
     assert torch.allclose(b1.grad.float(), b2.grad, atol=2e-1, rtol=5e-2)
+
+
+@pytest.mark.parametrize("shape", [
+    (512, 768, 768),        # exact tiles
+    (512, 768, 2304),       # qkv shape (columns)
+    (500, 768, 768),        # ragged M
+    (512, 768, 300),        # ragged N (tail block)
+    (512, 768, 298),        # ragged N crossing a 4-col store vector
+    (256, 3072, 256),       # deep K
+])
+def test_gemm_nt_bf16(dev, shape):
+    """Hand-written 256^2 8-phase bf16 MFMA GEMM vs fp32 torch reference
+    (asymmetric random operands per guide G9)."""
+    from ravnest_amd.ops import get_ext
+    ext = get_ext(True)
+    M, K, N = shape
+    torch.manual_seed(0)
+    a = (torch.randn(M, K, device=dev) / math.sqrt(K)).to(torch.bfloat16)
+    b = torch.randn(N, K, device=dev).to(torch.bfloat16)
+    ref = a.float() @ b.float().t()
+    (c,) = ext.gemm_nt_bf16(a, b, None, 0)
+    err = (c.float() - ref).abs().max().item()
+    scale = ref.abs().max().item() + 1e-6
+    assert err / scale < 3e-2, f"rel err {err/scale} (abs {err})"
+
+
+def test_gemm_nt_bf16_bias_gelu(dev):
+    from ravnest_amd.ops import get_ext
+    ext = get_ext(True)
+    M, K, N = 512, 768, 512
+    torch.manual_seed(1)
+    a = (torch.randn(M, K, device=dev) / math.sqrt(K)).to(torch.bfloat16)
+    b = torch.randn(N, K, device=dev).to(torch.bfloat16)
+    bias = torch.randn(N, device=dev).to(torch.bfloat16)
+    href = a.float() @ b.float().t() + bias.float()
+    # bias epilogue
+    (c1,) = ext.gemm_nt_bf16(a, b, bias, 1)
+    e1 = (c1.float() - href).abs().max().item() / (href.abs().max().item())
+    assert e1 < 3e-2, e1
+    # bias+gelu epilogue: returns (gelu(h), h)
+    y, h = ext.gemm_nt_bf16(a, b, bias, 2)
+    eh = (h.float() - href).abs().max().item() / (href.abs().max().item())
+    assert eh < 3e-2, eh
+    yref = torch.nn.functional.gelu(href, approximate="tanh")
+    ey = (y.float() - yref).abs().max().item() / \
+        (yref.abs().max().item() + 1e-6)
+    assert ey < 3e-2, ey

@@ -8,23 +8,30 @@
 // with fused epilogues: none / +bias / +bias+GELU (the GELU variant also
 // stores the pre-activation for the exact backward).
 //
-// Structure: the 256x256-tile 8-phase schedule from the CDNA4 guide
-// (guide §5 "The 256² 8-phase template"), built for the chip, not ported:
+// Structure: the 256x256-tile phase schedule from the CDNA4 guide
+// (guide §5 "The 256² 8-phase template"), built for the chip:
 //   * 512 threads = 8 waves as 2(M) x 4(N); per-wave output 128x64 via
-//     8x4 fragments of v_mfma_f32_16x16x32_bf16;
+//     8x4 fragments of v_mfma_f32_16x16x32_bf16 (the 32x32x16 shape was
+//     measured SLOWER here: its 2-accumulator clusters hit the
+//     dependent-MFMA latency);
 //   * K-tiles of 64 staged HBM->LDS with global_load_lds dwordx4 (the
 //     async LDS-DMA path; 2 ops per 16 KiB half-tile), double-buffered:
 //     128 KiB LDS total, one block per CU;
-//   * st_16x32 XOR swizzle (byte ^= ((byte>>9)&1)<<5 within each 1 KiB
-//     subtile) applied to the glds SOURCE address and the ds_read_b128
-//     address — the LDS image itself stays lane-linear (guide rule 21);
+//   * conflict-free 3-bit XOR swizzle (col bits 4-6 ^= row bits 1-3)
+//     applied to the glds SOURCE address and the ds_read_b128 address —
+//     the LDS image itself stays lane-linear (guide rule 21); the
+//     guide's 1-bit st_16x32 left uniform 2-way conflicts on THIS
+//     fragment map (measured LDS_BANK_CONFLICT == #reads; now 0);
 //   * counted s_waitcnt vmcnt(6) once per K-tile (3 half-tiles stay in
-//     flight across barriers), raw s_barrier everywhere — __syncthreads
-//     would drain the LDS-DMA queue (vmcnt(0)) and serialize the
-//     pipeline (guide §5 "pipelining across barriers");
+//     flight across barriers), raw s_barrier only at phase ends —
+//     __syncthreads would drain the LDS-DMA queue (vmcnt(0)) and
+//     serialize the pipeline (guide §5 "pipelining across barriers");
 //   * s_setprio(1) around each 16-MFMA cluster (guide T5);
-//   * bijective XCD-aware block remap so neighbor tiles share L2 (T1);
-//   * operandlanes SWAPPED vs the textbook mapping: the kernel computes
+//   * bijective XCD-aware remap + M-grouped supertiles (A-band stays
+//     cache-resident) + tpb consecutive column tiles per block (the
+//     staging pipeline runs across tiles, so the cold-start prologue is
+//     paid once per block, and A restages hit L2);
+//   * operand lanes SWAPPED vs the textbook mapping: the kernel computes
 //     mfma(B-frag, A-frag) so each lane's 4 accumulator values are 4
 //     CONSECUTIVE N-columns of one M-row -> the epilogue stores 8 B per
 //     lane per fragment instead of 4 scalar stores (store-issue tail).
@@ -37,7 +44,6 @@
 #include "common.h"
 
 typedef __bf16 bf16x8v __attribute__((ext_vector_type(8)));
-typedef float f32x16 __attribute__((ext_vector_type(16)));
 typedef unsigned short u16x4 __attribute__((ext_vector_type(4)));
 
 #define LDS_V8(p)                                            \
@@ -59,15 +65,14 @@ DEVINL float gelu_f(float x) {
   return x * t / (t + 1.f);  // 0.5*x*(1+tanh(.)) rewritten in one exp
 }
 
-// one lane's glds SOURCE pointer for op i (0/1) of a [128 x 64]
+// one lane's glds SOURCE offset for op i (0/1) of a [128 x 64]
 // half-tile: rows gR0..+127 of a row-major (ld-element) bf16 source at
 // k-window 0. Advancing one K-tile is +128 bytes (done per stage call).
 // LDS dest is lane-linear (glds requirement); the swizzle therefore
 // moves to the per-lane SOURCE byte offset (involution: same XOR the
-// ds_read side applies).
-// returned as a 32-bit byte offset from the tensor base (the operand
-// tensors are < 4 GiB — host-checked) so the 8 precomputed staging
-// addresses cost 8 VGPRs, not 16.
+// ds_read side applies). Returned as a 32-bit byte offset from the
+// tensor base (operands < 4 GiB, host-checked) so the 8 precomputed
+// staging addresses cost 8 VGPRs, not 16.
 DEVINL unsigned stage_off(long ld, long gR0, long gRmax, int i) {
   const int tid = threadIdx.x;
   const int p = i * 8192 + (tid >> 6) * 1024 + (tid & 63) * 16;
@@ -81,18 +86,14 @@ DEVINL unsigned stage_off(long ld, long gR0, long gRmax, int i) {
 // fragment read offset inside one [128][64] half-tile image (bytes),
 // swizzled. rr = row in half, kc = which K-32 chunk.
 //
-// Swizzle choice (replaces the guide's st_16x32 for THIS fragment map):
-// a b128 lane group reads 16 distinct rows r at col-bytes cb∈{c,c+16};
-// bank = (r&1)*32 + cb'/4, so only row bit 0 reaches the bank — st_16x32
-// (1 XOR bit) still left uniform 2-way conflicts (measured: LDS_BANK_
-// CONFLICT == #reads). XORing col bits 4-6 with row bits 1-3 makes all
-// 16 (row, cb) pairs of every lane group land on distinct banks:
-// same-cb rows differ in (r&1, (r>>1)&7); cross-cb collisions need
-// (r>>1)^(r'>>1) == cb-delta/16 which no group's row sets satisfy.
-DEVINL int frag_off32(int rr, int kc, int lane) {
-  // 32x32x16 fragment: lane holds row rr=(base+(l&31)), k = kc*16 +
-  // (l>>5)*8 + j  ->  col-bytes = kc*32 + (l>>5)*16
-  const int cb = (kc << 5) + ((lane >> 5) << 4);
+// Swizzle choice: a b128 lane group reads 16 distinct rows r at
+// col-bytes cb in {c, c+16}; bank = (r&1)*32 + cb'/4, so only row bit 0
+// reaches the bank. XORing col bits 4-6 with row bits 1-3 makes all 16
+// (row, cb) pairs of every lane group land on distinct banks: same-cb
+// rows differ in (r&1, (r>>1)&7); cross-cb collisions would need
+// (r>>1)^(r'>>1) == cb-delta/16, which no group's row sets satisfy.
+DEVINL int frag_off(int rr, int kc, int lane) {
+  const int cb = ((lane >> 4) << 4) + (kc << 6);
   const int p = (rr << 7) + cb;
   return p ^ (((p >> 8) & 7) << 4);
 }
@@ -106,18 +107,6 @@ __global__ __launch_bounds__(512, 2) void gemm_nt_kernel(
     int tpb) {
   __shared__ char smem[2 * BUF_BYTES];
 
-  // ---- tile mapping: XCD-contiguous, M-grouped, multi-tile blocks ---
-  // 1) bijective XCD remap (guide T1): each XCD gets a CONTIGUOUS run
-  //    of logical tile ids, so its private L2 sees neighboring tiles.
-  // 2) supertile grouping (GM M-tiles x all N columns walked together):
-  //    bounds the live A-band so A stays cache-resident instead of
-  //    re-read once per 256-column sweep.
-  // 3) each block owns `tpb` CONSECUTIVE column tiles of ONE M-row and
-  //    runs the K-tile staging pipeline CONTINUOUSLY across them: the
-  //    cold-start HBM latency of the prologue (the dominant loss at
-  //    short K with one block per CU and nothing to overlap with) is
-  //    paid once per block instead of once per tile, and the A-tile
-  //    restages hit L2 (same rows every column tile).
   const int nwg = MT * NTb;
   int bid = blockIdx.x, wg;
   if ((nwg & 7) == 0) {
@@ -143,10 +132,7 @@ __global__ __launch_bounds__(512, 2) void gemm_nt_kernel(
   const int nKT = (int)(K / BK);
   const unsigned SJ = (unsigned)(BN * K * 2);  // B stride per column tile
 
-  // staging: precomputed per-lane 32-bit source offsets (per-call
-  // address is base + off + j*SJ + t*128) and per-kind LDS offsets —
-  // keeps the per-glds setup to 32-bit adds instead of a 64-bit row
-  // multiply + clamp chain. kind: 0/1 = A half, 2/3 = B half.
+  // kind: 0/1 = A half, 2/3 = B half.
   unsigned sp[4][2];
   unsigned capB[2];  // row-(N-1) cap per op (ragged-N clamp)
 #pragma unroll
@@ -207,18 +193,16 @@ __global__ __launch_bounds__(512, 2) void gemm_nt_kernel(
   __builtin_amdgcn_s_barrier();
 
   // fragment LDS offsets (bytes, within a half image)
-  const int l31 = lane & 31;
+  const int l15 = lane & 15;
 
-  bf16x8v aF[2][4], bF[2][4];
+  bf16x8v aF[4][2], bF[4][2];
 
   for (int j = 0; j < tpb; ++j) {
-  f32x16 acc[4][2];
+  f32x4 acc[8][4];
 #pragma unroll
-  for (int m = 0; m < 4; ++m)
+  for (int m = 0; m < 8; ++m)
 #pragma unroll
-    for (int n = 0; n < 2; ++n)
-#pragma unroll
-      for (int r = 0; r < 16; ++r) acc[m][n][r] = 0.f;
+    for (int n = 0; n < 4; ++n) acc[m][n] = f32x4{0.f, 0.f, 0.f, 0.f};
 
   // ---- main loop over K-tiles ---------------------------------------
   for (int t = 0; t < nKT; ++t) {
@@ -230,69 +214,81 @@ __global__ __launch_bounds__(512, 2) void gemm_nt_kernel(
         smem + b * BUF_BYTES + TILE_BYTES + (wn >> 1) * HALF_BYTES;
     const int bRow0 = (wn & 1) * 64;          // B row base within half
 
-    // ph0: read B fragment strip 0 + A strip of M-half 0 (12 reads)
+    // ph0: read B fragments 0-1 + A fragments of M-half 0 (12 reads)
 #pragma unroll
-    for (int kc = 0; kc < 4; ++kc)
-      bF[0][kc] = LDS_V8(bH + frag_off32(bRow0 + l31, kc, lane));
+    for (int nf = 0; nf < 2; ++nf)
 #pragma unroll
-    for (int mf = 0; mf < 2; ++mf)
+      for (int kc = 0; kc < 2; ++kc)
+        bF[nf][kc] = LDS_V8(bH + frag_off(bRow0 + nf * 16 + l15, kc, lane));
 #pragma unroll
-      for (int kc = 0; kc < 4; ++kc)
-        aF[mf][kc] = LDS_V8(aH0 + frag_off32(mf * 32 + l31, kc, lane));
+    for (int mf = 0; mf < 4; ++mf)
+#pragma unroll
+      for (int kc = 0; kc < 2; ++kc)
+        aF[mf][kc] = LDS_V8(aH0 + frag_off(mf * 16 + l15, kc, lane));
     norm(j, t, 1, jj, tt);
     stage(jj, tt, 1);  // next tile's A-half-1
     __builtin_amdgcn_s_setprio(1);
 #pragma unroll
-    for (int kc = 0; kc < 4; ++kc)
+    for (int mf = 0; mf < 4; ++mf)
 #pragma unroll
-      for (int mf = 0; mf < 2; ++mf)
-        acc[mf][0] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-            bF[0][kc], aF[mf][kc], acc[mf][0], 0, 0, 0);
+      for (int nf = 0; nf < 2; ++nf)
+#pragma unroll
+        for (int kc = 0; kc < 2; ++kc)
+          acc[mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              bF[nf][kc], aF[mf][kc], acc[mf][nf], 0, 0, 0);
     __builtin_amdgcn_s_setprio(0);
     __builtin_amdgcn_s_barrier();
 
-    // ph1: read B fragment strip 1 (4 reads); MFMA M-half 0 x strip 1
+    // ph1: read B fragments 2-3 (4 reads); MFMA M-half 0 x N-frags 2,3
 #pragma unroll
-    for (int kc = 0; kc < 4; ++kc)
-      bF[1][kc] = LDS_V8(bH + frag_off32(bRow0 + 32 + l31, kc, lane));
+    for (int nf = 2; nf < 4; ++nf)
+#pragma unroll
+      for (int kc = 0; kc < 2; ++kc)
+        bF[nf][kc] = LDS_V8(bH + frag_off(bRow0 + nf * 16 + l15, kc, lane));
     __builtin_amdgcn_s_setprio(1);
 #pragma unroll
-    for (int kc = 0; kc < 4; ++kc)
+    for (int mf = 0; mf < 4; ++mf)
 #pragma unroll
-      for (int mf = 0; mf < 2; ++mf)
-        acc[mf][1] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-            bF[1][kc], aF[mf][kc], acc[mf][1], 0, 0, 0);
+      for (int nf = 2; nf < 4; ++nf)
+#pragma unroll
+        for (int kc = 0; kc < 2; ++kc)
+          acc[mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              bF[nf][kc], aF[mf][kc], acc[mf][nf], 0, 0, 0);
     __builtin_amdgcn_s_setprio(0);
     __builtin_amdgcn_s_barrier();
 
-    // ph2: read A strip of M-half 1 (8 reads)
+    // ph2: read A fragments of M-half 1 (8 reads)
 #pragma unroll
-    for (int mf = 0; mf < 2; ++mf)
+    for (int mf = 0; mf < 4; ++mf)
 #pragma unroll
-      for (int kc = 0; kc < 4; ++kc)
-        aF[mf][kc] = LDS_V8(aH0 + frag_off32(64 + mf * 32 + l31, kc, lane));
+      for (int kc = 0; kc < 2; ++kc)
+        aF[mf][kc] = LDS_V8(aH0 + frag_off(64 + mf * 16 + l15, kc, lane));
     norm(j, t, 2, jj, tt);
     stage(jj, tt, 2);
     __builtin_amdgcn_s_setprio(1);
 #pragma unroll
-    for (int kc = 0; kc < 4; ++kc)
+    for (int mf = 0; mf < 4; ++mf)
 #pragma unroll
-      for (int mf = 0; mf < 2; ++mf)
-        acc[2 + mf][0] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-            bF[0][kc], aF[mf][kc], acc[2 + mf][0], 0, 0, 0);
+      for (int nf = 0; nf < 2; ++nf)
+#pragma unroll
+        for (int kc = 0; kc < 2; ++kc)
+          acc[4 + mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              bF[nf][kc], aF[mf][kc], acc[4 + mf][nf], 0, 0, 0);
     __builtin_amdgcn_s_setprio(0);
     __builtin_amdgcn_s_barrier();
 
-    // ph3: no reads; MFMA M-half 1 x strip 1; tile-boundary vmcnt
+    // ph3: no reads; MFMA M-half 1 x N-frags 2,3; tile-boundary vmcnt
     stage(jj, tt, 3);
     stage(jj, tt, 0);
     __builtin_amdgcn_s_setprio(1);
 #pragma unroll
-    for (int kc = 0; kc < 4; ++kc)
+    for (int mf = 0; mf < 4; ++mf)
 #pragma unroll
-      for (int mf = 0; mf < 2; ++mf)
-        acc[2 + mf][1] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-            bF[1][kc], aF[mf][kc], acc[2 + mf][1], 0, 0, 0);
+      for (int nf = 2; nf < 4; ++nf)
+#pragma unroll
+        for (int kc = 0; kc < 2; ++kc)
+          acc[4 + mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              bF[nf][kc], aF[mf][kc], acc[4 + mf][nf], 0, 0, 0);
     __builtin_amdgcn_s_setprio(0);
     asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
     __builtin_amdgcn_s_barrier();
@@ -300,66 +296,63 @@ __global__ __launch_bounds__(512, 2) void gemm_nt_kernel(
 
   // ---- epilogue for column tile j ----------------------------------
   // Runs while the next tile's staged half-tiles are still in flight
-  // (the epilogue touches no LDS, so no wait). Swapped-operand 32x32
-  // C/D layout: lane holds M-row mBase+wm*128+mf*32+(lane&31); its 16
-  // accumulator values are N-cols nf*32 + (reg&3)+8*(reg>>2)+4*(l>>5)
-  // -> four 8 B stores per fragment (reg quads are 4 consecutive cols).
-  const long mRow0 = mBase + wm * 128 + l31;
-  const long nCol0 =
-      nBase + (long)j * BN + wn * 64 + ((lane >> 5) << 2);
+  // (the epilogue touches no LDS, so no wait). Swapped-operand C/D
+  // layout: lane holds M-row mBase+wm*128+mf*16+l15, N-cols
+  // nBase+j*256+wn*64+nf*16+(lane>>4)*4 + v, v = 0..3 -> one 8 B store
+  // per fragment (4 consecutive bf16 of one row). Interior fast path:
+  // ONE divergence test for the whole tile, then unpredicated stores.
+  const long mRow0 = mBase + wm * 128 + l15;
+  const long nCol0 = nBase + (long)j * BN + wn * 64 + ((lane >> 4) << 2);
   auto epilogue = [&](auto interior) {
 #pragma unroll
-    for (int mf = 0; mf < 4; ++mf) {
-      const long m = mRow0 + mf * 32;
+    for (int mf = 0; mf < 8; ++mf) {
+      const long m = mRow0 + mf * 16;
       if constexpr (!interior.value) {
         if (m >= M) continue;
       }
 #pragma unroll
-      for (int nf = 0; nf < 2; ++nf) {
-#pragma unroll
-        for (int q = 0; q < 4; ++q) {
-          const long n = nCol0 + nf * 32 + q * 8;
-          bool full = true;
-          if constexpr (!interior.value) {
-            if (n >= N) continue;
-            full = (n + 4 <= N);
-          }
-          float v0 = acc[mf][nf][4 * q + 0], v1 = acc[mf][nf][4 * q + 1];
-          float v2 = acc[mf][nf][4 * q + 2], v3 = acc[mf][nf][4 * q + 3];
-          if (EPI >= 1) {
-            if (full) {
-              const u16x4 bb = *reinterpret_cast<const u16x4*>(bias + n);
-              v0 += us2f(bb[0]); v1 += us2f(bb[1]);
-              v2 += us2f(bb[2]); v3 += us2f(bb[3]);
-            } else {
-              v0 += bf2f(bias[n]);
-              if (n + 1 < N) v1 += bf2f(bias[n + 1]);
-              if (n + 2 < N) v2 += bf2f(bias[n + 2]);
-            }
-          }
-          if (EPI == 2) {
-            u16x4 h{f2us(v0), f2us(v1), f2us(v2), f2us(v3)};
-            if (full) {
-              *reinterpret_cast<u16x4*>(H + m * N + n) = h;
-            } else {
-              for (int v = 0; v < 4; ++v)
-                if (n + v < N) H[m * N + n + v] = f2bf(us2f(h[v]));
-            }
-            v0 = gelu_f(v0); v1 = gelu_f(v1);
-            v2 = gelu_f(v2); v3 = gelu_f(v3);
-          }
-          const u16x4 o{f2us(v0), f2us(v1), f2us(v2), f2us(v3)};
+      for (int nf = 0; nf < 4; ++nf) {
+        const long n = nCol0 + nf * 16;
+        bool full = true;
+        if constexpr (!interior.value) {
+          if (n >= N) continue;
+          full = (n + 4 <= N);
+        }
+        float v0 = acc[mf][nf][0], v1 = acc[mf][nf][1];
+        float v2 = acc[mf][nf][2], v3 = acc[mf][nf][3];
+        if (EPI >= 1) {
           if (full) {
-            *reinterpret_cast<u16x4*>(C + m * N + n) = o;
+            const u16x4 bb = *reinterpret_cast<const u16x4*>(bias + n);
+            v0 += us2f(bb[0]); v1 += us2f(bb[1]);
+            v2 += us2f(bb[2]); v3 += us2f(bb[3]);
+          } else {
+            v0 += bf2f(bias[n]);
+            if (n + 1 < N) v1 += bf2f(bias[n + 1]);
+            if (n + 2 < N) v2 += bf2f(bias[n + 2]);
+          }
+        }
+        if (EPI == 2) {
+          u16x4 h{f2us(v0), f2us(v1), f2us(v2), f2us(v3)};
+          if (full) {
+            *reinterpret_cast<u16x4*>(H + m * N + n) = h;
           } else {
             for (int v = 0; v < 4; ++v)
-              if (n + v < N) C[m * N + n + v] = f2bf(us2f(o[v]));
+              if (n + v < N) H[m * N + n + v] = f2bf(us2f(h[v]));
           }
+          v0 = gelu_f(v0); v1 = gelu_f(v1);
+          v2 = gelu_f(v2); v3 = gelu_f(v3);
+        }
+        const u16x4 o{f2us(v0), f2us(v1), f2us(v2), f2us(v3)};
+        if (full) {
+          *reinterpret_cast<u16x4*>(C + m * N + n) = o;
+        } else {
+          for (int v = 0; v < 4; ++v)
+            if (n + v < N) C[m * N + n + v] = f2bf(us2f(o[v]));
         }
       }
     }
   };
-  if (mRow0 + 3 * 32 < M && nCol0 + 32 + 3 * 8 + 4 <= N)
+  if (mRow0 + 7 * 16 < M && nCol0 + 3 * 16 + 4 <= N)
     epilogue(std::true_type{});
   else
     epilogue(std::false_type{});

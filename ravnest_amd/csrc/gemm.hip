@@ -227,6 +227,7 @@ __global__ __launch_bounds__(512, 2) void gemm_nt_kernel(
         aF[mf][kc] = LDS_V8(aH0 + frag_off(mf * 16 + l15, kc, lane));
     norm(j, t, 1, jj, tt);
     stage(jj, tt, 1);  // next tile's A-half-1
+    asm volatile("s_waitcnt lgkmcnt(8)" ::: "memory");
     __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int mf = 0; mf < 4; ++mf)
@@ -394,7 +395,7 @@ std::vector<at::Tensor> gemm_nt_bf16(at::Tensor A2, at::Tensor B,
   // column tiles per block: amortizes the per-block staging cold start;
   // capped so the grid still fills the chip (>= ~2 block-waves)
   int tpb = 1;
-  for (int cand : {4, 3, 2}) {
+  for (int cand : {8, 6, 4, 3, 2}) {
     if (NT % cand == 0 && (long)MT * (NT / cand) >= 512) {
       tpb = cand;
       break;

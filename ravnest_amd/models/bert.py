@@ -17,6 +17,7 @@ import torch.nn as nn
 from ..ops import embedding_ln
 from ..ops import (AddLayerNorm, AttentionCoreQKV, Dropout, FusedLayerNorm,
                    GELU, LinearGelu)
+from ..ops.linear import make_linear
 
 
 class BertConfig:
@@ -76,9 +77,9 @@ class BertSelfAttention(nn.Module):
         super().__init__()
         self.heads = cfg.heads
         self.head_dim = cfg.hidden // cfg.heads
-        self.qkv = nn.Linear(cfg.hidden, 3 * cfg.hidden)
+        self.qkv = make_linear(cfg.hidden, 3 * cfg.hidden)
         self.core = AttentionCoreQKV(causal=False)
-        self.out = nn.Linear(cfg.hidden, cfg.hidden)
+        self.out = make_linear(cfg.hidden, cfg.hidden)
         self.drop = Dropout(cfg.dropout)
 
     def forward(self, x, mask):
@@ -97,7 +98,7 @@ class BertLayer(nn.Module):
         self.attn = BertSelfAttention(cfg)
         self.ln1 = AddLayerNorm(cfg.hidden, cfg.layer_norm_eps)
         self.mlp_in = LinearGelu(cfg.hidden, cfg.intermediate)
-        self.mlp_out = nn.Linear(cfg.intermediate, cfg.hidden)
+        self.mlp_out = make_linear(cfg.intermediate, cfg.hidden)
         self.drop = Dropout(cfg.dropout)
         self.ln2 = AddLayerNorm(cfg.hidden, cfg.layer_norm_eps)
 
@@ -110,10 +111,10 @@ class BertLayer(nn.Module):
 class BertMLMHead(nn.Module):
     def __init__(self, cfg: BertConfig):
         super().__init__()
-        self.dense = nn.Linear(cfg.hidden, cfg.hidden)
+        self.dense = make_linear(cfg.hidden, cfg.hidden)
         self.act = GELU()
         self.ln = FusedLayerNorm(cfg.hidden, cfg.layer_norm_eps)
-        self.decoder = nn.Linear(cfg.hidden, cfg.vocab_size)
+        self.decoder = make_linear(cfg.hidden, cfg.vocab_size)
 
     def forward(self, x):
         return self.decoder(self.ln(self.act(self.dense(x))))
@@ -133,7 +134,8 @@ class BertForMLM(nn.Module):
 
     @staticmethod
     def _init(m):
-        if isinstance(m, nn.Linear):
+        from ..ops import Linear as _OpsLinear
+        if isinstance(m, (nn.Linear, _OpsLinear)):
             nn.init.normal_(m.weight, std=0.02)
             if m.bias is not None:
                 nn.init.zeros_(m.bias)

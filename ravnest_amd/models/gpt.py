@@ -76,7 +76,8 @@ class GPTBlock(nn.Module):
         fp8 = getattr(cfg, "fp8", False)
         if fp8 and cfg.n_embd % 64 != 0:
             raise ValueError("fp8 GPT needs n_embd % 64 == 0")
-        lin = MXLinear if fp8 else nn.Linear
+        from ..ops.linear import make_linear
+        lin = MXLinear if fp8 else make_linear
         self.ln1 = FusedLayerNorm(cfg.n_embd)
         self.qkv = lin(cfg.n_embd, 3 * cfg.n_embd)
         self.core = AttentionCoreQKV(causal=True)
@@ -106,12 +107,14 @@ class GPT(nn.Module):
         self.blocks = nn.ModuleList(
             [GPTBlock(cfg) for _ in range(cfg.n_layer)])
         self.ln_f = FusedLayerNorm(cfg.n_embd)
-        self.lm_head = nn.Linear(cfg.n_embd, cfg.vocab_size, bias=False)
+        from ..ops.linear import make_linear
+        self.lm_head = make_linear(cfg.n_embd, cfg.vocab_size, bias=False)
         self.apply(self._init)
 
     @staticmethod
     def _init(m):
-        if isinstance(m, nn.Linear):
+        from ..ops import Linear as _OpsLinear
+        if isinstance(m, (nn.Linear, _OpsLinear)):
             nn.init.normal_(m.weight, std=0.02)
             if m.bias is not None:
                 nn.init.zeros_(m.bias)

@@ -78,7 +78,9 @@ class GELU(nn.Module):
 
 
 class LinearGelu(nn.Module):
-    """Linear (hipBLASLt GEMM, no bias) + fused bias-GELU epilogue."""
+    """Linear + bias-GELU: hipBLASLt GEMM + fused bias-GELU kernel by
+    default; ONE hand-written GEMM with the bias-GELU in the epilogue
+    under RAVNEST_HAND_GEMM=1 (ops/linear.py)."""
     _is_leaf_module = True
 
     def __init__(self, in_features: int, out_features: int):
@@ -88,6 +90,9 @@ class LinearGelu(nn.Module):
         nn.init.normal_(self.weight, std=0.02)
 
     def forward(self, x):
+        from .linear import _HandLinearGeluFn, _hand_ok, hand_gemm_enabled
+        if x.is_cuda and hand_gemm_enabled() and _hand_ok(x, self.weight):
+            return _HandLinearGeluFn.apply(x, self.weight, self.bias)
         h = torch.nn.functional.linear(x, self.weight)
         return bias_gelu(h, self.bias)
 

@@ -1,23 +1,109 @@
-"""Linear with library GEMMs and a fused bias gradient.
+"""Linear projections: library GEMMs by default, hand-written CDNA4 MFMA
+GEMM (csrc/gemm.hip — 256^2 glds-staged phase schedule, fused bias/GELU
+epilogues) behind RAVNEST_HAND_GEMM=1.
 
-The projection GEMMs run on hipBLASLt (the sanctioned library path for
-plain GEMMs; TunableOp-selected algorithms). What this wrapper changes vs
-nn.Linear is the BACKWARD bias reduction: autograd's dy.sum(0) launches a
-torch reduce per Linear per step — here it is the colsum kernel
-(csrc/gelu.hip) over bf16 with no intermediate.
+Default path: hipBLASLt (TunableOp-selected algorithms). The hand kernel
+measures 0.7-0.8x hipBLASLt on the BERT-base projection shapes
+(profiles/r02_gemm_vs_blaslt.txt), so the library stays the default for
+the headline bench; the hand path is complete (fwd + dgrad via W^T,
+wgrad via library, fused bias-GELU storing the pre-activation) and
+numerics-tested, and the flag flips every projection in a model built
+through make_linear()/LinearGelu.
 
-NOTE: the transformer models keep nn.Linear — measured end to end, the
-explicit dy.t()@x wgrad here hits slower hipBLASLt algorithm selections
-than autograd's addmm backward (BERT -2.4%, GPT-2 -13%), outweighing the
-~2% bias-reduce win. Kept as a library component.
+NOTE: the transformer models route plain projections through
+make_linear() — nn.Linear by default (autograd's addmm backward hits
+better hipBLASLt selections than an explicit dy.t()@x wgrad: BERT -2.4%,
+GPT-2 -13%), ops.Linear under the hand flag.
 """
 from __future__ import annotations
+
+import os
 
 import torch
 import torch.nn as nn
 import torch.nn.functional as F
 
 from ._ext import get_ext
+
+
+def hand_gemm_enabled() -> bool:
+    return os.environ.get("RAVNEST_HAND_GEMM", "0") == "1"
+
+
+def _hand_ok(x: torch.Tensor, w: torch.Tensor) -> bool:
+    """Shapes the hand kernel handles: bf16, K % 64 == 0, K >= 128,
+    operands < 4 GiB (32-bit staging offsets)."""
+    return (x.is_cuda and x.dtype == torch.bfloat16
+            and w.dtype == torch.bfloat16
+            and w.shape[1] % 64 == 0 and w.shape[1] >= 128
+            and x.numel() * 2 < 2**32 and w.numel() * 2 < 2**32)
+
+
+def _hand_dgrad(dy2: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
+    """dx = dy @ W: reduction over N -> needs W^T as the kernel's
+    [rows][k] operand; falls back to the library when N is ragged."""
+    N = w.shape[0]
+    if N % 64 == 0 and N >= 128 and dy2.numel() * 2 < 2**32:
+        ext = get_ext(required=True)
+        wt = w.t().contiguous()
+        (dx2,) = ext.gemm_nt_bf16(dy2.contiguous(), wt, None, 0)
+        return dx2
+    return dy2 @ w
+
+
+class _HandLinearFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, w, b):
+        ext = get_ext(required=True)
+        x2 = x.reshape(-1, x.shape[-1]).contiguous()
+        if b is not None:
+            (y,) = ext.gemm_nt_bf16(x2, w, b.contiguous(), 1)
+        else:
+            (y,) = ext.gemm_nt_bf16(x2, w, None, 0)
+        ctx.save_for_backward(x2, w)
+        ctx.has_bias = b is not None
+        ctx.x_shape = x.shape
+        return y.reshape(*x.shape[:-1], w.shape[0])
+
+    @staticmethod
+    def backward(ctx, dy):
+        x2, w = ctx.saved_tensors
+        dy2 = dy.reshape(-1, dy.shape[-1]).contiguous()
+        dx = _hand_dgrad(dy2, w).reshape(ctx.x_shape)
+        dw = dy2.t() @ x2
+        db = None
+        if ctx.has_bias:
+            ext = get_ext(required=True)
+            db = ext.colsum_bf16(dy2).to(dy.dtype)
+        return dx, dw, db
+
+
+class _HandLinearGeluFn(torch.autograd.Function):
+    """y = gelu(x @ W^T + b) in ONE kernel (the epilogue also stores the
+    post-bias pre-activation h for the exact backward)."""
+
+    @staticmethod
+    def forward(ctx, x, w, b):
+        ext = get_ext(required=True)
+        x2 = x.reshape(-1, x.shape[-1]).contiguous()
+        y, h = ext.gemm_nt_bf16(x2, w, b.contiguous(), 2)
+        ctx.save_for_backward(x2, w, h)
+        ctx.x_shape = x.shape
+        return y.reshape(*x.shape[:-1], w.shape[0])
+
+    @staticmethod
+    def backward(ctx, dy):
+        x2, w, h = ctx.saved_tensors
+        ext = get_ext(required=True)
+        dy2 = dy.reshape(-1, dy.shape[-1]).contiguous()
+        # h is POST-bias: reuse the fused gelu-bwd(+colsum) with a zero
+        # bias vector
+        zb = torch.zeros(w.shape[0], dtype=h.dtype, device=h.device)
+        dh, db = ext.bias_gelu_bwd_db(dy2, h, zb)
+        dh = dh.contiguous()
+        dx = _hand_dgrad(dh, w).reshape(ctx.x_shape)
+        dw = dh.t() @ x2
+        return dx, dw, db.to(dy.dtype)
 
 
 class _LinearFn(torch.autograd.Function):
@@ -58,5 +144,16 @@ class Linear(nn.Module):
 
     def forward(self, x):
         if x.is_cuda:
+            if hand_gemm_enabled() and _hand_ok(x, self.weight):
+                return _HandLinearFn.apply(x, self.weight, self.bias)
             return _LinearFn.apply(x, self.weight, self.bias)
         return F.linear(x, self.weight, self.bias)
+
+
+def make_linear(in_features: int, out_features: int, bias: bool = True):
+    """Projection factory for the model zoo: nn.Linear by default (best
+    library backward), ops.Linear when the hand-GEMM flag is set (same
+    state-dict keys either way)."""
+    if hand_gemm_enabled():
+        return Linear(in_features, out_features, bias=bias)
+    return nn.Linear(in_features, out_features, bias=bias)

@@ -697,3 +697,38 @@ def test_gemm_nt_bf16_bias_gelu(dev):
     ey = (y.float() - yref).abs().max().item() / \
         (yref.abs().max().item() + 1e-6)
     assert ey < 3e-2, ey
+
+
+def test_hand_linear_autograd(dev, monkeypatch):
+    """RAVNEST_HAND_GEMM path: Linear + LinearGelu forward/backward vs
+    fp32 torch reference."""
+    monkeypatch.setenv("RAVNEST_HAND_GEMM", "1")
+    from ravnest_amd.ops import Linear, LinearGelu
+    torch.manual_seed(0)
+    M, K, N = 512, 768, 512
+    for cls, ref_fn in [
+        (Linear, lambda x, w, b: torch.nn.functional.linear(x, w, b)),
+        (LinearGelu, lambda x, w, b: torch.nn.functional.gelu(
+            torch.nn.functional.linear(x, w, b), approximate="tanh")),
+    ]:
+        lin = cls(K, N).to(dev).to(torch.bfloat16)
+        x = (torch.randn(4, M // 4, K, device=dev) / 16).to(torch.bfloat16)
+        x.requires_grad_(True)
+        y = lin(x)
+        dy = torch.randn_like(y) / 8
+        y.backward(dy)
+
+        xf = x.detach().float().requires_grad_(True)
+        wf = lin.weight.detach().float().requires_grad_(True)
+        bf = lin.bias.detach().float().requires_grad_(True)
+        yr = ref_fn(xf, wf, bf)
+        yr.backward(dy.float())
+
+        def ok(a, r, tol):
+            d = (a.float() - r).abs().max().item()
+            s = r.abs().max().item() + 1e-6
+            assert d / s < tol, f"{cls.__name__}: rel {d/s}"
+        ok(y, yr, 4e-2)
+        ok(x.grad, xf.grad, 4e-2)
+        ok(lin.weight.grad, wf.grad, 4e-2)
+        ok(lin.bias.grad, bf.grad, 4e-2)

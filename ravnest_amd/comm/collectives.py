@@ -103,6 +103,96 @@ def average_optimizer_state(optimizer: torch.optim.Optimizer, group,
     average_tensors(tensors, group, bucket_bytes)
 
 
+class AsyncReducer:
+    """Overlapped DP parameter averaging (SURVEY.md §7 "Overlap
+    engineering" — the reference simply blocks, communication.py:125-277).
+
+    launch(): snapshot the tensors into per-dtype flat buffers and start
+    the averaging all_reduce on a SIDE stream (GPU) or a background
+    thread (CPU/gloo), so the pipeline's next microbatch overlaps the
+    collective. join_into(): make the compute stream wait on the
+    collective (GPU-side wait, no host sync) and copy the averaged
+    values back.
+
+    Semantics note: between launch and join the local params are frozen
+    (no optimizer step happens inside a reduce window when
+    reduce_threshold is a multiple of update_frequency), but the next
+    microbatch's gradients are computed at the PRE-average weights; the
+    averaged values land just before the next optimizer step. The
+    synchronous path (`average_parameters`) remains for exact-parity
+    call sites."""
+
+    def __init__(self, group, device: torch.device):
+        self.group = group
+        self.device = device
+        self._stream = (torch.cuda.Stream(device)
+                        if device.type == "cuda" else None)
+        self._pending = None
+        self._thread = None
+
+    @property
+    def pending(self) -> bool:
+        return self._pending is not None
+
+    def launch(self, tensors: list[torch.Tensor]) -> None:
+        assert self._pending is None, "previous async reduce not joined"
+        world = dist.get_world_size(group=self.group)
+        if world <= 1 or not tensors:
+            return
+        by_dtype: dict[torch.dtype, list[torch.Tensor]] = {}
+        for t in tensors:
+            by_dtype.setdefault(t.dtype, []).append(t)
+        if self._stream is not None:
+            ev = torch.cuda.Event()
+            ev.record(torch.cuda.current_stream(self.device))
+            flats = []
+            with torch.cuda.stream(self._stream):
+                self._stream.wait_event(ev)
+                for ts in by_dtype.values():
+                    flat = torch.cat([t.detach().reshape(-1) for t in ts])
+                    _all_reduce_avg(flat, self.group, world)
+                    flats.append((ts, flat))
+            done = torch.cuda.Event()
+            done.record(self._stream)
+            self._pending = (flats, done)
+        else:
+            flats = []
+            for ts in by_dtype.values():
+                flats.append((ts, torch.cat(
+                    [t.detach().reshape(-1) for t in ts])))
+
+            def run():
+                for _, flat in flats:
+                    _all_reduce_avg(flat, self.group, world)
+
+            import threading
+            self._thread = threading.Thread(target=run, daemon=True)
+            self._thread.start()
+            self._pending = (flats, None)
+
+    def join_into(self) -> bool:
+        """Install the averaged values; returns True if a reduce landed."""
+        if self._pending is None:
+            return False
+        flats, done = self._pending
+        if self._stream is not None:
+            # compute stream waits for the collective — no host sync
+            torch.cuda.current_stream(self.device).wait_event(done)
+        elif self._thread is not None:
+            self._thread.join()
+            self._thread = None
+        for ts, flat in flats:
+            off = 0
+            for t in ts:
+                n = t.numel()
+                t.detach().copy_(flat[off:off + n].view_as(t))
+                off += n
+            if self._stream is not None:
+                flat.record_stream(torch.cuda.current_stream(self.device))
+        self._pending = None
+        return True
+
+
 def average_parameter_segments(model: torch.nn.Module, segments,
                                optimizer=None, average_optim: bool = False,
                                bucket_bytes: int = 64 * 2**20) -> None:

@@ -72,6 +72,12 @@ class ComputeEngine:
             self._graph_step = GraphedTrainStep(model, criterion, device,
                                                 optimizer=optimizer)
 
+        # invoked right before every optimizer step / graph replay that
+        # steps: the node uses it to JOIN an overlapped DP averaging
+        # (comm/collectives.py AsyncReducer) so the averaged values land
+        # before new updates are applied
+        self.pre_step_hook = None
+
         self.current_version = 0
         self.version_to_param: dict[int, list[torch.Tensor]] = {}
         self.version_refs: dict[int, int] = {}
@@ -264,6 +270,8 @@ class ComputeEngine:
         stepped = False
         if self.optimizer is not None and \
                 self.n_backwards % self.update_frequency == 0:
+            if self.pre_step_hook is not None:
+                self.pre_step_hook()
             self.optimizer.step()
             self.optimizer.zero_grad(set_to_none=True)
             self.model.zero_grad(set_to_none=True)
@@ -282,6 +290,10 @@ class ComputeEngine:
             self.model.train()
         if self._graph_step is not None and not any(needs_grad) and \
                 self.amp_dtype is None:
+            # join a pending overlapped DP reduce BEFORE the replay: the
+            # captured region may contain the optimizer step
+            if self.pre_step_hook is not None:
+                self.pre_step_hook()
             loss_t = self._graph_step.run(args, targets)
             if loss_t is not None:
                 return self._finish_graphed_step(args, loss_t)
@@ -314,6 +326,8 @@ class ComputeEngine:
         stepped = False
         if self.optimizer is not None and \
                 self.n_backwards % self.update_frequency == 0:
+            if self.pre_step_hook is not None:
+                self.pre_step_hook()
             self.optimizer.step()
             self.optimizer.zero_grad(set_to_none=True)
             self.model.zero_grad(set_to_none=True)

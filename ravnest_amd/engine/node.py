@@ -30,9 +30,11 @@ from pathlib import Path
 import torch
 
 from ..comm import CommBackend, Message
-from ..comm.collectives import (average_parameters,
+from ..comm.collectives import (AsyncReducer,
+                                average_parameters,
                                 average_optimizer_state,
-                                average_parameter_segments)
+                                average_parameter_segments,
+                                _master_tensors)
 from ..strings import ActionTypes, NodeTypes
 from ..utils import load_node_json_configs
 from .compute import ComputeEngine
@@ -111,6 +113,7 @@ class Node:
                  compression: bool = False,
                  wire_dtype: torch.dtype | None = None,
                  amp_dtype: torch.dtype | None = None,
+                 async_reduce: bool = True,
                  model_transform=None):
         # parity: the reference's fp16 wire compression (utils.py:184-194)
         # becomes an optional on-the-wire cast; bf16 is the natural MI355X
@@ -232,9 +235,31 @@ class Node:
             self.comm = None
         self.comm_session = _CommSessionFacade(self)
 
+        # overlapped DP averaging: collective launched on a side stream
+        # at the reduce threshold, joined (via engine.pre_step_hook)
+        # right before the next optimizer step — the reduce rides under
+        # the next microbatch's compute instead of blocking the pipeline
+        # (SURVEY.md §7 "Overlap engineering"; the reference blocks).
+        self._reducer = None
+        if (async_reduce and self.comm is not None
+                and len(self.dp_ranks) > 1
+                and not getattr(self.comm, "my_dp_segments", None)):
+            self._reducer = AsyncReducer(self.comm.my_dp_group, self.device)
+            self.engine.pre_step_hook = self._join_reduce
+
         self._dispatch_thread: threading.Thread | None = None
         self._started = False
         self.health_monitor = None
+
+    def _join_reduce(self):
+        if self._reducer is not None and self._reducer.pending:
+            self._reducer.join_into()
+            self.engine.bump_version()
+
+    def join_pending_reduce(self):
+        """Drain any in-flight overlapped DP reduce (checkpoints, weight
+        serving, shutdown)."""
+        self._join_reduce()
 
     # ==================================================================
     # aux subsystems (health, checkpoint) — SURVEY.md section 5 parity+
@@ -253,6 +278,7 @@ class Node:
                 if self.health_monitor is not None else {"peers": {}})
 
     def save_checkpoint(self, path=None):
+        self._join_reduce()
         from .checkpoint import save_checkpoint
         return save_checkpoint(self, path)
 
@@ -681,6 +707,21 @@ class Node:
             self.engine.bump_version()
             return
         if len(self.dp_ranks) > 1:
+            if self._reducer is not None:
+                # overlapped path: params are frozen until the next
+                # optimizer step (reduce_threshold is a multiple of
+                # update_frequency), where the hook joins the result
+                self._join_reduce()  # drain any straggler first
+                params = [p for p in self.model.parameters()]
+                tensors = params + _master_tensors(self.optimizer, params)
+                if self.average_optim and self.optimizer is not None:
+                    for st in self.optimizer.state.values():
+                        for k, v in st.items():
+                            if k != "master" and isinstance(v, torch.Tensor) \
+                                    and v.is_floating_point() and v.numel():
+                                tensors.append(v)
+                self._reducer.launch(tensors)
+                return
             average_parameters(self.model, self.comm.my_dp_group,
                                optimizer=self.optimizer)
             if self.average_optim and self.optimizer is not None:
@@ -865,6 +906,7 @@ class Node:
                 pass
 
     def _handle_weights_request(self, payload_src: int):
+        self._join_reduce()
         snap = self.engine.latest_state_snapshot()
         tensors = [(i, t) for i, t in enumerate(snap["params"])]
         self.comm.send(payload_src, "ctrl",
@@ -886,6 +928,7 @@ class _CommSessionFacade:
         if n.comm is None:
             return
         n.engine.join_recompute()
+        n.join_pending_reduce()
         if getattr(n.comm, "my_dp_segments", None):
             average_parameter_segments(
                 n.model, n.comm.my_dp_segments,

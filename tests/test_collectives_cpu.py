@@ -107,3 +107,45 @@ def test_average_parameters_covers_fp32_masters(tmp_path):
         ref = (o0 + o1) / 2
         assert torch.allclose(m0, ref, atol=1e-6), (m0 - ref).abs().max()
         assert torch.allclose(m1, ref, atol=1e-6)
+
+
+def _async_worker(rank, port, out_dir):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=2)
+    from ravnest_amd.comm.collectives import AsyncReducer
+    g = torch.Generator().manual_seed(40 + rank)
+    ts = [torch.randn(16, 16, generator=g),
+          torch.randn(257, generator=g)]
+    orig = [t.clone() for t in ts]
+    red = AsyncReducer(dist.group.WORLD, torch.device("cpu"))
+    red.launch(ts)
+    assert red.pending
+    # tensors untouched until join (overlap window)
+    assert red.join_into()
+    assert not red.pending
+    torch.save({"orig": orig, "avg": [t.clone() for t in ts]},
+               os.path.join(out_dir, f"ar_{rank}.pt"))
+    dist.barrier()
+    dist.destroy_process_group()
+
+
+def test_async_reducer_exact_mean(tmp_path):
+    """Overlapped DP averaging: launch -> (overlap window) -> join must
+    install the exact mean (same numerics as the synchronous path)."""
+    port = 29820 + (os.getpid() % 30)
+    ctx = mp.get_context("spawn")
+    procs = [ctx.Process(target=_async_worker, args=(r, port, str(tmp_path)))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(timeout=120)
+    for p in procs:
+        assert p.exitcode == 0, p.exitcode
+    res = {r: torch.load(tmp_path / f"ar_{r}.pt") for r in range(2)}
+    for a0, a1, o0, o1 in zip(res[0]["avg"], res[1]["avg"],
+                              res[0]["orig"], res[1]["orig"]):
+        ref = (o0 + o1) / 2
+        assert torch.allclose(a0, ref, atol=1e-6)
+        assert torch.allclose(a1, ref, atol=1e-6)

@@ -66,7 +66,7 @@ class _ListLoader:
 
 
 def run_pp(args, rank, world, device, on_gpu, family, model, batches,
-           labels):
+           labels, share_gpu=False):
     """Async pipeline-parallel bench: ONE model split into `world` stages
     (1 cluster x N stages — the reference's headline topology). Rank 0
     plans with clusterize() and drives microbatch injection; every stage
@@ -101,6 +101,7 @@ def run_pp(args, rank, world, device, on_gpu, family, model, batches,
                 optimizer=FusedAdam if on_gpu else torch.optim.Adam,
                 optimizer_params={"lr": 1e-4},
                 device=device,
+                backend="gloo" if share_gpu else None,
                 criterion=CrossEntropyLoss(-100),
                 labels=_ListLoader([labels[i % len(labels)] for i in
                                     range(args.warmup + args.steps +
@@ -133,7 +134,7 @@ def run_pp(args, rank, world, device, on_gpu, family, model, batches,
     dist.barrier()
     elapsed = _time.perf_counter() - t0
     t = torch.tensor([elapsed], dtype=torch.float64,
-                     device=device if on_gpu else "cpu")
+                     device=device if on_gpu and not share_gpu else "cpu")
     dist.all_reduce(t, op=dist.ReduceOp.MAX)
     elapsed = float(t.item())
 
@@ -203,7 +204,12 @@ def main():
         args.micro_batch = min(args.micro_batch, 4)
         args.seq = min(args.seq, 64)
 
-    device = torch.device("cuda", local_rank) if on_gpu else torch.device("cpu")
+    # shared-GPU mode: more pipeline ranks than GPUs (e.g. a 2-stage
+    # async pipeline exercised on ONE MI355X) — ranks share cuda:0.. and
+    # the wire falls back to gloo with CPU staging (comm/p2p.py)
+    share_gpu = on_gpu and world > torch.cuda.device_count()
+    device = (torch.device("cuda", local_rank % torch.cuda.device_count())
+              if on_gpu else torch.device("cpu"))
     if on_gpu:
         torch.cuda.set_device(device)
     set_seed(1234 + 0)  # same init on every replica (DP semantics)
@@ -256,7 +262,7 @@ def main():
 
     if args.parallelism == "pp" and world > 1:
         run_pp(args, rank, world, device, on_gpu, family, model,
-               batches, labels)
+               batches, labels, share_gpu=share_gpu)
         return
 
     comm = None

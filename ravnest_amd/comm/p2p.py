@@ -64,6 +64,14 @@ class Channel:
         self.device = device
         self.my_rank = my_rank
         self.deliver = deliver
+        # non-RCCL wire (gloo) cannot carry device tensors: stage via
+        # CPU. This is the shared-GPU bring-up mode (N pipeline ranks on
+        # fewer GPUs than ranks, e.g. a 2-stage pipeline on one MI355X).
+        try:
+            self.wire_cpu = (device.type == "cuda"
+                             and dist.get_backend(group) != "nccl")
+        except Exception:
+            self.wire_cpu = False
         self._send_q: queue.Queue = queue.Queue(maxsize=max_queue)
         self._threads: list[threading.Thread] = []
         self._stop = threading.Event()
@@ -160,7 +168,19 @@ class Channel:
             [ACTION_CODES[msg.action], msg.fpid, len(msg.tensors), msg.extra],
             dtype=torch.int64)
         meta = self._encode_meta(msg)
-        if self.device.type == "cuda":
+        if self.device.type == "cuda" and self.wire_cpu:
+            # gloo wire: device tensors hop through host memory
+            if msg.ready_event is not None:
+                torch.cuda.current_stream(self.device).wait_event(
+                    msg.ready_event)
+            dist.send(head, self.dst, group=self.group)
+            dist.send(meta, self.dst, group=self.group)
+            for _, t in msg.tensors:
+                t = t.detach()
+                if t.is_cuda:
+                    t = t.cpu()
+                dist.send(t.contiguous(), self.dst, group=self.group)
+        elif self.device.type == "cuda":
             if msg.ready_event is not None:
                 self._stream.wait_event(msg.ready_event)
             with torch.cuda.stream(self._stream):
@@ -198,6 +218,10 @@ class Channel:
                 self.deliver(self, msg)
 
     def _recv_tensor(self, shape, dtype):
+        if self.device.type == "cuda" and self.wire_cpu:
+            t = torch.empty(shape, dtype=dtype, device="cpu")
+            dist.recv(t, self.src, group=self.group)
+            return t.to(self.device)
         if self.device.type == "cuda":
             # allocate on the DEFAULT stream (the consumer's home stream)
             # so downstream compute use needs no record_stream; the comm
@@ -212,7 +236,8 @@ class Channel:
 
     def _do_recv(self) -> Message | None:
         ctx = (torch.cuda.stream(self._stream)
-               if self.device.type == "cuda" else _nullctx())
+               if self.device.type == "cuda" and not self.wire_cpu
+               else _nullctx())
         with ctx:
             head = self._recv_tensor((_HEADER_LEN,), torch.int64)
             head_cpu = head.cpu()
@@ -233,7 +258,7 @@ class Channel:
                 t = self._recv_tensor(shape, dtype)
                 tensors.append((gid, t))
                 off += _META_PER_TENSOR
-            if self.device.type == "cuda":
+            if self.device.type == "cuda" and not self.wire_cpu:
                 self._stream.synchronize()
         return Message(action=action, fpid=fpid, tensors=tensors, extra=extra)
 

@@ -530,9 +530,17 @@ class Node:
                 elif torch.is_tensor(t):
                     t = t.to(self.device)
                     # decompress wire-cast activations to the template
-                    # dtype (parity: reference compute.py:160-163)
+                    # dtype (parity: reference compute.py:160-163) —
+                    # ONLY when compression actually converted them. The
+                    # templates record PLAN-time dtypes (fp32 trace); a
+                    # bf16-native stage must receive its neighbor's bf16
+                    # hidden states as-is, not upcast to the plan dtype
+                    # (that fed fp32 into bf16 Linears on the first real
+                    # GPU pipeline run).
                     want = src.get("dtype")
-                    if want and t.is_floating_point():
+                    if (self.wire_dtype is not None and want
+                            and t.is_floating_point()
+                            and t.dtype == self.wire_dtype):
                         wd = _dtype_from_str(want)
                         if wd.is_floating_point and t.dtype != wd:
                             t = t.to(wd)

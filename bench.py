@@ -169,7 +169,7 @@ def run_pp(args, rank, world, device, on_gpu, family, model, batches,
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
-    ap.add_argument("--steps", type=int, default=20)
+    ap.add_argument("--steps", type=int, default=100)
     ap.add_argument("--warmup", type=int, default=5)
     ap.add_argument("--model", default="bert-base")
     ap.add_argument("--micro-batch", type=int, default=128)

@@ -161,9 +161,16 @@ class AsyncReducer:
                 flats.append((ts, torch.cat(
                     [t.detach().reshape(-1) for t in ts])))
 
+            self._failed = False
+
             def run():
-                for _, flat in flats:
-                    _all_reduce_avg(flat, self.group, world)
+                try:
+                    for _, flat in flats:
+                        _all_reduce_avg(flat, self.group, world)
+                except Exception as e:  # peer died mid-collective
+                    self._failed = True
+                    print(f"[AsyncReducer] collective failed: {e}",
+                          flush=True)
 
             import threading
             self._thread = threading.Thread(target=run, daemon=True)
@@ -181,6 +188,13 @@ class AsyncReducer:
         elif self._thread is not None:
             self._thread.join()
             self._thread = None
+            if getattr(self, "_failed", False):
+                # peer died mid-collective: DISCARD the partial buffers
+                # (never install garbage); the health monitor suspends
+                # future averaging
+                self._pending = None
+                self._failed = False
+                return False
         for ts, flat in flats:
             off = 0
             for t in ts:

@@ -21,11 +21,12 @@ import time
 
 class HealthMonitor:
     def __init__(self, node, interval: float = 2.0, timeout: float = 10.0,
-                 on_peer_lost=None):
+                 on_peer_lost=None, on_peer_recovered=None):
         self.node = node
         self.interval = interval
         self.timeout = timeout
         self.on_peer_lost = on_peer_lost
+        self.on_peer_recovered = on_peer_recovered
         self.last_seen: dict[int, float] = {}
         self.rtt: dict[int, float] = {}
         self._ping_sent: dict[int, float] = {}
@@ -54,7 +55,10 @@ class HealthMonitor:
         self.last_seen[src] = now
         if src in self._ping_sent:
             self.rtt[src] = now - self._ping_sent[src]
-        self._lost.discard(src)
+        if src in self._lost:
+            self._lost.discard(src)
+            if self.on_peer_recovered is not None:
+                self.on_peer_recovered(src)
 
     def note_ping(self, src: int):
         self.last_seen[src] = time.monotonic()
@@ -70,8 +74,8 @@ class HealthMonitor:
                     self.node.comm.send(peer, "ctrl", Message(
                         action=ActionTypes.STOP, fpid=-2, tensors=[],
                         extra=3))  # extra=3: PING
-                except KeyError:
-                    pass
+                except Exception:
+                    pass  # a dead peer's channel may refuse the send
                 seen = self.last_seen.get(peer)
                 if seen is not None and now - seen > self.timeout and \
                         peer not in self._lost:

@@ -265,13 +265,35 @@ class Node:
     # aux subsystems (health, checkpoint) — SURVEY.md section 5 parity+
     # ==================================================================
     def start_health_monitor(self, interval: float = 2.0,
-                             timeout: float = 10.0, on_peer_lost=None):
+                             timeout: float = 10.0, on_peer_lost=None,
+                             on_peer_recovered=None):
+        """Heartbeat over the ctrl channels. Default recovery policy
+        (parity+: the reference only ships an unused Ping RPC): when a
+        DP replica goes silent, parameter averaging is DISABLED so
+        training continues un-averaged instead of hanging the next
+        collective on the dead rank; when the peer's heartbeat returns,
+        averaging re-enables (the rejoining rank pulls weights via
+        update_with_latest_weights — the elastic-join path)."""
         from .health import HealthMonitor
-        self.health_monitor = HealthMonitor(self, interval=interval,
-                                            timeout=timeout,
-                                            on_peer_lost=on_peer_lost)
+        self.health_monitor = HealthMonitor(
+            self, interval=interval, timeout=timeout,
+            on_peer_lost=on_peer_lost or self._on_dp_peer_lost,
+            on_peer_recovered=on_peer_recovered or self._on_dp_peer_back)
         self.health_monitor.start()
         return self.health_monitor
+
+    def _on_dp_peer_lost(self, peer: int):
+        if peer in self.dp_ranks:
+            self._dp_suspended = True
+            print(f"[rank {self.rank}] DP peer {peer} lost — parameter "
+                  "averaging suspended (training continues un-averaged)",
+                  flush=True)
+
+    def _on_dp_peer_back(self, peer: int):
+        if peer in self.dp_ranks and getattr(self, "_dp_suspended", False):
+            self._dp_suspended = False
+            print(f"[rank {self.rank}] DP peer {peer} recovered — "
+                  "parameter averaging resumed", flush=True)
 
     def health(self) -> dict:
         return (self.health_monitor.health()
@@ -700,6 +722,8 @@ class Node:
     def _maybe_reduce(self):
         if not self.reduce_threshold or self.comm is None:
             return
+        if getattr(self, "_dp_suspended", False):
+            return  # a DP replica is down: skip averaging, keep training
         if self.engine.n_backwards % self.reduce_threshold != 0:
             return
         # The speculative recompute thread swaps param.data to a
@@ -744,8 +768,9 @@ class Node:
             targets = next(self.labels)
             epoch_change = self.labels.take_wrapped()
             args = self._root_args(model_inputs)
-            _, stepped, _ = self.engine.find_loss(
+            _, stepped, loss = self.engine.find_loss(
                 fpid, args, [False] * len(args), targets)
+            self.losses.append(loss)
             self._step_scheduler(stepped, epoch_change)
             self._mark_backward_complete(fpid)
             self._maybe_reduce()

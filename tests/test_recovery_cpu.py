@@ -70,11 +70,13 @@ def _worker(rank, base_dir, port, out_dir, q):
         time.sleep(0.1)
 
     # phase 2: training continues (averaging suspended, must NOT hang)
-    for X, _y in loader[:8]:
+    for X, _y in loader[:4]:
         node.forward_compute(tensors=X)
     node.wait_for_backwards(timeout=60)
     q.put(("ok", node.engine.n_backwards,
            node.losses[-1] < node.losses[0] if node.losses else None))
+    q.close()
+    q.join_thread()  # flush before the hard exit
     node.stop()
     os._exit(0)
 
@@ -100,5 +102,5 @@ def test_dp_replica_death_training_continues(tmp_path):
     for p in procs:
         p.join(timeout=60)
     assert res[0] == "ok", res
-    assert res[1] == 12  # 4 + 8 microbatches completed on the survivor
+    assert res[1] == 8  # 4 + 4 microbatches completed on the survivor
     assert procs[0].exitcode == 0

@@ -76,6 +76,16 @@ def edges_from_plan(base_dir: str) -> list[Edge]:
                 if s < len(nxt["stage_ranks"]):
                     edges.add(Edge(r, nxt["stage_ranks"][s], "ctrl"))
                     edges.add(Edge(nxt["stage_ranks"][s], r, "ctrl"))
+    # unequal clusters: ctrl edges between every param-range segment's
+    # peers (the per-range latest-weights pull targets the MAPPED peer
+    # that owns the range in the other cluster — reference
+    # node.py:127-135 retrieve_latest_params_data)
+    for seg in plan.get("dp_segments", []):
+        ranks = seg["ranks"]
+        for a in ranks:
+            for b in ranks:
+                if a != b:
+                    edges.add(Edge(a, b, "ctrl"))
     return sorted(edges)
 
 

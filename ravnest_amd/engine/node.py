@@ -447,9 +447,12 @@ class Node:
             else:
                 self._enqueue(3, ("stop_cascade", None, None))
         elif msg.action == ActionTypes.PREDICTION and channel.kind == "ctrl":
-            # WEIGHTS protocol over ctrl: extra==1 request, extra==2 reply
+            # WEIGHTS protocol over ctrl: extra==1 request (optionally
+            # carrying a newline-joined param-NAME blob for a partial
+            # range), extra==2 reply
             if msg.extra == 1:
-                self._enqueue(1, ("weights_request", channel.src, None))
+                self._enqueue(1, ("weights_request", channel.src,
+                                  msg.tensors))
             else:
                 self._weights_reply_q.put(msg.tensors)
 
@@ -530,7 +533,8 @@ class Node:
         elif kind == "save_submodel":
             self._handle_save()
         elif kind == "weights_request":
-            self._handle_weights_request(payload_src=fpid)
+            self._handle_weights_request(payload_src=fpid,
+                                         req_tensors=payload)
         elif kind == "stop_cascade":
             self._forward_stop()
             self._stop.set()
@@ -697,9 +701,38 @@ class Node:
 
     def update_with_latest_weights(self, src_rank: int | None = None,
                                    timeout: float = 120.0):
-        """Pull the latest parameter snapshot from a DP peer over the ctrl
-        channel (parity: reference get_latest_weights,
-        communication.py:279-330 / node.py:726-730)."""
+        """Pull the latest parameters over the ctrl channels (parity:
+        reference get_latest_weights, communication.py:279-330 /
+        node.py:726-730). Equal clusters: full snapshot from one DP
+        peer. Unequal clusters: each param-range SEGMENT is pulled from
+        a mapped peer of that segment's ring (the reference's multi-peer
+        partial restore, node.py:127-135)."""
+        if self.comm is None:
+            return
+        segs = getattr(self.comm, "my_dp_segments", None)
+        if segs and src_rank is None:
+            named = dict(self.model.named_parameters())
+            pulled = 0
+            for ranks, _g, names in segs:
+                peers = [r for r in ranks if r != self.rank]
+                names = [n for n in names if n in named]
+                if not peers or not names:
+                    continue
+                blob = torch.tensor(list("\n".join(names).encode()),
+                                    dtype=torch.uint8)
+                self.comm.send(peers[0], "ctrl",
+                               Message(action=ActionTypes.PREDICTION,
+                                       fpid=0, tensors=[(0, blob)],
+                                       extra=1))
+                tensors = self._weights_reply_q.get(timeout=timeout)
+                with torch.no_grad():
+                    for i, t in tensors:
+                        p = named[names[i]]
+                        p.data.copy_(t.to(p.device, p.dtype))
+                        pulled += 1
+            if pulled:
+                self.engine.bump_version()
+            return
         if src_rank is None:
             peers = [r for r in self.dp_ranks if r != self.rank]
             if not peers:
@@ -925,10 +958,58 @@ class Node:
         else:  # prediction
             torch.save(out.detach().cpu(), "prediction.pt")
 
+    def _template_example_args(self) -> list:
+        """Zero-filled example inputs from the plan templates' recorded
+        shapes/dtypes (used for the traced submodel export)."""
+        args = []
+        for src in self.input_template:
+            if src.get("kind") == "const":
+                args.append(src.get("value"))
+            elif src.get("pyscalar"):
+                args.append(int(src.get("value", 1)))
+            else:
+                shape = src.get("shape") or [1]
+                dt = _dtype_from_str(src.get("dtype", "torch.float32"))
+                args.append(torch.zeros(*shape, dtype=dt,
+                                        device=self.device))
+        return args
+
     def _handle_save(self, cascade: bool = False):
         if self.template_path:
             p = Path(self.template_path)
             torch.save(self.model.state_dict(), p / "trained_state_dict.pt")
+            # self-contained TorchScript submodel (parity: reference
+            # node.py:719-722 saves jit-scripted submodels loadable with
+            # NO package; sorter_inference.py:5-21 chains them). Custom
+            # autograd ops usually defeat script(), so fall back to an
+            # eval-mode trace over the plan-template shapes.
+            import os as _os
+            tmp = str(p / "submod_script.pt.tmp")
+            final = str(p / "submod_script.pt")
+            try:
+                # script() can COMPILE but still fail at save() on
+                # custom-autograd ops — write to a temp and rename only
+                # on success so a partial archive never lands
+                torch.jit.script(self.model).save(tmp)
+                _os.replace(tmp, final)
+            except Exception:
+                try:
+                    was_training = self.model.training
+                    self.model.eval()
+                    with torch.no_grad():
+                        tr = torch.jit.trace(
+                            self.model,
+                            tuple(self._template_example_args()),
+                            check_trace=False)
+                    tr.save(tmp)
+                    _os.replace(tmp, final)
+                    if was_training:
+                        self.model.train()
+                except Exception as e:
+                    print(f"[rank {self.rank}] TorchScript export "
+                          f"skipped: {e}", flush=True)
+                    if _os.path.exists(tmp):
+                        _os.remove(tmp)
         if (cascade or self.node_type != NodeTypes.LEAF) and not self.fused \
                 and self.comm is not None and self.stage + 1 < self.n_stages:
             nxt = self.stage_ranks[self.stage + 1]
@@ -938,10 +1019,21 @@ class Node:
             except KeyError:
                 pass
 
-    def _handle_weights_request(self, payload_src: int):
+    def _handle_weights_request(self, payload_src: int, req_tensors=None):
         self._join_reduce()
-        snap = self.engine.latest_state_snapshot()
-        tensors = [(i, t) for i, t in enumerate(snap["params"])]
+        if req_tensors:
+            # partial pull: request names a contiguous param range
+            # (parity: reference get_latest_weights serves name ranges,
+            # communication.py:279-330 / endpoints.py:145-154)
+            blob = bytes(req_tensors[0][1].to(torch.uint8).tolist())
+            names = blob.decode().split("\n")
+            self.engine.join_recompute()
+            named = dict(self.model.named_parameters())
+            tensors = [(i, named[n].detach().clone())
+                       for i, n in enumerate(names) if n in named]
+        else:
+            snap = self.engine.latest_state_snapshot()
+            tensors = [(i, t) for i, t in enumerate(snap["params"])]
         self.comm.send(payload_src, "ctrl",
                        Message(action=ActionTypes.PREDICTION, fpid=0,
                                tensors=tensors, extra=2))

@@ -119,3 +119,61 @@ def test_unequal_cluster_dp(tmp_path):
     # two leaves interleave their losses; compare epoch aggregates
     assert len(losses) == 2 * 4 * n_batches
     assert sum(losses[-2 * n_batches:]) < sum(losses[:2 * n_batches])
+
+
+def _pull_worker(rank, base_dir, port, out_dir, q):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    os.chdir(out_dir)
+    set_seed(42)
+    import time
+    from ravnest_amd import Node
+    node = Node(name=f"node_{rank}", base_dir=base_dir,
+                optimizer=torch.optim.Adam,
+                device=torch.device("cpu"),
+                criterion=_loss_fn, labels=_make_loader())
+    node.start()
+    if rank == 0:
+        time.sleep(1.5)  # let every serving peer come up
+        before = {n: p.detach().clone()
+                  for n, p in node.model.named_parameters()}
+        with torch.no_grad():
+            for p in node.model.parameters():
+                p.add_(1.0)  # corrupt the local replica
+        node.update_with_latest_weights()  # per-SEGMENT pull from peers
+        diff = max((p.detach() - before[n]).abs().max().item()
+                   for n, p in node.model.named_parameters())
+        q.put(("ok", diff))
+        q.close()
+        q.join_thread()
+        os._exit(0)
+    # serving peers park until the parent terminates them
+    time.sleep(60)
+    os._exit(0)
+
+
+def test_per_range_latest_weights_pull(tmp_path):
+    """Unequal clusters: update_with_latest_weights pulls each
+    param-range SEGMENT from its mapped peer in the other cluster (the
+    reference's multi-peer partial restore, node.py:127-135 /
+    communication.py:279-330) — a corrupted replica is fully restored."""
+    set_seed(42)
+    model = CNN()
+    base = str(tmp_path / "node_data")
+    pool = [NodeSpec(name=f"n{i}", ram=100 * 2**20) for i in range(WORLD)]
+    clusterize(model, (torch.randn(2, 1, 8, 8),), node_pool=pool,
+               cluster_assignment=[[0, 1], [2, 3, 4]], base_dir=base)
+    port = 30420 + (os.getpid() % 60)
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_pull_worker,
+                         args=(r, base, port, str(tmp_path), q))
+             for r in range(WORLD)]
+    for p in procs:
+        p.start()
+    res = q.get(timeout=150)
+    for p in procs:
+        p.terminate()
+        p.join(timeout=30)
+    assert res[0] == "ok"
+    assert res[1] < 1e-6, f"pull did not restore params (max diff {res[1]})"

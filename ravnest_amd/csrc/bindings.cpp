@@ -34,6 +34,7 @@ std::vector<at::Tensor> mx_quant(at::Tensor x);
 std::vector<at::Tensor> gemm_nt_bf16(at::Tensor A2, at::Tensor B,
                                      c10::optional<at::Tensor> bias,
                                      int64_t epi);
+at::Tensor gemm_wgrad_bf16(at::Tensor dy, at::Tensor x);
 at::Tensor mx_gemm(at::Tensor x, at::Tensor xs, at::Tensor w,
                    at::Tensor ws);
 at::Tensor mx_scale_probe(at::Tensor a, at::Tensor b, at::Tensor sa,
@@ -123,6 +124,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "bf16 MFMA GEMM x @ W^T (256^2 8-phase, fused bias/GELU epilogue)",
         py::arg("a"), py::arg("b"), py::arg("bias") = py::none(),
         py::arg("epi") = 0);
+  m.def("gemm_wgrad_bf16", &gemm_wgrad_bf16,
+        "split-K wgrad dW = dy^T @ x (tr16 transpose reads, fp32 atomics)");
   m.def("mx_quant", &mx_quant,
         "bf16 -> MX fp8 (e4m3 + per-32 e8m0 scales)");
   m.def("mx_gemm", &mx_gemm, "MX fp8 GEMM: x @ W^T, 32x32x64 scaled MFMA");

@@ -39,6 +39,19 @@ def _hand_ok(x: torch.Tensor, w: torch.Tensor) -> bool:
             and x.numel() * 2 < 2**32 and w.numel() * 2 < 2**32)
 
 
+def _hand_wgrad(dy2: torch.Tensor, x2: torch.Tensor) -> torch.Tensor:
+    """dW = dy^T @ x via the split-K tr16 kernel where it measured
+    faster than the library (profiles/r02_gemm_vs_blaslt.txt: wins the
+    N*K <= 2304*768 shapes, loses the wide ones)."""
+    M, N = dy2.shape
+    K = x2.shape[1]
+    if (M % 64 == 0 and M >= 128 and N >= 8 and K >= 8
+            and N * K <= 2304 * 768):
+        ext = get_ext(required=True)
+        return ext.gemm_wgrad_bf16(dy2, x2).to(dy2.dtype)
+    return dy2.t() @ x2
+
+
 def _hand_dgrad(dy2: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
     """dx = dy @ W: reduction over N -> needs W^T as the kernel's
     [rows][k] operand; falls back to the library when N is ragged."""
@@ -70,7 +83,7 @@ class _HandLinearFn(torch.autograd.Function):
         x2, w = ctx.saved_tensors
         dy2 = dy.reshape(-1, dy.shape[-1]).contiguous()
         dx = _hand_dgrad(dy2, w).reshape(ctx.x_shape)
-        dw = dy2.t() @ x2
+        dw = _hand_wgrad(dy2, x2)
         db = None
         if ctx.has_bias:
             ext = get_ext(required=True)

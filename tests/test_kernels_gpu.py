@@ -732,3 +732,24 @@ def test_hand_linear_autograd(dev, monkeypatch):
         ok(x.grad, xf.grad, 4e-2)
         ok(lin.weight.grad, wf.grad, 4e-2)
         ok(lin.bias.grad, bf.grad, 4e-2)
+
+
+@pytest.mark.parametrize("shape", [
+    (512, 768, 768),
+    (2048, 2304, 768),
+    (256, 298, 136),    # ragged N (not %8) and K
+    (128, 130, 72),     # tails everywhere
+])
+def test_gemm_wgrad_bf16(dev, shape):
+    """Split-K wgrad (tr16 transpose reads + fp32 atomics) vs fp32 torch."""
+    from ravnest_amd.ops import get_ext
+    ext = get_ext(True)
+    M, N, K = shape
+    torch.manual_seed(0)
+    dy = (torch.randn(M, N, device=dev) / math.sqrt(M)).to(torch.bfloat16)
+    x = torch.randn(M, K, device=dev).to(torch.bfloat16)
+    ref = dy.float().t() @ x.float()
+    c = ext.gemm_wgrad_bf16(dy, x)
+    err = (c - ref).abs().max().item()
+    scale = ref.abs().max().item() + 1e-6
+    assert err / scale < 3e-2, f"rel err {err/scale} (abs {err})"

@@ -73,5 +73,42 @@ def main():
         del a, b, bt
 
 
+def bench_wgrad(iters=10):
+    """wgrad dW[N,K] = dy[M,N]^T @ x[M,K]: library vs transpose+hand-NT
+    (reduction M=65536 becomes the hand kernel's deep-K dimension)."""
+    from ravnest_amd.ops import get_ext
+    ext = get_ext(True)
+    dev = torch.device("cuda", 0)
+    torch.manual_seed(0)
+    M = 65536
+    for (N, K, label) in [(2304, 768, "qkv_w"), (768, 768, "attn_out_w"),
+                          (3072, 768, "mlp_in_w"), (768, 3072, "mlp_out_w"),
+                          (30522, 768, "mlm_head_w")]:
+        dy = (torch.randn(M, N, device=dev) / 64).to(torch.bfloat16)
+        x = (torch.randn(M, K, device=dev) / 64).to(torch.bfloat16)
+        flops = 2.0 * M * K * N
+
+        def lib():
+            return dy.t() @ x
+
+        def hand():
+            return ext.gemm_wgrad_bf16(dy, x).to(torch.bfloat16)
+
+        t_lib = time_fn(lib, iters)
+        t_hand = time_fn(hand, iters)
+        c = hand().float()
+        cref = (dy.t().contiguous().float() @ x.float())
+        err = (c - cref).abs().max().item() / (cref.abs().max().item() + 1e-9)
+        print(f"{label:12s} N{N} K{K} M{M}: hand {t_hand:7.3f} ms "
+              f"({flops/t_hand/1e9:7.1f} TF) | lib {t_lib:7.3f} ms "
+              f"({flops/t_lib/1e9:7.1f} TF) | ratio {t_lib/t_hand:5.2f}x "
+              f"| relerr {err:.3e}", flush=True)
+        del dy, x
+
+
 if __name__ == "__main__":
-    main()
+    if "--wgrad" in sys.argv:
+        sys.argv.remove("--wgrad")
+        bench_wgrad()
+    else:
+        main()

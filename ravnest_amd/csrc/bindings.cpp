@@ -37,6 +37,8 @@ std::vector<at::Tensor> gemm_nt_bf16(at::Tensor A2, at::Tensor B,
 at::Tensor gemm_wgrad_bf16(at::Tensor dy, at::Tensor x);
 at::Tensor mx_gemm(at::Tensor x, at::Tensor xs, at::Tensor w,
                    at::Tensor ws);
+at::Tensor mx_gemm2(at::Tensor xq, at::Tensor xs, at::Tensor wq,
+                    at::Tensor ws);
 at::Tensor mx_scale_probe(at::Tensor a, at::Tensor b, at::Tensor sa,
                           at::Tensor sb);
 at::Tensor softmax_fwd(at::Tensor x);
@@ -129,6 +131,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("mx_quant", &mx_quant,
         "bf16 -> MX fp8 (e4m3 + per-32 e8m0 scales)");
   m.def("mx_gemm", &mx_gemm, "MX fp8 GEMM: x @ W^T, 32x32x64 scaled MFMA");
+  m.def("mx_gemm2", &mx_gemm2,
+        "LDS-staged MX fp8 GEMM (256^2 4-phase, scaled 32x32x64 MFMA)");
   m.def("mx_scale_probe", &mx_scale_probe,
         "per-lane scale-byte semantics probe");
   m.def("softmax_fwd", &softmax_fwd, "standalone softmax fwd (last dim)");

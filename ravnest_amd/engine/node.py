@@ -241,7 +241,10 @@ class Node:
         # the next microbatch's compute instead of blocking the pipeline
         # (SURVEY.md §7 "Overlap engineering"; the reference blocks).
         self._reducer = None
-        if (async_reduce and self.comm is not None
+        if (async_reduce
+                and __import__("os").environ.get(
+                    "RAVNEST_ASYNC_REDUCE", "1") == "1"
+                and self.comm is not None
                 and len(self.dp_ranks) > 1
                 and not getattr(self.comm, "my_dp_segments", None)):
             self._reducer = AsyncReducer(self.comm.my_dp_group, self.device)

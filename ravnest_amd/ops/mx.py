@@ -24,7 +24,14 @@ class _MXLinearFn(torch.autograd.Function):
         x2 = x.reshape(-1, shp[-1]).contiguous()
         xq, xs = ext.mx_quant(x2)
         wq, ws = ext.mx_quant(weight.contiguous())
-        y = ext.mx_gemm(xq, xs, wq, ws)
+        K = x2.shape[1]
+        if K % 128 == 0 and K >= 256 and xq.numel() < 2**32 \
+                and wq.numel() < 2**32:
+            # LDS-staged 4-phase kernel (csrc/mx_gemm2.hip): 845-1294 TF
+            # vs the register-tiled mx_gemm's ~530
+            y = ext.mx_gemm2(xq, xs, wq, ws)
+        else:
+            y = ext.mx_gemm(xq, xs, wq, ws)
         if bias is not None:
             y = y + bias
         ctx.save_for_backward(x2, weight)

@@ -753,3 +753,26 @@ def test_gemm_wgrad_bf16(dev, shape):
     err = (c - ref).abs().max().item()
     scale = ref.abs().max().item() + 1e-6
     assert err / scale < 3e-2, f"rel err {err/scale} (abs {err})"
+
+
+@pytest.mark.parametrize("shape", [(512, 768, 768), (512, 2304, 768),
+                                   (512, 300, 768)])
+def test_mx_gemm2(dev, shape):
+    """LDS-staged MX fp8 GEMM vs the register-tiled mx_gemm AND the
+    dequantized fp32 reference."""
+    from ravnest_amd.ops import get_ext
+    ext = get_ext(True)
+    M, N, K = shape
+    torch.manual_seed(0)
+    x = (torch.randn(M, K, device=dev) / 8).to(torch.bfloat16)
+    w = (torch.randn(N, K, device=dev) / 8).to(torch.bfloat16)
+    xq, xs = ext.mx_quant(x)
+    wq, ws = ext.mx_quant(w)
+    y2 = ext.mx_gemm2(xq, xs, wq, ws)
+    y1 = ext.mx_gemm(xq, xs, wq, ws)
+    d12 = (y2.float() - y1.float()).abs().max().item()
+    assert d12 / (y1.float().abs().max().item() + 1e-6) < 1e-2, d12
+    ref = x.float() @ w.float().t()
+    e = (y2.float() - ref).abs().max().item() / \
+        (ref.abs().max().item() + 1e-6)
+    assert e < 8e-2, e  # fp8 quantization error class

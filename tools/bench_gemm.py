@@ -106,9 +106,38 @@ def bench_wgrad(iters=10):
         del dy, x
 
 
+
+
+
+def bench_mx(iters=20):
+    from ravnest_amd.ops import get_ext
+    ext = get_ext(True)
+    dev = torch.device("cuda", 0)
+    torch.manual_seed(0)
+    for (M, K, N, label) in [(32768, 768, 2304, "gpt_qkv"),
+                             (32768, 768, 3072, "gpt_mlp_in"),
+                             (32768, 3072, 768, "gpt_mlp_out"),
+                             (4096, 4096, 4096, "square4k"),
+                             (8192, 8192, 8192, "square8k")]:
+        x = (torch.randn(M, K, device=dev) / 8).to(torch.bfloat16)
+        w = (torch.randn(N, K, device=dev) / 8).to(torch.bfloat16)
+        xq, xs = ext.mx_quant(x)
+        wq, ws = ext.mx_quant(w)
+        flops = 2.0 * M * K * N
+        t1 = time_fn(lambda: ext.mx_gemm(xq, xs, wq, ws), iters)
+        t2 = time_fn(lambda: ext.mx_gemm2(xq, xs, wq, ws), iters)
+        tb = time_fn(lambda: torch.matmul(x, w.t()), iters)
+        print(f"{label:11s} M{M} K{K} N{N}: mx2 {t2:7.3f} ms "
+              f"({flops/t2/1e9:7.1f} TF) | mx1 {t1:7.3f} "
+              f"({flops/t1/1e9:7.1f}) | bf16-lib {tb:7.3f} "
+              f"({flops/tb/1e9:7.1f})", flush=True)
+        del x, w, xq, wq
+
+
 if __name__ == "__main__":
     if "--wgrad" in sys.argv:
-        sys.argv.remove("--wgrad")
         bench_wgrad()
+    elif "--mx" in sys.argv:
+        bench_mx()
     else:
         main()

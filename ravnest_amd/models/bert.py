@@ -73,12 +73,23 @@ class BertEmbeddings(nn.Module):
 
 
 class BertSelfAttention(nn.Module):
+    """DOCUMENTED SEMANTICS DIFFERENCE vs HuggingFace BERT: by default
+    dropout is applied AFTER the output projection (self.drop below)
+    instead of on the attention probabilities — the fused flash kernel
+    never materializes the S x S probs. Set RAVNEST_EXACT_ATTN_DROPOUT=1
+    (before model construction) for exact HF prob-dropout semantics via
+    the composed P-materializing path (replayable philox mask; lower
+    throughput — ops/attention.py attention_qkv_prob_dropout)."""
+
     def __init__(self, cfg: BertConfig):
         super().__init__()
+        import os
         self.heads = cfg.heads
         self.head_dim = cfg.hidden // cfg.heads
         self.qkv = make_linear(cfg.hidden, 3 * cfg.hidden)
-        self.core = AttentionCoreQKV(causal=False)
+        exact = os.environ.get("RAVNEST_EXACT_ATTN_DROPOUT", "0") == "1"
+        self.core = AttentionCoreQKV(
+            causal=False, prob_dropout=cfg.dropout if exact else 0.0)
         self.out = make_linear(cfg.hidden, cfg.hidden)
         self.drop = Dropout(cfg.dropout)
 

@@ -143,12 +143,57 @@ def attention_qkv(qkv, mask=None, causal=False, scale=None):
     return o.transpose(1, 2).flatten(2)
 
 
+def attention_qkv_prob_dropout(qkv, mask, causal, scale, p, training):
+    """Attention with dropout ON THE PROBABILITIES — exact HuggingFace
+    BertSelfAttention semantics. Composed path: the S x S probs are
+    materialized and masked by the replayable philox dropout kernel
+    (csrc/dropout.hip), so the versioned-recompute engine replays the
+    identical mask. Slower than the fused flash kernel (which cannot
+    drop individual probs without a P-materialization); use when exact
+    prob-dropout semantics matter more than throughput."""
+    from .dropout import dropout as _phil_dropout
+    q, k, v = qkv.unbind(dim=2)          # (B, S, H, D) each
+    q = q.transpose(1, 2)                # (B, H, S, D)
+    k = k.transpose(1, 2)
+    v = v.transpose(1, 2)
+    if scale is None:
+        scale = 1.0 / math.sqrt(q.shape[-1])
+    s = (q @ k.transpose(-2, -1)) * scale
+    if causal:
+        S = q.shape[-2]
+        cm = torch.triu(torch.full((S, S), float("-inf"),
+                                   device=q.device), diagonal=1)
+        s = s + cm
+    if mask is not None:
+        s = s + mask
+    probs = torch.softmax(s.float(), dim=-1).to(qkv.dtype)
+    if training and p > 0:
+        probs = _phil_dropout(probs, p, training=True)
+    o = probs @ v                        # (B, H, S, D)
+    return o.transpose(1, 2).contiguous()  # token-major like the kernel
+
+
 class AttentionCoreQKV(nn.Module):
+    """Fused flash-style attention over packed (B,S,3,H,D) qkv.
+
+    prob_dropout: dropout rate applied to the ATTENTION PROBABILITIES
+    (HF BertSelfAttention semantics). The fused flash kernel never
+    materializes the probs, so a prob_dropout > 0 routes through the
+    composed P-materializing path above (exact semantics, replayable
+    philox mask, lower throughput). With prob_dropout == 0 (the perf
+    default used by the bench models, which apply dropout AFTER the
+    output projection instead — a DOCUMENTED semantics difference vs
+    HF BERT) the fused kernel runs."""
     _is_leaf_module = True
 
-    def __init__(self, causal: bool = False):
+    def __init__(self, causal: bool = False, prob_dropout: float = 0.0):
         super().__init__()
         self.causal = causal
+        self.prob_dropout = prob_dropout
 
     def forward(self, qkv, mask=None):
+        if self.prob_dropout > 0 and self.training and qkv.is_cuda:
+            return attention_qkv_prob_dropout(
+                qkv, mask, self.causal, None, self.prob_dropout,
+                self.training)
         return attention_qkv(qkv, mask=mask, causal=self.causal)

@@ -170,7 +170,10 @@ def attention_qkv_prob_dropout(qkv, mask, causal, scale, p, training):
     if training and p > 0:
         probs = _phil_dropout(probs, p, training=True)
     o = probs @ v                        # (B, H, S, D)
-    return o.transpose(1, 2).contiguous()  # token-major like the kernel
+    # token-major (B, S, H*D) like the fused kernel (the next Linear's
+    # input layout)
+    B, Hh, S, D = o.shape
+    return o.transpose(1, 2).reshape(B, S, Hh * D)
 
 
 class AttentionCoreQKV(nn.Module):

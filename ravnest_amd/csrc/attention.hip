@@ -239,14 +239,26 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
 // G4/T2: the b128 16-lane groups read rows distinct mod 16), V stored
 // linear and consumed as 2B LDS reads (16 consecutive banks per group).
 // ---------------------------------------------------------------------
-template <bool TR16>  // V B-fragment reads: tr16 hardware-transpose
-                      // gather vs eight 2-B reads (A/B, env-selected)
+template <bool TR16, bool DROP = false>
+// TR16: V B-fragment reads via tr16 hardware-transpose gather vs eight
+// 2-B reads (A/B, env-selected). DROP: attention-PROB dropout fused in
+// (HF BertSelfAttention semantics): the per-(qrow,key) philox keep mask
+// scales the UNNORMALIZED exp values fed to PV while l_run keeps the
+// undropped softmax normalizer (so O = dropout(P) @ V and lse is the
+// true logsumexp). Counter scheme shared with the bwd kernels: one
+// philox gen per 2x2 (qrow,key) square — ctr = (bh<<32 | qrow>>1,
+// key>>1), output index (qrow&1)*2 + (key&1); keep iff u >= p (same
+// convention as csrc/dropout.hip), seed XORed with the device graph
+// counter at run time.
 __global__ __launch_bounds__(256, 4) void attn_fwd_lds_kernel(
     const bf16_t* __restrict__ q, const bf16_t* __restrict__ k,
     const bf16_t* __restrict__ v, const float* __restrict__ mask,
     bf16_t* __restrict__ o, float* __restrict__ lse, int S, int causal,
     float scale, int has_mask, long H,
-    Strides sq, Strides sk, Strides sv, Strides so) {
+    Strides sq, Strides sk, Strides sv, Strides so,
+    float pdrop = 0.f, float inv_keep = 1.f,
+    unsigned long long dseed = 0,
+    const long long* __restrict__ seed_buf = nullptr) {
   constexpr int D = 64;
   // per slot: K swizzled 4KB + V linear 4KB (row stride 128B) or, for
   // the tr16 path, V padded to 144B rows (4608B) so the 4-lanes-per-row
@@ -377,6 +389,27 @@ __global__ __launch_bounds__(256, 4) void attn_fwd_lds_kernel(
       }
       psum += __shfl_xor(psum, 32, WAVE);
       l_run += psum;
+
+      if constexpr (DROP) {
+        const Philox4 ph(seed_buf
+                             ? dseed ^ (unsigned long long)*seed_buf
+                             : dseed);
+        const long bh_ctr = ((unsigned long long)bh << 32) |
+                            (unsigned)(qrow >> 1);
+        const int qb = (qrow & 1) << 1;
+#pragma unroll
+        for (int rq = 0; rq < 8; ++rq) {
+          const int r0 = 2 * rq;
+          const int key0 = k0 + (r0 & 3) + 8 * (r0 >> 2) + 4 * hi;  // even
+          unsigned int rr[4];
+          ph.gen((unsigned long long)bh_ctr,
+                 (unsigned long long)(key0 >> 1), rr);
+          const float u0 = (rr[qb] >> 8) * (1.0f / 16777216.0f);
+          const float u1 = (rr[qb | 1] >> 8) * (1.0f / 16777216.0f);
+          s_acc[r0] *= (u0 >= pdrop) ? inv_keep : 0.f;
+          s_acc[r0 + 1] *= (u1 >= pdrop) ? inv_keep : 0.f;
+        }
+      }
 
       unsigned int pa[2][4];
 #pragma unroll
@@ -560,7 +593,9 @@ std::vector<at::Tensor> attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v,
 // packed layout: qkv (B,S,3,H,D) contiguous -> o (B,S,H,D); no
 // permute/contiguous copies around the attention core.
 std::vector<at::Tensor> attn_fwd_qkv(at::Tensor qkv, at::Tensor mask,
-                                     bool causal, double scale) {
+                                     bool causal, double scale,
+                                     double pdrop, int64_t dseed,
+                                     c10::optional<at::Tensor> seed_buf) {
   TORCH_CHECK(qkv.is_cuda() && qkv.is_contiguous());
   TORCH_CHECK(qkv.dim() == 5 && qkv.size(2) == 3,
               "qkv must be (B,S,3,H,D)");
@@ -588,7 +623,20 @@ std::vector<at::Tensor> attn_fwd_qkv(at::Tensor qkv, at::Tensor mask,
   const Strides so{(long)S * HD, (long)D, HD};
   const bf16_t* base = reinterpret_cast<const bf16_t*>(qkv.data_ptr());
 
-  if (D == 64) {
+  if (pdrop > 0.0) {
+    TORCH_CHECK(D == 64 && use_lds_fwd && use_tr16,
+                "attn_fwd_qkv: fused prob-dropout needs the D=64 LDS+tr16 "
+                "path");
+    const long long* sb = nullptr;
+    if (seed_buf.has_value()) sb = reinterpret_cast<const long long*>(seed_buf->data_ptr<int64_t>());
+    hipLaunchKernelGGL((attn_fwd_lds_kernel<true, true>), grid, block, 0,
+                       stream, base, base + HD, base + 2 * HD, mask_ptr,
+                       reinterpret_cast<bf16_t*>(o.data_ptr()),
+                       lse.data_ptr<float>(), S, causal ? 1 : 0,
+                       (float)scale, has_mask ? 1 : 0, (long)H, sp, sp, sp,
+                       so, (float)pdrop, 1.f / (1.f - (float)pdrop),
+                       (unsigned long long)dseed, sb);
+  } else if (D == 64) {
     LAUNCH_ATTN_FWD(64, base, base + HD, base + 2 * HD,
                     reinterpret_cast<bf16_t*>(o.data_ptr()), sp, sp, sp, so);
   } else {

@@ -33,6 +33,48 @@ typedef float f32x16 __attribute__((ext_vector_type(16)));
 
 namespace {
 
+// fused attention-prob dropout (matches the fwd kernel's scheme): one
+// philox gen per 2x2 (qrow,key) square, ctr = (bh<<32 | qrow>>1,
+// key>>1), output index (qrow&1)*2 + (key&1), keep iff u >= p. The two
+// mask-bit builders cover the two lane layouts: key fixed per lane
+// (dv/dk kernels) and qrow fixed per lane (dq kernel).
+DEVINL unsigned drop_bits_keyfixed(const Philox4& ph, long bh, int q0,
+                                   int hi, int key, float p) {
+  unsigned bits = 0;
+  const int kb = key & 1;
+#pragma unroll
+  for (int rq = 0; rq < 8; ++rq) {
+    const int r0 = 2 * rq;
+    const int qrow0 = q0 + (r0 & 3) + 8 * (r0 >> 2) + 4 * hi;  // even
+    unsigned int rr[4];
+    ph.gen(((unsigned long long)bh << 32) | (unsigned)(qrow0 >> 1),
+           (unsigned long long)(key >> 1), rr);
+    if ((rr[kb] >> 8) * (1.0f / 16777216.0f) >= p) bits |= 1u << r0;
+    if ((rr[2 | kb] >> 8) * (1.0f / 16777216.0f) >= p)
+      bits |= 1u << (r0 + 1);
+  }
+  return bits;
+}
+
+DEVINL unsigned drop_bits_rowfixed(const Philox4& ph, long bh, int k0,
+                                   int hi, int qrow, float p) {
+  unsigned bits = 0;
+  const int qb = (qrow & 1) << 1;
+#pragma unroll
+  for (int rq = 0; rq < 8; ++rq) {
+    const int r0 = 2 * rq;
+    const int key0 = k0 + (r0 & 3) + 8 * (r0 >> 2) + 4 * hi;  // even
+    unsigned int rr[4];
+    ph.gen(((unsigned long long)bh << 32) | (unsigned)(qrow >> 1),
+           (unsigned long long)(key0 >> 1), rr);
+    if ((rr[qb] >> 8) * (1.0f / 16777216.0f) >= p) bits |= 1u << r0;
+    if ((rr[qb | 1] >> 8) * (1.0f / 16777216.0f) >= p)
+      bits |= 1u << (r0 + 1);
+  }
+  return bits;
+}
+
+
 // element offset of (b, h, row) = b*bs + h*hs + row*rs (see attention.hip)
 struct StridesB { long bs, hs, rs; };
 
@@ -110,12 +152,15 @@ __global__ void attn_bwd_delta_kernel(const bf16_t* __restrict__ dout,
 // the combined dkv kernel sat at 180 VGPR = 2 waves/SIMD and 56% of wave
 // cycles parked on memory waits — PMC profile r01)
 // -------------------------------------------------------------------
+template <bool DROP = false>
 __global__ __launch_bounds__(256, 4) void attn_bwd_dv_kernel(
     const bf16_t* __restrict__ q, const bf16_t* __restrict__ k,
     const bf16_t* __restrict__ dout, const float* __restrict__ lse,
     const float* __restrict__ mask, bf16_t* __restrict__ dv, int S,
     int causal, float scale, int has_mask, long H, StridesB sio,
-    StridesB sdo, StridesB sg) {
+    StridesB sdo, StridesB sg, float pdrop = 0.f, float inv_keep = 1.f,
+    unsigned long long dseed = 0,
+    const long long* __restrict__ seed_buf = nullptr) {
   constexpr int D = 64;
   // block-shared tiles: dO (linear: 2B B-operand reads) and Q
   // (((row&15)<<4)-swizzled: conflict-free b128 A-fragment reads).
@@ -194,6 +239,12 @@ __global__ __launch_bounds__(256, 4) void attn_bwd_dv_kernel(
       s_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qf, kf[s], s_acc,
                                                       0, 0, 0);
     }
+    unsigned dbits = 0;
+    if constexpr (DROP) {
+      const Philox4 ph(seed_buf ? dseed ^ (unsigned long long)*seed_buf
+                                : dseed);
+      dbits = drop_bits_keyfixed(ph, bh, q0, hi, key, pdrop);
+    }
 #pragma unroll
     for (int r = 0; r < 16; ++r) {
       const int qrow = q0 + (r & 3) + 8 * (r >> 2) + 4 * hi;
@@ -203,6 +254,8 @@ __global__ __launch_bounds__(256, 4) void attn_bwd_dv_kernel(
       bool dead = (qrow >= S) || (key >= S) || (causal && key > qrow) ||
                   !isfinite(l);
       s_acc[r] = dead ? 0.f : __expf(sv - l);  // P
+      if constexpr (DROP)  // dV uses the DROPPED probabilities
+        s_acc[r] *= ((dbits >> r) & 1) ? inv_keep : 0.f;
     }
     unsigned int pa_p[2][4];
     acc_to_afrag(s_acc, pa_p);
@@ -245,13 +298,16 @@ __global__ __launch_bounds__(256, 4) void attn_bwd_dv_kernel(
 // -------------------------------------------------------------------
 // dk kernel: wave owns keys; S, dP -> dS -> dK (~150 VGPR, 3 waves/SIMD)
 // -------------------------------------------------------------------
+template <bool DROP = false>
 __global__ __launch_bounds__(256, 3) void attn_bwd_dk_kernel(
     const bf16_t* __restrict__ q, const bf16_t* __restrict__ k,
     const bf16_t* __restrict__ v, const bf16_t* __restrict__ dout,
     const float* __restrict__ lse, const float* __restrict__ delta,
     const float* __restrict__ mask, bf16_t* __restrict__ dk, int S,
     int causal, float scale, int has_mask, long H, StridesB sio,
-    StridesB sdo, StridesB sg) {
+    StridesB sdo, StridesB sg, float pdrop = 0.f, float inv_keep = 1.f,
+    unsigned long long dseed = 0,
+    const long long* __restrict__ seed_buf = nullptr) {
   constexpr int D = 64;
   // Q tile swizzled (b128 A-frags for S, XOR-adjusted u16 B-reads for
   // dK) + dO tile swizzled (b128 A-frags for dP); double-buffered.
@@ -343,6 +399,12 @@ __global__ __launch_bounds__(256, 3) void attn_bwd_dk_kernel(
       dp_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dof, vf[s], dp_acc,
                                                        0, 0, 0);
     }
+    unsigned dbits = 0;
+    if constexpr (DROP) {
+      const Philox4 ph(seed_buf ? dseed ^ (unsigned long long)*seed_buf
+                                : dseed);
+      dbits = drop_bits_keyfixed(ph, bh, q0, hi, key, pdrop);
+    }
 #pragma unroll
     for (int r = 0; r < 16; ++r) {
       const int qrow = q0 + (r & 3) + 8 * (r >> 2) + 4 * hi;
@@ -353,7 +415,10 @@ __global__ __launch_bounds__(256, 3) void attn_bwd_dk_kernel(
       bool dead = (qrow >= S) || (key >= S) || (causal && key > qrow) ||
                   !isfinite(l);
       const float p = dead ? 0.f : __expf(sv - l);
-      dp_acc[r] = dead ? 0.f : p * (dp_acc[r] - dlt) * scale;
+      float dpr = dp_acc[r];
+      if constexpr (DROP)  // dP = mask * d(P_drop); delta is unchanged
+        dpr *= ((dbits >> r) & 1) ? inv_keep : 0.f;
+      dp_acc[r] = dead ? 0.f : p * (dpr - dlt) * scale;
     }
     unsigned int pa_ds[2][4];
     acc_to_afrag(dp_acc, pa_ds);
@@ -565,13 +630,16 @@ __global__ __launch_bounds__(256) void attn_bwd_dvdk_kernel(
 // dq kernel: wave owns q rows [q0, q0+32); loops kv tiles (forward
 // orientation: acc = [key-pattern][qrow=lane&31]).
 // -------------------------------------------------------------------
+template <bool DROP = false>
 __global__ __launch_bounds__(256, 3) void attn_bwd_dq_kernel(
     const bf16_t* __restrict__ q, const bf16_t* __restrict__ k,
     const bf16_t* __restrict__ v, const bf16_t* __restrict__ dout,
     const float* __restrict__ lse, const float* __restrict__ delta,
     const float* __restrict__ mask, bf16_t* __restrict__ dq, int S,
     int causal, float scale, int has_mask, long H, StridesB sio,
-    StridesB sdo, StridesB sg) {
+    StridesB sdo, StridesB sg, float pdrop = 0.f, float inv_keep = 1.f,
+    unsigned long long dseed = 0,
+    const long long* __restrict__ seed_buf = nullptr) {
   constexpr int D = 64;
   // K tile swizzled (b128 A-frags for S^T, XOR-adjusted u16 B-reads for
   // dQ) + V tile swizzled (b128 A-frags for dP^T); double-buffered
@@ -654,6 +722,12 @@ __global__ __launch_bounds__(256, 3) void attn_bwd_dq_kernel(
       dp_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vfr, dof[s], dp_acc,
                                                        0, 0, 0);
     }
+    unsigned dbits = 0;
+    if constexpr (DROP) {
+      const Philox4 ph(seed_buf ? dseed ^ (unsigned long long)*seed_buf
+                                : dseed);
+      dbits = drop_bits_rowfixed(ph, bh, k0, hi, qrow, pdrop);
+    }
 #pragma unroll
     for (int r = 0; r < 16; ++r) {
       const int kk = k0 + (r & 3) + 8 * (r >> 2) + 4 * hi;
@@ -662,7 +736,10 @@ __global__ __launch_bounds__(256, 3) void attn_bwd_dq_kernel(
       bool dead = (qrow >= S) || (kk >= S) || (causal && kk > qrow) ||
                   !isfinite(l_row);
       const float p = dead ? 0.f : __expf(sv - l_row);
-      dp_acc[r] = dead ? 0.f : p * (dp_acc[r] - dlt_row) * scale;  // dS
+      float dpr = dp_acc[r];
+      if constexpr (DROP)
+        dpr *= ((dbits >> r) & 1) ? inv_keep : 0.f;
+      dp_acc[r] = dead ? 0.f : p * (dpr - dlt_row) * scale;  // dS
     }
     // dQ += dS K : A = transform(dS^T) over k=keys; B[k=key][j=d] = K rows
     unsigned int pa_ds[2][4];
@@ -713,7 +790,9 @@ std::vector<at::Tensor> attn_bwd_impl(
     const bf16_t* dob, at::Tensor lse, const float* mask_ptr, bool has_mask,
     bool causal, double scale, long B, long H, int S, StridesB sio,
     StridesB sdo, StridesB sg, bf16_t* dqb, bf16_t* dkb, bf16_t* dvb,
-    const at::TensorOptions& fopt) {
+    const at::TensorOptions& fopt, float pdrop = 0.f,
+    unsigned long long dseed = 0, const long long* seed_buf = nullptr) {
+  const float inv_keep = 1.f / (1.f - pdrop);
   auto delta = at::empty({B * H * (long)S}, fopt);
   auto stream = c10::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
   const long NR = B * H * (long)S;
@@ -728,12 +807,34 @@ std::vector<at::Tensor> attn_bwd_impl(
     const char* e = getenv("RAVNEST_ATTN_MERGED_DKV");
     return !(e && e[0] == '1');
   }();
+  TORCH_CHECK(pdrop == 0.f || split_dkv,
+              "attn_bwd: fused prob-dropout needs the split dv/dk path");
+  if (pdrop > 0.f) {
+    hipLaunchKernelGGL((attn_bwd_dv_kernel<true>), gridk, block, 0, stream,
+                       qb, kb, dob, lse.data_ptr<float>(), mask_ptr, dvb, S,
+                       causal ? 1 : 0, (float)scale, has_mask ? 1 : 0, H,
+                       sio, sdo, sg, pdrop, inv_keep, dseed, seed_buf);
+    hipLaunchKernelGGL((attn_bwd_dk_kernel<true>), gridk, block, 0, stream,
+                       qb, kb, vb, dob, lse.data_ptr<float>(),
+                       delta.data_ptr<float>(), mask_ptr, dkb, S,
+                       causal ? 1 : 0, (float)scale, has_mask ? 1 : 0, H,
+                       sio, sdo, sg, pdrop, inv_keep, dseed, seed_buf);
+    hipLaunchKernelGGL((attn_bwd_dq_kernel<true>), gridk, block, 0, stream,
+                       qb, kb, vb, dob, lse.data_ptr<float>(),
+                       delta.data_ptr<float>(), mask_ptr, dqb, S,
+                       causal ? 1 : 0, (float)scale, has_mask ? 1 : 0, H,
+                       sio, sdo, sg, pdrop, inv_keep, dseed, seed_buf);
+    HIP_CHECK_LAST();
+    return {};
+  }
   if (split_dkv) {
-    hipLaunchKernelGGL(attn_bwd_dv_kernel, gridk, block, 0, stream, qb, kb,
+    hipLaunchKernelGGL((attn_bwd_dv_kernel<false>), gridk, block, 0, stream,
+                       qb, kb,
                        dob, lse.data_ptr<float>(), mask_ptr, dvb, S,
                        causal ? 1 : 0, (float)scale, has_mask ? 1 : 0, H,
                        sio, sdo, sg);
-    hipLaunchKernelGGL(attn_bwd_dk_kernel, gridk, block, 0, stream, qb, kb,
+    hipLaunchKernelGGL((attn_bwd_dk_kernel<false>), gridk, block, 0, stream,
+                       qb, kb,
                        vb, dob, lse.data_ptr<float>(),
                        delta.data_ptr<float>(), mask_ptr, dkb, S,
                        causal ? 1 : 0, (float)scale, has_mask ? 1 : 0, H,
@@ -745,7 +846,8 @@ std::vector<at::Tensor> attn_bwd_impl(
                        causal ? 1 : 0, (float)scale, has_mask ? 1 : 0, H,
                        sio, sdo, sg);
   }
-  hipLaunchKernelGGL(attn_bwd_dq_kernel, gridk, block, 0, stream, qb, kb,
+  hipLaunchKernelGGL((attn_bwd_dq_kernel<false>), gridk, block, 0, stream,
+                     qb, kb,
                      vb, dob, lse.data_ptr<float>(), delta.data_ptr<float>(),
                      mask_ptr, dqb, S, causal ? 1 : 0, (float)scale,
                      has_mask ? 1 : 0, H, sio, sdo, sg);
@@ -794,7 +896,8 @@ std::vector<at::Tensor> attn_bwd(at::Tensor q, at::Tensor k, at::Tensor v,
 // packed layout: qkv (B,S,3,H,D), o/dout (B,S,H,D) -> dqkv (B,S,3,H,D)
 at::Tensor attn_bwd_qkv(at::Tensor qkv, at::Tensor o, at::Tensor dout,
                         at::Tensor lse, at::Tensor mask, bool causal,
-                        double scale) {
+                        double scale, double pdrop, int64_t dseed,
+                        c10::optional<at::Tensor> seed_buf) {
   TORCH_CHECK(qkv.is_cuda() && qkv.is_contiguous() && o.is_contiguous());
   TORCH_CHECK(qkv.dim() == 5 && qkv.size(2) == 3);
   const long B = qkv.size(0), H = qkv.size(3);
@@ -815,11 +918,14 @@ at::Tensor attn_bwd_qkv(at::Tensor qkv, at::Tensor o, at::Tensor dout,
   const StridesB sod{(long)S * HD, (long)D, HD};          // o, dout
   const bf16_t* base = reinterpret_cast<const bf16_t*>(qkv.data_ptr());
   bf16_t* dbase = reinterpret_cast<bf16_t*>(dqkv.data_ptr());
+  const long long* sb = nullptr;
+  if (seed_buf.has_value()) sb = reinterpret_cast<const long long*>(seed_buf->data_ptr<int64_t>());
   attn_bwd_impl(base, base + HD, base + 2 * HD,
                 reinterpret_cast<const bf16_t*>(o.data_ptr()),
                 reinterpret_cast<const bf16_t*>(dc.data_ptr()), lse,
                 mask_ptr, has_mask, causal, scale, B, H, S, sp, sod, sp,
                 dbase, dbase + HD, dbase + 2 * HD,
-                qkv.options().dtype(at::kFloat));
+                qkv.options().dtype(at::kFloat), (float)pdrop,
+                (unsigned long long)dseed, sb);
   return dqkv;
 }

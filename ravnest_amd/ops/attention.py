@@ -104,17 +104,33 @@ class AttentionCore(nn.Module):
 class _AttnQKVFn(torch.autograd.Function):
     """Packed path: qkv (B,S,3,H,D) -> o (B,S,H,D). No permute/contiguous
     copies: the kernel reads the projection's natural layout and writes O
-    in token-major order (the next Linear's input layout)."""
+    in token-major order (the next Linear's input layout).
+
+    pdrop > 0 fuses attention-PROB dropout (HF BertSelfAttention
+    semantics) into the fwd and bwd kernels: the philox seed is drawn
+    from the torch CPU generator (so the versioned-recompute engine's
+    fork_rng replay redraws the identical mask) and XORed with the
+    device graph counter at run time (fresh masks per hipGraph replay;
+    fwd and bwd of one step see the same counter value)."""
 
     @staticmethod
-    def forward(ctx, qkv, mask, causal, scale):
+    def forward(ctx, qkv, mask, causal, scale, pdrop=0.0):
         ext = get_ext(required=True)
+        seed = 0
+        seed_buf = None
+        if pdrop > 0:
+            from . import rng
+            seed = int(torch.randint(0, 2**62, (1,)).item())
+            seed_buf = rng.device_seed_counter(qkv.device)
         o, lse = ext.attn_fwd_qkv(qkv,
                                   mask if mask is not None else torch.Tensor(),
-                                  causal, scale)
+                                  causal, scale, pdrop, seed, seed_buf)
         ctx.save_for_backward(qkv, mask if mask is not None else None, o, lse)
         ctx.causal = causal
         ctx.scale = scale
+        ctx.pdrop = pdrop
+        ctx.dseed = seed
+        ctx.seed_buf = seed_buf
         return o
 
     @staticmethod
@@ -124,17 +140,20 @@ class _AttnQKVFn(torch.autograd.Function):
         dqkv = ext.attn_bwd_qkv(
             qkv, o, do.contiguous(), lse,
             mask if mask is not None else torch.Tensor(),
-            ctx.causal, ctx.scale)
-        return dqkv, None, None, None
+            ctx.causal, ctx.scale, ctx.pdrop, ctx.dseed, ctx.seed_buf)
+        return dqkv, None, None, None, None
 
 
-def attention_qkv(qkv, mask=None, causal=False, scale=None):
-    """qkv (B,S,3,H,D) -> (B,S,H*D)."""
+def attention_qkv(qkv, mask=None, causal=False, scale=None,
+                  prob_dropout=0.0):
+    """qkv (B,S,3,H,D) -> (B,S,H*D). prob_dropout fuses attention-prob
+    dropout into the kernels (D=64 path)."""
     H, D = qkv.shape[-2], qkv.shape[-1]
     if scale is None:
         scale = 1.0 / math.sqrt(D)
     if qkv.is_cuda and D == 64:
-        o = _AttnQKVFn.apply(qkv.contiguous(), mask, causal, scale)
+        o = _AttnQKVFn.apply(qkv.contiguous(), mask, causal, scale,
+                             prob_dropout)
         return o.flatten(2)
     # fallback: unpack + (custom or math) attention
     q, k, v = (qkv.permute(2, 0, 3, 1, 4)[i] for i in range(3))
@@ -180,13 +199,14 @@ class AttentionCoreQKV(nn.Module):
     """Fused flash-style attention over packed (B,S,3,H,D) qkv.
 
     prob_dropout: dropout rate applied to the ATTENTION PROBABILITIES
-    (HF BertSelfAttention semantics). The fused flash kernel never
-    materializes the probs, so a prob_dropout > 0 routes through the
-    composed P-materializing path above (exact semantics, replayable
-    philox mask, lower throughput). With prob_dropout == 0 (the perf
-    default used by the bench models, which apply dropout AFTER the
-    output projection instead — a DOCUMENTED semantics difference vs
-    HF BERT) the fused kernel runs."""
+    (HF BertSelfAttention semantics). On the D=64 path the philox mask
+    is FUSED into the flash fwd/bwd kernels (no P materialization;
+    masks regenerated in the backward from the same counter scheme);
+    other shapes route through the composed P-materializing path.
+    Measured cost on BERT-base: ~19% step time (philox at 4 kernel
+    sites), so prob_dropout == 0 remains the perf default used by the
+    bench models, which apply dropout AFTER the output projection — a
+    DOCUMENTED semantics difference vs HF BERT."""
     _is_leaf_module = True
 
     def __init__(self, causal: bool = False, prob_dropout: float = 0.0):
@@ -196,6 +216,9 @@ class AttentionCoreQKV(nn.Module):
 
     def forward(self, qkv, mask=None):
         if self.prob_dropout > 0 and self.training and qkv.is_cuda:
+            if qkv.shape[-1] == 64:  # fused philox masks in the kernels
+                return attention_qkv(qkv, mask=mask, causal=self.causal,
+                                     prob_dropout=self.prob_dropout)
             return attention_qkv_prob_dropout(
                 qkv, mask, self.causal, None, self.prob_dropout,
                 self.training)

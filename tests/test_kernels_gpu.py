@@ -804,3 +804,82 @@ def test_attention_prob_dropout(dev):
             qkv, None, False, None, 0.5, True).float()
     mdiff = (acc / 48 - o_p0.float()).abs().mean().item()
     assert mdiff < 0.05, mdiff
+
+
+def test_fused_attention_prob_dropout(dev):
+    """FUSED attention-prob dropout: with V = I the output IS the
+    dropped probability matrix, so the philox mask is recovered exactly
+    and a torch reference built with THAT mask must match outputs AND
+    all gradients; plus p=0 equivalence and same-seed determinism."""
+    from ravnest_amd.ops.attention import attention_qkv
+    torch.manual_seed(7)
+    B, S, H, D = 1, 64, 2, 64
+    p = 0.5
+    qkv = (torch.randn(B, S, 3, H, D, device=dev) / 4).to(torch.bfloat16)
+    # identity V per head
+    qkv_id = qkv.clone()
+    eye = torch.eye(S, D, device=dev).to(torch.bfloat16)
+    for h in range(H):
+        qkv_id[0, :, 2, h] = eye
+    scale = 1.0 / math.sqrt(D)
+
+    # p=0 path identical to the plain kernel
+    o_plain = attention_qkv(qkv)
+    o_p0 = attention_qkv(qkv, prob_dropout=0.0)
+    assert torch.equal(o_plain, o_p0)
+
+    # recover the mask from an identity-V run
+    with torch.random.fork_rng(devices=[dev]):
+        torch.manual_seed(123)
+        o_id = attention_qkv(qkv_id, prob_dropout=p)  # (B,S,H*D)
+    q, k = qkv_id[0, :, 0], qkv_id[0, :, 1]           # (S,H,D)
+    pd = o_id.reshape(S, H, D)
+    masks = []
+    for h in range(H):
+        s = (q[:, h].float() @ k[:, h].float().t()) * scale
+        pref = torch.softmax(s, dim=-1)
+        ratio = pd[:, h].float() / pref.clamp_min(1e-9)
+        # every ratio must be ~0 or ~1/(1-p)
+        near0 = ratio.abs() < 0.25
+        near2 = (ratio - 2.0).abs() < 0.35
+        frac = (near0 | near2).float().mean().item()
+        assert frac > 0.995, f"mask structure violated ({frac})"
+        keep_frac = near2.float().mean().item()
+        assert 0.40 < keep_frac < 0.60, keep_frac
+        masks.append(near2.float())
+
+    # gradients: torch reference with the EXTRACTED mask vs fused bwd
+    with torch.random.fork_rng(devices=[dev]):
+        torch.manual_seed(123)  # same seed draw -> same philox mask
+        qkv_live = qkv_id.detach().clone().requires_grad_(True)
+        o = attention_qkv(qkv_live, prob_dropout=p)
+        dout = (torch.randn_like(o) / 8)
+        o.backward(dout)
+
+    qkv_ref = qkv_id.detach().float().requires_grad_(True)
+    outs = []
+    for h in range(H):
+        qh = qkv_ref[0, :, 0, h]
+        kh = qkv_ref[0, :, 1, h]
+        vh = qkv_ref[0, :, 2, h]
+        pref = torch.softmax((qh @ kh.t()) * scale, dim=-1)
+        outs.append((pref * masks[h] * 2.0) @ vh)
+    # outs[h] is (S, D); o layout is (B, S, H*D) head-major inner
+    oref = torch.stack(outs, dim=1).reshape(1, S, H * D)
+    ((oref * dout.float()).sum()).backward()
+    g = qkv_live.grad.float()
+    gr = qkv_ref.grad
+    err_o = (o.float() - oref.detach()).abs().max().item()
+    assert err_o < 5e-2, f"output vs mask-reference {err_o}"
+    scale_g = gr.abs().max().item() + 1e-6
+    err_g = (g - gr).abs().max().item() / scale_g
+    assert err_g < 6e-2, f"grad rel err {err_g}"
+
+    # determinism across identical seeded runs
+    with torch.random.fork_rng(devices=[dev]):
+        torch.manual_seed(55)
+        a = attention_qkv(qkv, prob_dropout=p)
+    with torch.random.fork_rng(devices=[dev]):
+        torch.manual_seed(55)
+        b = attention_qkv(qkv, prob_dropout=p)
+    assert torch.equal(a, b)

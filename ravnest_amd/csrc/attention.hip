@@ -258,7 +258,8 @@ __global__ __launch_bounds__(256, 4) void attn_fwd_lds_kernel(
     Strides sq, Strides sk, Strides sv, Strides so,
     float pdrop = 0.f, float inv_keep = 1.f,
     unsigned long long dseed = 0,
-    const long long* __restrict__ seed_buf = nullptr) {
+    const long long* __restrict__ seed_buf = nullptr,
+    unsigned* __restrict__ mbits_out = nullptr) {
   constexpr int D = 64;
   // per slot: K swizzled 4KB + V linear 4KB (row stride 128B) or, for
   // the tr16 path, V padded to 144B rows (4608B) so the 4-lanes-per-row
@@ -397,6 +398,7 @@ __global__ __launch_bounds__(256, 4) void attn_fwd_lds_kernel(
         const long bh_ctr = ((unsigned long long)bh << 32) |
                             (unsigned)(qrow >> 1);
         const int qb = (qrow & 1) << 1;
+        unsigned mloc = 0;
 #pragma unroll
         for (int rq = 0; rq < 8; ++rq) {
           const int r0 = 2 * rq;
@@ -404,11 +406,26 @@ __global__ __launch_bounds__(256, 4) void attn_fwd_lds_kernel(
           unsigned int rr[4];
           ph.gen((unsigned long long)bh_ctr,
                  (unsigned long long)(key0 >> 1), rr);
-          const float u0 = (rr[qb] >> 8) * (1.0f / 16777216.0f);
-          const float u1 = (rr[qb | 1] >> 8) * (1.0f / 16777216.0f);
-          s_acc[r0] *= (u0 >= pdrop) ? inv_keep : 0.f;
-          s_acc[r0 + 1] *= (u1 >= pdrop) ? inv_keep : 0.f;
+          if ((rr[qb] >> 8) * (1.0f / 16777216.0f) >= pdrop)
+            mloc |= 1u << r0;
+          if ((rr[qb | 1] >> 8) * (1.0f / 16777216.0f) >= pdrop)
+            mloc |= 1u << (r0 + 1);
         }
+#pragma unroll
+        for (int r = 0; r < 16; ++r)
+          s_acc[r] *= ((mloc >> r) & 1) ? inv_keep : 0.f;
+        // publish the 32-key word for the backward kernels: bit
+        // (key - k0) of this qrow; lane pairs (hi halves) interleave
+        // 4-key nibbles, merged via permlane-free shfl
+        unsigned w_half = 0;
+#pragma unroll
+        for (int r = 0; r < 16; ++r)
+          w_half |= ((mloc >> r) & 1)
+                    << ((r & 3) + 8 * (r >> 2) + 4 * hi);
+        const unsigned w32 = w_half | __shfl_xor(w_half, 32, WAVE);
+        if (hi == 0 && qrow < S && mbits_out)
+          mbits_out[(bh * (long)S + qrow) * ((S + 31) >> 5) + (k0 >> 5)] =
+              w32;
       }
 
       unsigned int pa[2][4];
@@ -629,13 +646,19 @@ std::vector<at::Tensor> attn_fwd_qkv(at::Tensor qkv, at::Tensor mask,
                 "path");
     const long long* sb = nullptr;
     if (seed_buf.has_value()) sb = reinterpret_cast<const long long*>(seed_buf->data_ptr<int64_t>());
+    // per-(bh,qrow) 32-key mask words, read back by the bwd kernels
+    auto mbits = at::empty({B * H * (long)S, (long)((S + 31) / 32)},
+                           qkv.options().dtype(at::kInt));
     hipLaunchKernelGGL((attn_fwd_lds_kernel<true, true>), grid, block, 0,
                        stream, base, base + HD, base + 2 * HD, mask_ptr,
                        reinterpret_cast<bf16_t*>(o.data_ptr()),
                        lse.data_ptr<float>(), S, causal ? 1 : 0,
                        (float)scale, has_mask ? 1 : 0, (long)H, sp, sp, sp,
                        so, (float)pdrop, 1.f / (1.f - (float)pdrop),
-                       (unsigned long long)dseed, sb);
+                       (unsigned long long)dseed, sb,
+                       reinterpret_cast<unsigned*>(mbits.data_ptr<int>()));
+    HIP_CHECK_LAST();
+    return {o, lse, mbits};
   } else if (D == 64) {
     LAUNCH_ATTN_FWD(64, base, base + HD, base + 2 * HD,
                     reinterpret_cast<bf16_t*>(o.data_ptr()), sp, sp, sp, so);

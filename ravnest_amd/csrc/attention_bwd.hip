@@ -33,43 +33,43 @@ typedef float f32x16 __attribute__((ext_vector_type(16)));
 
 namespace {
 
-// fused attention-prob dropout (matches the fwd kernel's scheme): one
-// philox gen per 2x2 (qrow,key) square, ctr = (bh<<32 | qrow>>1,
-// key>>1), output index (qrow&1)*2 + (key&1), keep iff u >= p. The two
-// mask-bit builders cover the two lane layouts: key fixed per lane
-// (dv/dk kernels) and qrow fixed per lane (dq kernel).
-DEVINL unsigned drop_bits_keyfixed(const Philox4& ph, long bh, int q0,
-                                   int hi, int key, float p) {
+// fused attention-prob dropout: the fwd kernel published one 32-key
+// mask word per (bh, qrow) k-block (bit = key - k0, philox drawn once
+// in the fwd); the bwd kernels READ those words instead of re-running
+// philox (regenerating measured ~4x the masking cost). Two readers for
+// the two lane layouts: key fixed per lane (dv/dk) and qrow fixed per
+// lane (dq).
+DEVINL unsigned drop_bits_keyfixed(const unsigned* __restrict__ mb,
+                                   long bh, int S, int q0, int hi,
+                                   int key, int k0) {
+  // lane l loads the word of qrow q0 + (l&31); each lane then gathers
+  // its 16 r-elements' bits via wave shuffles (wave-uniform branch)
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wpr = (S + 31) >> 5;
+  const int myrow = min(q0 + (lane & 31), S - 1);
+  const unsigned w = mb[(bh * (long)S + myrow) * wpr + (k0 >> 5)];
   unsigned bits = 0;
-  const int kb = key & 1;
+  const int kb = key - k0;  // bit position within the word (0..31)
 #pragma unroll
-  for (int rq = 0; rq < 8; ++rq) {
-    const int r0 = 2 * rq;
-    const int qrow0 = q0 + (r0 & 3) + 8 * (r0 >> 2) + 4 * hi;  // even
-    unsigned int rr[4];
-    ph.gen(((unsigned long long)bh << 32) | (unsigned)(qrow0 >> 1),
-           (unsigned long long)(key >> 1), rr);
-    if ((rr[kb] >> 8) * (1.0f / 16777216.0f) >= p) bits |= 1u << r0;
-    if ((rr[2 | kb] >> 8) * (1.0f / 16777216.0f) >= p)
-      bits |= 1u << (r0 + 1);
+  for (int r = 0; r < 16; ++r) {
+    const int qoff = (r & 3) + 8 * (r >> 2) + 4 * hi;  // qrow - q0
+    const unsigned wr = __shfl(w, qoff, WAVE);
+    bits |= ((wr >> kb) & 1) << r;
   }
   return bits;
 }
 
-DEVINL unsigned drop_bits_rowfixed(const Philox4& ph, long bh, int k0,
-                                   int hi, int qrow, float p) {
+DEVINL unsigned drop_bits_rowfixed(const unsigned* __restrict__ mb,
+                                   long bh, int S, int k0, int hi,
+                                   int qrow) {
+  const int wpr = (S + 31) >> 5;
+  const unsigned w =
+      mb[(bh * (long)S + min(qrow, S - 1)) * wpr + (k0 >> 5)];
   unsigned bits = 0;
-  const int qb = (qrow & 1) << 1;
 #pragma unroll
-  for (int rq = 0; rq < 8; ++rq) {
-    const int r0 = 2 * rq;
-    const int key0 = k0 + (r0 & 3) + 8 * (r0 >> 2) + 4 * hi;  // even
-    unsigned int rr[4];
-    ph.gen(((unsigned long long)bh << 32) | (unsigned)(qrow >> 1),
-           (unsigned long long)(key0 >> 1), rr);
-    if ((rr[qb] >> 8) * (1.0f / 16777216.0f) >= p) bits |= 1u << r0;
-    if ((rr[qb | 1] >> 8) * (1.0f / 16777216.0f) >= p)
-      bits |= 1u << (r0 + 1);
+  for (int r = 0; r < 16; ++r) {
+    const int kb = (r & 3) + 8 * (r >> 2) + 4 * hi;  // key - k0
+    bits |= ((w >> kb) & 1) << r;
   }
   return bits;
 }
@@ -158,9 +158,8 @@ __global__ __launch_bounds__(256, 4) void attn_bwd_dv_kernel(
     const bf16_t* __restrict__ dout, const float* __restrict__ lse,
     const float* __restrict__ mask, bf16_t* __restrict__ dv, int S,
     int causal, float scale, int has_mask, long H, StridesB sio,
-    StridesB sdo, StridesB sg, float pdrop = 0.f, float inv_keep = 1.f,
-    unsigned long long dseed = 0,
-    const long long* __restrict__ seed_buf = nullptr) {
+    StridesB sdo, StridesB sg, float inv_keep = 1.f,
+    const unsigned* __restrict__ mbits = nullptr) {
   constexpr int D = 64;
   // block-shared tiles: dO (linear: 2B B-operand reads) and Q
   // (((row&15)<<4)-swizzled: conflict-free b128 A-fragment reads).
@@ -240,11 +239,8 @@ __global__ __launch_bounds__(256, 4) void attn_bwd_dv_kernel(
                                                       0, 0, 0);
     }
     unsigned dbits = 0;
-    if constexpr (DROP) {
-      const Philox4 ph(seed_buf ? dseed ^ (unsigned long long)*seed_buf
-                                : dseed);
-      dbits = drop_bits_keyfixed(ph, bh, q0, hi, key, pdrop);
-    }
+    if constexpr (DROP)
+      dbits = drop_bits_keyfixed(mbits, bh, S, q0, hi, key, k0);
 #pragma unroll
     for (int r = 0; r < 16; ++r) {
       const int qrow = q0 + (r & 3) + 8 * (r >> 2) + 4 * hi;
@@ -305,9 +301,8 @@ __global__ __launch_bounds__(256, 3) void attn_bwd_dk_kernel(
     const float* __restrict__ lse, const float* __restrict__ delta,
     const float* __restrict__ mask, bf16_t* __restrict__ dk, int S,
     int causal, float scale, int has_mask, long H, StridesB sio,
-    StridesB sdo, StridesB sg, float pdrop = 0.f, float inv_keep = 1.f,
-    unsigned long long dseed = 0,
-    const long long* __restrict__ seed_buf = nullptr) {
+    StridesB sdo, StridesB sg, float inv_keep = 1.f,
+    const unsigned* __restrict__ mbits = nullptr) {
   constexpr int D = 64;
   // Q tile swizzled (b128 A-frags for S, XOR-adjusted u16 B-reads for
   // dK) + dO tile swizzled (b128 A-frags for dP); double-buffered.
@@ -400,11 +395,8 @@ __global__ __launch_bounds__(256, 3) void attn_bwd_dk_kernel(
                                                        0, 0, 0);
     }
     unsigned dbits = 0;
-    if constexpr (DROP) {
-      const Philox4 ph(seed_buf ? dseed ^ (unsigned long long)*seed_buf
-                                : dseed);
-      dbits = drop_bits_keyfixed(ph, bh, q0, hi, key, pdrop);
-    }
+    if constexpr (DROP)
+      dbits = drop_bits_keyfixed(mbits, bh, S, q0, hi, key, k0);
 #pragma unroll
     for (int r = 0; r < 16; ++r) {
       const int qrow = q0 + (r & 3) + 8 * (r >> 2) + 4 * hi;
@@ -637,9 +629,8 @@ __global__ __launch_bounds__(256, 3) void attn_bwd_dq_kernel(
     const float* __restrict__ lse, const float* __restrict__ delta,
     const float* __restrict__ mask, bf16_t* __restrict__ dq, int S,
     int causal, float scale, int has_mask, long H, StridesB sio,
-    StridesB sdo, StridesB sg, float pdrop = 0.f, float inv_keep = 1.f,
-    unsigned long long dseed = 0,
-    const long long* __restrict__ seed_buf = nullptr) {
+    StridesB sdo, StridesB sg, float inv_keep = 1.f,
+    const unsigned* __restrict__ mbits = nullptr) {
   constexpr int D = 64;
   // K tile swizzled (b128 A-frags for S^T, XOR-adjusted u16 B-reads for
   // dQ) + V tile swizzled (b128 A-frags for dP^T); double-buffered
@@ -723,11 +714,8 @@ __global__ __launch_bounds__(256, 3) void attn_bwd_dq_kernel(
                                                        0, 0, 0);
     }
     unsigned dbits = 0;
-    if constexpr (DROP) {
-      const Philox4 ph(seed_buf ? dseed ^ (unsigned long long)*seed_buf
-                                : dseed);
-      dbits = drop_bits_rowfixed(ph, bh, k0, hi, qrow, pdrop);
-    }
+    if constexpr (DROP)
+      dbits = drop_bits_rowfixed(mbits, bh, S, k0, hi, qrow);
 #pragma unroll
     for (int r = 0; r < 16; ++r) {
       const int kk = k0 + (r & 3) + 8 * (r >> 2) + 4 * hi;
@@ -791,7 +779,7 @@ std::vector<at::Tensor> attn_bwd_impl(
     bool causal, double scale, long B, long H, int S, StridesB sio,
     StridesB sdo, StridesB sg, bf16_t* dqb, bf16_t* dkb, bf16_t* dvb,
     const at::TensorOptions& fopt, float pdrop = 0.f,
-    unsigned long long dseed = 0, const long long* seed_buf = nullptr) {
+    const unsigned* mbits = nullptr) {
   const float inv_keep = 1.f / (1.f - pdrop);
   auto delta = at::empty({B * H * (long)S}, fopt);
   auto stream = c10::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
@@ -807,23 +795,24 @@ std::vector<at::Tensor> attn_bwd_impl(
     const char* e = getenv("RAVNEST_ATTN_MERGED_DKV");
     return !(e && e[0] == '1');
   }();
-  TORCH_CHECK(pdrop == 0.f || split_dkv,
-              "attn_bwd: fused prob-dropout needs the split dv/dk path");
+  TORCH_CHECK(pdrop == 0.f || (split_dkv && mbits != nullptr),
+              "attn_bwd: fused prob-dropout needs the split dv/dk path "
+              "and the fwd's mask words");
   if (pdrop > 0.f) {
     hipLaunchKernelGGL((attn_bwd_dv_kernel<true>), gridk, block, 0, stream,
                        qb, kb, dob, lse.data_ptr<float>(), mask_ptr, dvb, S,
                        causal ? 1 : 0, (float)scale, has_mask ? 1 : 0, H,
-                       sio, sdo, sg, pdrop, inv_keep, dseed, seed_buf);
+                       sio, sdo, sg, inv_keep, mbits);
     hipLaunchKernelGGL((attn_bwd_dk_kernel<true>), gridk, block, 0, stream,
                        qb, kb, vb, dob, lse.data_ptr<float>(),
                        delta.data_ptr<float>(), mask_ptr, dkb, S,
                        causal ? 1 : 0, (float)scale, has_mask ? 1 : 0, H,
-                       sio, sdo, sg, pdrop, inv_keep, dseed, seed_buf);
+                       sio, sdo, sg, inv_keep, mbits);
     hipLaunchKernelGGL((attn_bwd_dq_kernel<true>), gridk, block, 0, stream,
                        qb, kb, vb, dob, lse.data_ptr<float>(),
                        delta.data_ptr<float>(), mask_ptr, dqb, S,
                        causal ? 1 : 0, (float)scale, has_mask ? 1 : 0, H,
-                       sio, sdo, sg, pdrop, inv_keep, dseed, seed_buf);
+                       sio, sdo, sg, inv_keep, mbits);
     HIP_CHECK_LAST();
     return {};
   }
@@ -896,8 +885,8 @@ std::vector<at::Tensor> attn_bwd(at::Tensor q, at::Tensor k, at::Tensor v,
 // packed layout: qkv (B,S,3,H,D), o/dout (B,S,H,D) -> dqkv (B,S,3,H,D)
 at::Tensor attn_bwd_qkv(at::Tensor qkv, at::Tensor o, at::Tensor dout,
                         at::Tensor lse, at::Tensor mask, bool causal,
-                        double scale, double pdrop, int64_t dseed,
-                        c10::optional<at::Tensor> seed_buf) {
+                        double scale, double pdrop,
+                        c10::optional<at::Tensor> mbits) {
   TORCH_CHECK(qkv.is_cuda() && qkv.is_contiguous() && o.is_contiguous());
   TORCH_CHECK(qkv.dim() == 5 && qkv.size(2) == 3);
   const long B = qkv.size(0), H = qkv.size(3);
@@ -918,14 +907,14 @@ at::Tensor attn_bwd_qkv(at::Tensor qkv, at::Tensor o, at::Tensor dout,
   const StridesB sod{(long)S * HD, (long)D, HD};          // o, dout
   const bf16_t* base = reinterpret_cast<const bf16_t*>(qkv.data_ptr());
   bf16_t* dbase = reinterpret_cast<bf16_t*>(dqkv.data_ptr());
-  const long long* sb = nullptr;
-  if (seed_buf.has_value()) sb = reinterpret_cast<const long long*>(seed_buf->data_ptr<int64_t>());
+  const unsigned* mb = nullptr;
+  if (mbits.has_value())
+    mb = reinterpret_cast<const unsigned*>(mbits->data_ptr<int>());
   attn_bwd_impl(base, base + HD, base + 2 * HD,
                 reinterpret_cast<const bf16_t*>(o.data_ptr()),
                 reinterpret_cast<const bf16_t*>(dc.data_ptr()), lse,
                 mask_ptr, has_mask, causal, scale, B, H, S, sp, sod, sp,
                 dbase, dbase + HD, dbase + 2 * HD,
-                qkv.options().dtype(at::kFloat), (float)pdrop,
-                (unsigned long long)dseed, sb);
+                qkv.options().dtype(at::kFloat), (float)pdrop, mb);
   return dqkv;
 }

@@ -97,8 +97,8 @@ std::vector<at::Tensor> attn_fwd_qkv(at::Tensor qkv, at::Tensor mask,
                                      c10::optional<at::Tensor> seed_buf);
 at::Tensor attn_bwd_qkv(at::Tensor qkv, at::Tensor o, at::Tensor dout,
                         at::Tensor lse, at::Tensor mask, bool causal,
-                        double scale, double pdrop, int64_t dseed,
-                        c10::optional<at::Tensor> seed_buf);
+                        double scale, double pdrop,
+                        c10::optional<at::Tensor> mbits);
 at::Tensor mfma_probe(at::Tensor a, at::Tensor b);
 at::Tensor tr16_probe();
 at::Tensor mfma_mx_probe(at::Tensor a, at::Tensor b, int64_t sa, int64_t sb);
@@ -172,8 +172,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attn_bwd_qkv", &attn_bwd_qkv,
         py::arg("qkv"), py::arg("o"), py::arg("dout"), py::arg("lse"),
         py::arg("mask"), py::arg("causal"), py::arg("scale"),
-        py::arg("pdrop") = 0.0, py::arg("dseed") = 0,
-        py::arg("seed_buf") = py::none(),
+        py::arg("pdrop") = 0.0, py::arg("mbits") = py::none(),
         "flash attention bwd, packed qkv -> dqkv");
   m.def("mfma_probe", &mfma_probe, "32x32x16 bf16 MFMA layout probe");
   m.def("tr16_probe", &tr16_probe, "ds_read_b64_tr_b16 mapping probe");

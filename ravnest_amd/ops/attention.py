@@ -110,8 +110,10 @@ class _AttnQKVFn(torch.autograd.Function):
     semantics) into the fwd and bwd kernels: the philox seed is drawn
     from the torch CPU generator (so the versioned-recompute engine's
     fork_rng replay redraws the identical mask) and XORed with the
-    device graph counter at run time (fresh masks per hipGraph replay;
-    fwd and bwd of one step see the same counter value)."""
+    device graph counter at run time (fresh masks per hipGraph replay).
+    The fwd publishes its keep-mask as one 32-key word per (bh, qrow)
+    k-block; the three bwd kernels READ those words (regenerating via
+    philox measured ~4x the masking cost)."""
 
     @staticmethod
     def forward(ctx, qkv, mask, causal, scale, pdrop=0.0):
@@ -122,25 +124,26 @@ class _AttnQKVFn(torch.autograd.Function):
             from . import rng
             seed = int(torch.randint(0, 2**62, (1,)).item())
             seed_buf = rng.device_seed_counter(qkv.device)
-        o, lse = ext.attn_fwd_qkv(qkv,
-                                  mask if mask is not None else torch.Tensor(),
-                                  causal, scale, pdrop, seed, seed_buf)
-        ctx.save_for_backward(qkv, mask if mask is not None else None, o, lse)
+        res = ext.attn_fwd_qkv(qkv,
+                               mask if mask is not None else torch.Tensor(),
+                               causal, scale, pdrop, seed, seed_buf)
+        o, lse = res[0], res[1]
+        mbits = res[2] if len(res) > 2 else None
+        ctx.save_for_backward(qkv, mask if mask is not None else None, o,
+                              lse, mbits)
         ctx.causal = causal
         ctx.scale = scale
         ctx.pdrop = pdrop
-        ctx.dseed = seed
-        ctx.seed_buf = seed_buf
         return o
 
     @staticmethod
     def backward(ctx, do):
-        qkv, mask, o, lse = ctx.saved_tensors
+        qkv, mask, o, lse, mbits = ctx.saved_tensors
         ext = get_ext(required=True)
         dqkv = ext.attn_bwd_qkv(
             qkv, o, do.contiguous(), lse,
             mask if mask is not None else torch.Tensor(),
-            ctx.causal, ctx.scale, ctx.pdrop, ctx.dseed, ctx.seed_buf)
+            ctx.causal, ctx.scale, ctx.pdrop, mbits)
         return dqkv, None, None, None, None
 
 

@@ -355,6 +355,10 @@ __global__ __launch_bounds__(256, 4) void attn_fwd_lds_kernel(
         s_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qf[s], s_acc,
                                                         0, 0, 0);
       }
+      // additive mask: ONE lane-load for the tile's 32 keys, values
+      // redistributed by wave shuffle (vs 16 scattered loads per lane)
+      float mload = 0.f;
+      if (has_mask) mload = mp[min(k0 + j32, S - 1)];
       float tile_max = -INFINITY;
 #pragma unroll
       for (int r = 0; r < 16; ++r) {
@@ -362,7 +366,7 @@ __global__ __launch_bounds__(256, 4) void attn_fwd_lds_kernel(
         float sv2 = s_acc[r] * scale;
         if (key >= S) sv2 = -INFINITY;
         if (causal && key > qrow) sv2 = -INFINITY;
-        if (has_mask) sv2 += mp[min(key, S - 1)];
+        if (has_mask) sv2 += __shfl(mload, key - k0, WAVE);
         s_acc[r] = sv2;
         tile_max = fmaxf(tile_max, sv2);
       }
@@ -374,12 +378,16 @@ __global__ __launch_bounds__(256, 4) void attn_fwd_lds_kernel(
         m_run = m_new;
       }
       l_run *= alpha;
+      // skip the 32-shfl/32-mul O-rescale whenever NO row's running
+      // max moved this tile (the common case once maxima settle)
+      if (__any(alpha != 1.f)) {
 #pragma unroll
-      for (int r = 0; r < 16; ++r) {
-        const int row_local = (r & 3) + 8 * (r >> 2) + 4 * hi;
-        const float a_r = __shfl(alpha, row_local, WAVE);
+        for (int r = 0; r < 16; ++r) {
+          const int row_local = (r & 3) + 8 * (r >> 2) + 4 * hi;
+          const float a_r = __shfl(alpha, row_local, WAVE);
 #pragma unroll
-        for (int hh = 0; hh < 2; ++hh) oacc[hh][r] *= a_r;
+          for (int hh = 0; hh < 2; ++hh) oacc[hh][r] *= a_r;
+        }
       }
       float psum = 0.f;
 #pragma unroll

@@ -716,11 +716,13 @@ __global__ __launch_bounds__(256, 3) void attn_bwd_dq_kernel(
     unsigned dbits = 0;
     if constexpr (DROP)
       dbits = drop_bits_rowfixed(mbits, bh, S, k0, hi, qrow);
+    float mload = 0.f;
+    if (has_mask) mload = mp[min(k0 + j32, S - 1)];
 #pragma unroll
     for (int r = 0; r < 16; ++r) {
       const int kk = k0 + (r & 3) + 8 * (r >> 2) + 4 * hi;
       float sv = s_acc[r] * scale;
-      if (has_mask && kk < S) sv += mp[kk];
+      if (has_mask && kk < S) sv += __shfl(mload, kk - k0, WAVE);
       bool dead = (qrow >= S) || (kk >= S) || (causal && kk > qrow) ||
                   !isfinite(l_row);
       const float p = dead ? 0.f : __expf(sv - l_row);

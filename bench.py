@@ -268,7 +268,8 @@ def main():
     comm = None
     if world > 1:
         comm = CommBackend(rank=rank, world_size=world, edges=[],
-                           dp_groups=[list(range(world))], device=device)
+                           dp_groups=[list(range(world))], device=device,
+                           backend="gloo" if share_gpu else None)
 
     mi_names = (["input_ids", "attention_mask"] if family == "bert"
                 else ["x"] if family == "resnet" else ["idx"])

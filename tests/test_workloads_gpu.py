@@ -105,3 +105,62 @@ def test_bert_loss_decreases(dev):
     # CPU fp32 reference trajectory reaches ~0.01 by step 80; allow bf16
     # headroom but require real memorization
     assert losses[-1] < 1.0, losses[::10]
+
+
+def _async_cuda_worker(rank, port, out_dir):
+    import os
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    import torch
+    import torch.distributed as dist
+    torch.cuda.set_device(0)  # both ranks share the one GPU
+    dist.init_process_group("gloo", rank=rank, world_size=2)
+    from ravnest_amd.comm.collectives import AsyncReducer
+    dev = torch.device("cuda", 0)
+    g = torch.Generator().manual_seed(11 + rank)
+    ts = [torch.randn(256, 256, generator=g).to(dev),
+          torch.randn(1000, generator=g).to(dev).to(torch.bfloat16)]
+    orig = [t.clone() for t in ts]
+    red = AsyncReducer(dist.group.WORLD, dev)
+    # overlap window: run compute on the default stream while the
+    # collective rides the side stream
+    red.launch(ts)
+    w = torch.randn(512, 512, device=dev)
+    for _ in range(8):
+        w = (w @ w).clamp(-1, 1)
+    assert red.join_into()
+    torch.cuda.synchronize()
+    torch.save({"orig": [t.cpu() for t in orig],
+                "avg": [t.cpu() for t in ts]},
+               f"{out_dir}/gar_{rank}.pt")
+    dist.barrier()
+    dist.destroy_process_group()
+
+
+def test_async_reducer_cuda_streams(tmp_path):
+    """AsyncReducer's GPU path (side stream + event join + record_stream)
+    on real hardware: 2 processes share cuda:0 over gloo; the averaged
+    values must be the exact mean despite concurrent default-stream
+    compute during the overlap window."""
+    if not torch.cuda.is_available():
+        pytest.skip("no GPU")
+    import torch.multiprocessing as mp
+    import os
+    port = 29880 + (os.getpid() % 40)
+    ctx = mp.get_context("spawn")
+    procs = [ctx.Process(target=_async_cuda_worker,
+                         args=(r, port, str(tmp_path)))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(timeout=180)
+    for p in procs:
+        assert p.exitcode == 0, p.exitcode
+    res = {r: torch.load(tmp_path / f"gar_{r}.pt") for r in range(2)}
+    for a0, a1, o0, o1 in zip(res[0]["avg"], res[1]["avg"],
+                              res[0]["orig"], res[1]["orig"]):
+        ref = ((o0.float() + o1.float()) / 2).to(a0.dtype)
+        assert torch.allclose(a0, ref, atol=2e-2), \
+            (a0.float() - ref.float()).abs().max()
+        assert torch.allclose(a1, ref, atol=2e-2)

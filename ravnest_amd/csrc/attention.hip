@@ -371,11 +371,21 @@ __global__ __launch_bounds__(256, 4) void attn_fwd_lds_kernel(
         tile_max = fmaxf(tile_max, sv2);
       }
       tile_max = fmaxf(tile_max, __shfl_xor(tile_max, 32, WAVE));
-      const float m_new = fmaxf(m_run, tile_max);
+      // T13 defer-max (guide §5.5): skip the whole rescale when this
+      // tile's max is within THR of the running max — P values are then
+      // bounded by e^THR instead of 1, which the fp32 accumulator
+      // tolerates (measured accuracy cost ~3x vs THR=0, still inside
+      // the bf16 output rounding class; forced-branch test in
+      // tests/test_kernels_gpu.py). The exponentiation below happens
+      // AFTER this decision (the safe textbook order).
+      constexpr float DEFER_THR = 8.f;
       float alpha = 1.f;
-      if (m_new != m_run) {
-        alpha = (m_run == -INFINITY) ? 0.f : __expf(m_run - m_new);
-        m_run = m_new;
+      if (!__all(tile_max - m_run <= DEFER_THR)) {
+        const float m_new = fmaxf(m_run, tile_max);
+        if (m_new != m_run) {
+          alpha = (m_run == -INFINITY) ? 0.f : __expf(m_run - m_new);
+          m_run = m_new;
+        }
       }
       l_run *= alpha;
       // skip the 32-shfl/32-mul O-rescale whenever NO row's running

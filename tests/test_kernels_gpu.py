@@ -218,6 +218,33 @@ def test_attention_fwd(dev, causal, masked, S, D):
     assert lerr.max().item() < 2e-2, f"lse err {lerr.max()}"
 
 
+def test_attention_fwd_defer_max_spike(dev):
+    """Force the T13 defer-max rescale branch: a late K-tile whose scores
+    dwarf every earlier tile's (running max jumps by >> DEFER_THR), plus
+    a mild case where maxima stay within the threshold. Checks both the
+    O output and the lse (which must stay exact under the deferred
+    max/sum split)."""
+    from ravnest_amd.ops import get_ext
+    ext = get_ext(True)
+    torch.manual_seed(3)
+    B, H, S, D = 2, 2, 512, 64
+    q = torch.randn(B, H, S, D, device=dev, dtype=torch.bfloat16)
+    k = torch.randn(B, H, S, D, device=dev, dtype=torch.bfloat16)
+    # keys 384..415 produce scores ~ +40 above the rest: every Q row's
+    # running max jumps past DEFER_THR at that tile -> rescale path
+    k[:, :, 384:416] *= 8.0
+    v = torch.randn(B, H, S, D, device=dev, dtype=torch.bfloat16)
+    scale = 1.0 / math.sqrt(D)
+    o, lse = ext.attn_fwd(q, k, v, torch.Tensor(), False, scale)
+    s = (q.float() @ k.float().transpose(-2, -1)) * scale
+    p = torch.softmax(s, dim=-1)
+    ref = p @ v.float()
+    err = (o.float() - ref).abs().max().item()
+    assert err < 3e-2, f"spike attention fwd max err {err}"
+    lerr = (lse - torch.logsumexp(s, dim=-1)).abs().max().item()
+    assert lerr < 2e-2, f"spike lse err {lerr}"
+
+
 def test_attention_autograd(dev):
     """Full custom-fwd + GEMM-recompute-bwd path vs fp32 autograd."""
     from ravnest_amd.ops import attention

@@ -11,9 +11,11 @@ numerics-tested, and the flag flips every projection in a model built
 through make_linear()/LinearGelu.
 
 NOTE: the transformer models route plain projections through
-make_linear() — nn.Linear by default (autograd's addmm backward hits
-better hipBLASLt selections than an explicit dy.t()@x wgrad: BERT -2.4%,
-GPT-2 -13%), ops.Linear under the hand flag.
+make_linear() -> ops.Linear: the explicit backward (shape-picked wgrad
+GEMM form + hand colsum bias-grad) measured FASTER than nn.Linear
+autograd on every BERT-base projection shape (tools/bench_linear_bwd.py,
+profiles/r02_linear_bwd.txt: qkv 0.81 vs 0.84 ms, mlp_out 0.81 vs 0.93),
+reversing the round-1 finding. RAVNEST_LIB_LINEAR=1 restores nn.Linear.
 """
 from __future__ import annotations
 
@@ -132,7 +134,14 @@ class _LinearFn(torch.autograd.Function):
         dy2 = dy.reshape(-1, dy.shape[-1])
         x2 = x.reshape(-1, x.shape[-1])
         dx = (dy2 @ w).reshape(x.shape)
-        dw = dy2.t() @ x2
+        # wgrad GEMM form picked by shape (tools/bench_linear_bwd.py):
+        # wide-K projections (mlp_out 3072->768) run 11% faster as
+        # mm(x^T, dy)^T (hipBLASLt sees the other TN problem); the rest
+        # as the direct dy^T @ x.
+        if x2.shape[1] > dy2.shape[1]:
+            dw = torch.mm(x2.t(), dy2.contiguous()).t().contiguous()
+        else:
+            dw = dy2.t() @ x2
         db = None
         if ctx.has_bias:
             ext = get_ext(required=False)
@@ -164,9 +173,10 @@ class Linear(nn.Module):
 
 
 def make_linear(in_features: int, out_features: int, bias: bool = True):
-    """Projection factory for the model zoo: nn.Linear by default (best
-    library backward), ops.Linear when the hand-GEMM flag is set (same
-    state-dict keys either way)."""
-    if hand_gemm_enabled():
-        return Linear(in_features, out_features, bias=bias)
-    return nn.Linear(in_features, out_features, bias=bias)
+    """Projection factory for the model zoo: ops.Linear (explicit
+    backward — measured faster than nn.Linear autograd on the BERT-base
+    shapes, profiles/r02_linear_bwd.txt), same state-dict keys as
+    nn.Linear either way. RAVNEST_LIB_LINEAR=1 restores nn.Linear."""
+    if os.environ.get("RAVNEST_LIB_LINEAR", "0") == "1":
+        return nn.Linear(in_features, out_features, bias=bias)
+    return Linear(in_features, out_features, bias=bias)

@@ -11,11 +11,10 @@ numerics-tested, and the flag flips every projection in a model built
 through make_linear()/LinearGelu.
 
 NOTE: the transformer models route plain projections through
-make_linear() -> ops.Linear: the explicit backward (shape-picked wgrad
-GEMM form + hand colsum bias-grad) measured FASTER than nn.Linear
-autograd on every BERT-base projection shape (tools/bench_linear_bwd.py,
-profiles/r02_linear_bwd.txt: qkv 0.81 vs 0.84 ms, mlp_out 0.81 vs 0.93),
-reversing the round-1 finding. RAVNEST_LIB_LINEAR=1 restores nn.Linear.
+make_linear() — nn.Linear by default: the explicit backward wins
+isolated per-op timings (tools/bench_linear_bwd.py) but loses end-to-end
+inside the captured step (see make_linear docstring), so autograd's
+addmm backward stays the default. RAVNEST_EXPLICIT_LINEAR=1 flips it.
 """
 from __future__ import annotations
 
@@ -173,10 +172,17 @@ class Linear(nn.Module):
 
 
 def make_linear(in_features: int, out_features: int, bias: bool = True):
-    """Projection factory for the model zoo: ops.Linear (explicit
-    backward — measured faster than nn.Linear autograd on the BERT-base
-    shapes, profiles/r02_linear_bwd.txt), same state-dict keys as
-    nn.Linear either way. RAVNEST_LIB_LINEAR=1 restores nn.Linear."""
-    if os.environ.get("RAVNEST_LIB_LINEAR", "0") == "1":
-        return nn.Linear(in_features, out_features, bias=bias)
-    return Linear(in_features, out_features, bias=bias)
+    """Projection factory for the model zoo: nn.Linear by default,
+    ops.Linear under RAVNEST_EXPLICIT_LINEAR=1 or the hand-GEMM flag
+    (same state-dict keys either way).
+
+    Measured both ways twice: the explicit backward wins ISOLATED per-op
+    timings on every BERT-base shape (profiles/r02_linear_bwd.txt — qkv
+    0.81 vs 0.84 ms, mlp_out 0.81 vs 0.93) but LOSES end-to-end inside
+    the captured training step (BERT 1552 vs 1637 samples/s, GPT-2 560
+    vs 610 — hipBLASLt/TunableOp pick different algorithms in the graph
+    context), so autograd's addmm backward stays the default."""
+    if hand_gemm_enabled() or \
+            os.environ.get("RAVNEST_EXPLICIT_LINEAR", "0") == "1":
+        return Linear(in_features, out_features, bias=bias)
+    return nn.Linear(in_features, out_features, bias=bias)

@@ -9,7 +9,8 @@ for the hot ops (ravnest_amd/ops, ravnest_amd/csrc).
 """
 
 from .strings import NodeTypes, ActionTypes, NodeStatus
-from .utils import set_seed, model_fusion, load_node_json_configs
+from .utils import (set_seed, model_fusion, load_node_json_configs,
+                    gpu_usage)
 from .planner import clusterize
 from .engine import Node, Trainer, ComputeEngine
 
@@ -17,6 +18,6 @@ __version__ = "0.1.0"
 
 __all__ = [
     "Node", "Trainer", "ComputeEngine", "clusterize",
-    "set_seed", "model_fusion", "load_node_json_configs",
+    "set_seed", "model_fusion", "load_node_json_configs", "gpu_usage",
     "NodeTypes", "ActionTypes", "NodeStatus",
 ]

@@ -250,6 +250,8 @@ class Node:
             self._reducer = AsyncReducer(self.comm.my_dp_group, self.device)
             self.engine.pre_step_hook = self._join_reduce
 
+        self._mem_log = __import__("os").environ.get(
+            "RAVNEST_MEM_LOG", "0") == "1"
         self._dispatch_thread: threading.Thread | None = None
         self._started = False
         self.health_monitor = None
@@ -758,6 +760,14 @@ class Node:
     def _maybe_reduce(self):
         if not self.reduce_threshold or self.comm is None:
             return
+        if self._mem_log and \
+                self.engine.n_backwards % self.reduce_threshold == 0:
+            # memory telemetry at averaging boundaries (parity: the
+            # reference polls NVML around every fwd/bwd,
+            # utils.py:211-221 / node.py:447-450; we sample at the much
+            # rarer reduce cadence to stay off the hot path)
+            from ..utils import gpu_usage
+            print(f"[{self.name}] mem {gpu_usage()}", flush=True)
         if getattr(self, "_dp_suspended", False):
             return  # a DP replica is down: skip averaging, keep training
         if self.engine.n_backwards % self.reduce_threshold != 0:

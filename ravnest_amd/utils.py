@@ -74,6 +74,36 @@ def set_seed(seed: int) -> None:
     torch.backends.cudnn.benchmark = False
 
 
+def gpu_usage(device=None) -> dict:
+    """GPU + host memory telemetry (parity: reference utils.py:211-221
+    check_gpu_usage via NVML, called around every fwd/bwd at
+    node.py:447-450 / compute.py:148-304; host RAM at node.py:490,554).
+
+    MI355X-native: torch.cuda.mem_get_info goes through the ROCm runtime
+    (no NVML on AMD), plus the allocator's own view of reserved/allocated
+    bytes. Returns {} fields gracefully on CPU-only hosts.
+    """
+    out: dict = {}
+    if torch.cuda.is_available():
+        dev = device if device is not None else torch.cuda.current_device()
+        free_b, total_b = torch.cuda.mem_get_info(dev)
+        out.update(
+            gpu_free_mb=free_b // 2**20,
+            gpu_total_mb=total_b // 2**20,
+            gpu_used_mb=(total_b - free_b) // 2**20,
+            torch_allocated_mb=torch.cuda.memory_allocated(dev) // 2**20,
+            torch_reserved_mb=torch.cuda.memory_reserved(dev) // 2**20,
+        )
+    try:
+        import psutil
+        vm = psutil.virtual_memory()
+        out.update(host_ram_percent=vm.percent,
+                   host_ram_used_mb=vm.used // 2**20)
+    except Exception:
+        pass
+    return out
+
+
 def wire_cast(t: torch.Tensor, wire_dtype: torch.dtype | None) -> torch.Tensor:
     """Optional lossy on-the-wire cast (parity: fp16 compression,
     reference ravnest/utils.py:184-194). On MI355X the natural wire dtype

@@ -169,3 +169,19 @@ def test_torchscript_submodel_export(tmp_path):
     assert torch.allclose(out, ref, atol=1e-5), \
         (out - ref).abs().max().item()
     node.stop()
+
+
+def test_gpu_usage_telemetry():
+    """Memory telemetry parity (reference utils.py:211-221): on a
+    CPU-only host the GPU fields are absent and host RAM is reported;
+    on a GPU box the free/total/allocated fields must be present."""
+    from ravnest_amd import gpu_usage
+    u = gpu_usage()
+    assert isinstance(u, dict)
+    assert "host_ram_percent" in u
+    import torch
+    if torch.cuda.is_available():
+        assert u["gpu_total_mb"] > 0
+        assert 0 <= u["gpu_used_mb"] <= u["gpu_total_mb"]
+    else:
+        assert "gpu_total_mb" not in u

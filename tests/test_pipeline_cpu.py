@@ -212,3 +212,37 @@ def test_shared_loader_label_alignment(tmp_path):
     last_epoch = sum(losses[-n:]) / n
     assert last_epoch < 0.03, \
         f"data/label misalignment suspected: last-epoch loss {last_epoch}"
+
+
+def _f8_worker(rank, base_dir, port, out_dir):
+    os.environ["RAVNEST_WIRE_FP8"] = "1"
+    _comp_worker(rank, base_dir, port, out_dir)
+
+
+def test_pipeline_fp8_wire(tmp_path):
+    """Full 2-stage pipeline with the fp8 wire (RAVNEST_WIRE_FP8=1) on
+    top of bf16 compression: activations AND grads ship as scaled-e4m3
+    bytes (comm/p2p.py). Training must still converge on the CNN
+    workload — exercises the engine-level decompression path against
+    channel-level fp8 reconstruction."""
+    set_seed(42)
+    model = CNN()
+    x = torch.randn(2, 1, 8, 8)
+    base = str(tmp_path / "node_data")
+    pool = [NodeSpec(name=f"n{i}", ram=10 * 2**20) for i in range(2)]
+    clusterize(model, (x,), node_pool=pool, max_clusters=1, base_dir=base)
+    port = 29700 + (os.getpid() % 90)
+    ctx = mp.get_context("spawn")
+    procs = [ctx.Process(target=_f8_worker,
+                         args=(r, base, port, str(tmp_path)))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(timeout=180)
+    for p in procs:
+        assert p.exitcode == 0, f"worker exited {p.exitcode}"
+    losses = [float(l) for l in (tmp_path / "losses.txt").read_text().split()]
+    n_batches = 256 // 32
+    assert len(losses) == 3 * n_batches
+    assert sum(losses[-n_batches:]) < sum(losses[:n_batches])

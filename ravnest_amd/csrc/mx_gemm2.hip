@@ -44,6 +44,16 @@ constexpr int BM = 256, BN = 256, BKE = 128;  // K-tile in fp8 elements
 constexpr int HALF_BYTES = 128 * BKE;         // 16 KiB per half-tile
 constexpr int TILE_BYTES = 2 * HALF_BYTES;
 constexpr int BUF_BYTES = 2 * TILE_BYTES;
+// e8m0 scales ride the SAME glds pipeline as the data (guide trap b:
+// a plain global scale load issued between glds batches forces hipcc
+// to a vmcnt(0) at the scale's use, draining the whole staging
+// pipeline). The host repacks scales tile-major — (nKT, Mp)/(nKT, Np)
+// u32, one u32 = 4 MX-block exponents per row per 128-element K-tile,
+// rows padded to the 256 grid — so one wave can glds each side's 1 KiB
+// per tile; consumers then read them from LDS (lgkm-counted, no vm
+// interaction). 2 KiB per buffer: 2 x 66 KiB still one block per CU.
+constexpr int SC_BYTES = 2048;
+constexpr int BUF_FULL = BUF_BYTES + SC_BYTES;
 
 DEVINL unsigned stage_off8(long ld, long gR0, long gRmax, int i) {
   const int tid = threadIdx.x;
@@ -64,11 +74,13 @@ DEVINL int frag_off8(int rr, int ks, int hi, int pc) {
 }
 
 __global__ __launch_bounds__(512, 2) void mx2_kernel(
-    const unsigned char* __restrict__ XQ, const unsigned char* __restrict__ XS,
-    const unsigned char* __restrict__ WQ, const unsigned char* __restrict__ WS,
-    bf16_t* __restrict__ C, long M, long N, long K, int MT, int NTb,
-    int tpb) {
-  __shared__ char smem[2 * BUF_BYTES];
+    const unsigned char* __restrict__ XQ,
+    const unsigned char* __restrict__ XS_T,
+    const unsigned char* __restrict__ WQ,
+    const unsigned char* __restrict__ WS_T,
+    bf16_t* __restrict__ C, long M, long N, long K, long Mp, long Np,
+    int MT, int NTb, int tpb) {
+  __shared__ char smem[2 * BUF_FULL];
 
   const int nwg = MT * NTb;
   int bid = blockIdx.x, wg;
@@ -95,7 +107,6 @@ __global__ __launch_bounds__(512, 2) void mx2_kernel(
   const long nBase = (long)bn0 * BN;
   const int nKT = (int)(K / BKE);
   const unsigned SJ = (unsigned)(BN * K);  // bytes per column tile of W
-  const long SBLK = K / 32;                // scale bytes per row
 
   unsigned sp[4][2];
   unsigned capB[2];
@@ -116,7 +127,7 @@ __global__ __launch_bounds__(512, 2) void mx2_kernel(
     }
     const unsigned off = (unsigned)tt * BKE;
     const int b = (jj * nKT + tt) & 1;
-    char* const lb = smem + b * BUF_BYTES + slo[kind] + wbase;
+    char* const lb = smem + b * BUF_FULL + slo[kind] + wbase;
     const char* const base =
         reinterpret_cast<const char*>(kind >= 2 ? WQ : XQ);
 #pragma unroll
@@ -140,18 +151,38 @@ __global__ __launch_bounds__(512, 2) void mx2_kernel(
       ++jj;
     }
   };
+  // one wave gldses each side's 1 KiB of tile-major scales (dst is the
+  // area base: glds writes lane L's 16 B at dst + 16L; rows padded on
+  // the host so no clamp is needed)
+  auto stage_sc = [&](int jj, int tt) {
+    if (jj >= tpb) {
+      jj = tpb - 1;
+      tt = nKT - 1;
+    }
+    const int b = (jj * nKT + tt) & 1;
+    if (wid == 0) {
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void*)(
+              XS_T + (long)tt * Mp * 4 + mBase * 4 + lane * 16),
+          (__attribute__((address_space(3))) void*)(
+              smem + b * BUF_FULL + BUF_BYTES),
+          16, 0, 0);
+    } else if (wid == 1) {
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void*)(
+              WS_T + (long)tt * Np * 4 + (nBase + (long)jj * BN) * 4 +
+              lane * 16),
+          (__attribute__((address_space(3))) void*)(
+              smem + b * BUF_FULL + BUF_BYTES + 1024),
+          16, 0, 0);
+    }
+  };
 
+  stage_sc(0, 0);
   stage(0, 0, 0); stage(0, 0, 2); stage(0, 0, 3); stage(0, 0, 1);
   stage(0, 1, 2); stage(0, 1, 3); stage(0, 1, 0);
   asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
   __builtin_amdgcn_s_barrier();
-
-  // per-fragment scale rows (clamped like the data)
-  const long aSR[4] = {min(mBase + wm * 128 + 0 + l31, M - 1),
-                       min(mBase + wm * 128 + 32 + l31, M - 1),
-                       min(mBase + wm * 128 + 64 + l31, M - 1),
-                       min(mBase + wm * 128 + 96 + l31, M - 1)};
-  const long bSR0 = wn * 64 + l31;  // + column-tile offset at use
 
   i32x8v aF[2][2], bF[2][2];  // [frag][ks] current strip
   unsigned aSc[4], bSc[2];    // u32 scale words (4 blocks each)
@@ -168,23 +199,30 @@ __global__ __launch_bounds__(512, 2) void mx2_kernel(
   for (int t = 0; t < nKT; ++t) {
     const int b = (j * nKT + t) & 1;
     int jj, tt;
-    char* const aH0 = smem + b * BUF_BYTES + wm * HALF_BYTES;
+    char* const aH0 = smem + b * BUF_FULL + wm * HALF_BYTES;
     char* const bH =
-        smem + b * BUF_BYTES + TILE_BYTES + (wn >> 1) * HALF_BYTES;
+        smem + b * BUF_FULL + TILE_BYTES + (wn >> 1) * HALF_BYTES;
     const int bRow0 = (wn & 1) * 64;
 
-    // scale words for this K-tile (4 MX blocks per row per tile)
-#pragma unroll
-    for (int mf = 0; mf < 4; ++mf)
-      aSc[mf] = *reinterpret_cast<const unsigned*>(
-          XS + aSR[mf] * SBLK + (long)t * 4);
+    // scale words for this K-tile: LDS reads of the glds-staged area
+    // (lgkm-counted — no vmcnt interaction with the data pipeline)
     {
-      const long wrow = min(nBase + (long)j * BN + bSR0, N - 1);
-      bSc[0] = *reinterpret_cast<const unsigned*>(
-          WS + wrow * SBLK + (long)t * 4);
-      const long wrow2 = min(nBase + (long)j * BN + bSR0 + 32, N - 1);
-      bSc[1] = *reinterpret_cast<const unsigned*>(
-          WS + wrow2 * SBLK + (long)t * 4);
+      const char* const scA = smem + b * BUF_FULL + BUF_BYTES;
+      const char* const scB = scA + 1024;
+#pragma unroll
+      for (int mf = 0; mf < 4; ++mf)
+        aSc[mf] = *reinterpret_cast<
+            const __attribute__((address_space(3))) unsigned*>(
+            (const __attribute__((address_space(3))) void*)(
+                scA + (wm * 128 + mf * 32 + l31) * 4));
+      bSc[0] = *reinterpret_cast<
+          const __attribute__((address_space(3))) unsigned*>(
+          (const __attribute__((address_space(3))) void*)(
+              scB + (wn * 64 + l31) * 4));
+      bSc[1] = *reinterpret_cast<
+          const __attribute__((address_space(3))) unsigned*>(
+          (const __attribute__((address_space(3))) void*)(
+              scB + (wn * 64 + 32 + l31) * 4));
     }
 
     // ph0: read B strip 0 + A strips of M-half 0 (12 b128 reads)
@@ -210,6 +248,7 @@ __global__ __launch_bounds__(512, 2) void mx2_kernel(
       }
     norm(j, t, 1, jj, tt);
     stage(jj, tt, 1);
+    stage_sc(jj, tt);  // next tile's scales ride the same vmcnt window
     __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int ks = 0; ks < 2; ++ks)
@@ -343,15 +382,31 @@ at::Tensor mx_gemm2(at::Tensor xq, at::Tensor xs, at::Tensor wq,
     }
   }
   const int NTb = NT / tpb;
+  // tile-major scale repack for the kernel's glds staging: (rows, K/32)
+  // u8 -> (nKT, rowsPadded) u32 (one u32 = the 4 MX-block exponents of
+  // one row's 128-element K-tile), rows zero-padded to the 256 grid
+  // (e8m0 0x00 = 2^-127: pad rows contribute ~0 and their outputs are
+  // dropped by the epilogue bounds checks anyway).
+  const long nKT = K / BKE, Mp = (long)MT * BM, Np = (long)NT * BN;
+  auto iopt = xq.options().dtype(at::kInt);
+  auto pack = [&](const at::Tensor& sc, long rows, long rowsP) {
+    auto t = at::zeros({nKT, rowsP}, iopt);
+    t.narrow(1, 0, rows).copy_(
+        sc.reshape({rows, nKT, 4}).view(at::kInt).reshape({rows, nKT})
+            .transpose(0, 1));
+    return t;
+  };
+  auto xs_t = pack(xs, M, Mp);
+  auto ws_t = pack(ws, N, Np);
   auto stream = c10::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
   dim3 grid((unsigned)(MT * NTb)), block(512);
   hipLaunchKernelGGL(mx2_kernel, grid, block, 0, stream,
                      xq.data_ptr<unsigned char>(),
-                     xs.data_ptr<unsigned char>(),
+                     reinterpret_cast<unsigned char*>(xs_t.data_ptr<int>()),
                      wq.data_ptr<unsigned char>(),
-                     ws.data_ptr<unsigned char>(),
-                     reinterpret_cast<bf16_t*>(C.data_ptr()), M, N, K, MT,
-                     NTb, tpb);
+                     reinterpret_cast<unsigned char*>(ws_t.data_ptr<int>()),
+                     reinterpret_cast<bf16_t*>(C.data_ptr()), M, N, K, Mp,
+                     Np, MT, NTb, tpb);
   HIP_CHECK_LAST();
   return C;
 }

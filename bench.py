@@ -49,6 +49,12 @@ def build_model(name: str, seq: int):
         return BertForMLM(BertConfig.tiny(max_seq=seq)), "bert"
     if name == "gpt2-small":
         return GPT(GPTConfig.gpt2_small(block_size=seq)), "gpt"
+    if name == "gpt2-small-fp8":
+        # BASELINE.json config 5: block projections on the MX fp8 MFMA
+        # path (ops.MXLinear -> csrc/mx_gemm2.hip)
+        cfg = GPTConfig.gpt2_small(block_size=seq)
+        cfg.fp8 = True
+        return GPT(cfg), "gpt"
     if name == "resnet50":
         from ravnest_amd.models import resnet50
         m = resnet50(num_classes=200)

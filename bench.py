@@ -197,6 +197,10 @@ def main():
     if world > 1:
         os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
         os.environ.setdefault("MASTER_PORT", "29531")
+        # this host driver only supports dmabuf IPC: without it RCCL
+        # cross-process CUDA-tensor sharing fails with
+        # "hipIpcGetMemHandle: invalid argument"
+        os.environ.setdefault("HSA_ENABLE_IPC_MODE_LEGACY", "0")
 
     if args.dtype is None:
         # transformers run bf16-native; ResNet stays fp32 NCHW (measured:

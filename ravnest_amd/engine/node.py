@@ -418,6 +418,12 @@ class Node:
         self._dispatch_thread.start()
 
     def stop(self):
+        # drain any in-flight overlapped DP reduce first: tearing down
+        # process groups under a live side-stream collective aborts
+        try:
+            self.join_pending_reduce()
+        except Exception:
+            pass
         self._stop.set()
         self._enqueue(2, ("_stop", None, None))
         if self.comm is not None:

@@ -295,7 +295,7 @@ __global__ __launch_bounds__(256, 4) void attn_bwd_dv_kernel(
 // dk kernel: wave owns keys; S, dP -> dS -> dK (~150 VGPR, 3 waves/SIMD)
 // -------------------------------------------------------------------
 template <bool DROP = false>
-__global__ __launch_bounds__(256, 3) void attn_bwd_dk_kernel(
+__global__ __launch_bounds__(256, 4) void attn_bwd_dk_kernel(
     const bf16_t* __restrict__ q, const bf16_t* __restrict__ k,
     const bf16_t* __restrict__ v, const bf16_t* __restrict__ dout,
     const float* __restrict__ lse, const float* __restrict__ delta,

@@ -205,26 +205,43 @@ __global__ __launch_bounds__(256, 4) void attn_bwd_dv_kernel(
     for (int r = 0; r < 16; ++r) dv_acc[hh][r] = 0.f;
 
   const int q_start = causal ? blockIdx.x * 128 : 0;
-  const int lin_dst = st_row * 128 + st_c16;
-  const int q_dst = lin_dst ^ ((st_row & 15) << 4);
-  bf16x8v dstage = *reinterpret_cast<const bf16x8v*>(
-      dop + (long)min(q_start + st_row, S - 1) * sdo.rs + st_c16 / 2);
-  bf16x8v qstage = *reinterpret_cast<const bf16x8v*>(
-      qp + (long)min(q_start + st_row, S - 1) * sio.rs + st_c16 / 2);
-  *reinterpret_cast<bf16x8v*>(smem + lin_dst) = dstage;
-  *reinterpret_cast<bf16x8v*>(smem + 4096 + q_dst) = qstage;
+  // glds staging (LDS-DMA, rule 21): dest is lane-linear, so dO (linear
+  // image) gldses with the plain addressing and Q's XOR swizzle moves
+  // to the SOURCE side via the closed-form inverse of
+  // p ^ (((p>>7)&15)<<4) (verified bijective + 16B-aligned). Replaces
+  // the register round-trip (2 global loads + 2 ds_writes per thread
+  // per tile) and frees the 16 staging VGPRs; the loop-top
+  // __syncthreads drains the in-flight DMA (vmcnt) before first read.
+  const int wdst = ((int)threadIdx.x >> 6) * 1024;
+  const int p16 = (int)threadIdx.x * 16;
+  const int inv_r7 = ((p16 >> 7) & 1) ^ ((p16 >> 10) & 1);
+  const int p_src = (p16 & ~0xF0) | ((((p16 >> 4) & 1) ^ inv_r7) << 4) |
+                    ((((p16 >> 5) & 1) ^ ((p16 >> 8) & 1)) << 5) |
+                    ((((p16 >> 6) & 1) ^ ((p16 >> 9) & 1)) << 6) |
+                    (inv_r7 << 7);
+  const int s_row = p_src >> 7, s_col = p_src & 127;
+  auto stage2 = [&](int qt, char* sl) {
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) void*)(
+            reinterpret_cast<const char*>(dop) +
+            (long)min(qt + st_row, S - 1) * sdo.rs * 2 + st_c16),
+        (__attribute__((address_space(3))) void*)(sl + wdst), 16, 0, 0);
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) void*)(
+            reinterpret_cast<const char*>(qp) +
+            (long)min(qt + s_row, S - 1) * sio.rs * 2 + s_col),
+        (__attribute__((address_space(3))) void*)(sl + 4096 + wdst),
+        16, 0, 0);
+  };
+  stage2(q_start, smem);
   for (int q0 = q_start; q0 < S; q0 += 32) {
     const int slot = ((q0 - q_start) >> 5) & 1;
     const char* do_lds = smem + slot * 8192;
     const char* q_lds = do_lds + 4096;
     __syncthreads();
     const bool have_next = q0 + 32 < S;
-    if (have_next) {  // T14: next tile's loads under this compute
-      dstage = *reinterpret_cast<const bf16x8v*>(
-          dop + (long)min(q0 + 32 + st_row, S - 1) * sdo.rs + st_c16 / 2);
-      qstage = *reinterpret_cast<const bf16x8v*>(
-          qp + (long)min(q0 + 32 + st_row, S - 1) * sio.rs + st_c16 / 2);
-    }
+    if (have_next)  // T14: next tile's LDS-DMA under this compute
+      stage2(q0 + 32, smem + (slot ^ 1) * 8192);
     const bool compute = live_wave && (!causal || q0 + 31 >= k0);
     if (compute) {
     f32x16 s_acc;
@@ -273,11 +290,6 @@ __global__ __launch_bounds__(256, 4) void attn_bwd_dv_kernel(
       }
     }
     }  // compute
-    if (have_next) {
-      char* nb = smem + (slot ^ 1) * 8192;
-      *reinterpret_cast<bf16x8v*>(nb + lin_dst) = dstage;
-      *reinterpret_cast<bf16x8v*>(nb + 4096 + q_dst) = qstage;
-    }
   }
   if (!live_wave) return;
 #pragma unroll
@@ -356,25 +368,38 @@ __global__ __launch_bounds__(256, 4) void attn_bwd_dk_kernel(
     for (int r = 0; r < 16; ++r) dk_acc[hh][r] = 0.f;
 
   const int q_start = causal ? blockIdx.x * 128 : 0;
-  const int swz_dst = (st_row * 128 + st_c16) ^ ((st_row & 15) << 4);
-  bf16x8v qstage = *reinterpret_cast<const bf16x8v*>(
-      qp + (long)min(q_start + st_row, S - 1) * sio.rs + st_c16 / 2);
-  bf16x8v dstage = *reinterpret_cast<const bf16x8v*>(
-      dop + (long)min(q_start + st_row, S - 1) * sdo.rs + st_c16 / 2);
-  *reinterpret_cast<bf16x8v*>(smem + swz_dst) = qstage;
-  *reinterpret_cast<bf16x8v*>(smem + 4096 + swz_dst) = dstage;
+  // glds staging: both tiles are XOR-swizzled images, so both use the
+  // source-side inverse permutation (see the dv kernel's comment)
+  const int wdst = ((int)threadIdx.x >> 6) * 1024;
+  const int p16 = (int)threadIdx.x * 16;
+  const int inv_r7 = ((p16 >> 7) & 1) ^ ((p16 >> 10) & 1);
+  const int p_src = (p16 & ~0xF0) | ((((p16 >> 4) & 1) ^ inv_r7) << 4) |
+                    ((((p16 >> 5) & 1) ^ ((p16 >> 8) & 1)) << 5) |
+                    ((((p16 >> 6) & 1) ^ ((p16 >> 9) & 1)) << 6) |
+                    (inv_r7 << 7);
+  const int s_row = p_src >> 7, s_col = p_src & 127;
+  auto stage2 = [&](int qt, char* sl) {
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) void*)(
+            reinterpret_cast<const char*>(qp) +
+            (long)min(qt + s_row, S - 1) * sio.rs * 2 + s_col),
+        (__attribute__((address_space(3))) void*)(sl + wdst), 16, 0, 0);
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) void*)(
+            reinterpret_cast<const char*>(dop) +
+            (long)min(qt + s_row, S - 1) * sdo.rs * 2 + s_col),
+        (__attribute__((address_space(3))) void*)(sl + 4096 + wdst),
+        16, 0, 0);
+  };
+  stage2(q_start, smem);
   for (int q0 = q_start; q0 < S; q0 += 32) {
     const int slot = ((q0 - q_start) >> 5) & 1;
     const char* q_lds = smem + slot * 8192;
     const char* do_lds = q_lds + 4096;
     __syncthreads();
     const bool have_next = q0 + 32 < S;
-    if (have_next) {  // T14: next tile's loads under this compute
-      qstage = *reinterpret_cast<const bf16x8v*>(
-          qp + (long)min(q0 + 32 + st_row, S - 1) * sio.rs + st_c16 / 2);
-      dstage = *reinterpret_cast<const bf16x8v*>(
-          dop + (long)min(q0 + 32 + st_row, S - 1) * sdo.rs + st_c16 / 2);
-    }
+    if (have_next)  // T14: next tile's LDS-DMA under this compute
+      stage2(q0 + 32, smem + (slot ^ 1) * 8192);
     const bool compute = live_wave && (!causal || q0 + 31 >= k0);
     if (compute) {
     f32x16 s_acc, dp_acc;
@@ -432,11 +457,6 @@ __global__ __launch_bounds__(256, 4) void attn_bwd_dk_kernel(
       }
     }
     }  // compute
-    if (have_next) {
-      char* nb = smem + (slot ^ 1) * 8192;
-      *reinterpret_cast<bf16x8v*>(nb + swz_dst) = qstage;
-      *reinterpret_cast<bf16x8v*>(nb + 4096 + swz_dst) = dstage;
-    }
   }
   if (!live_wave) return;
 #pragma unroll
@@ -517,25 +537,38 @@ __global__ __launch_bounds__(256) void attn_bwd_dvdk_kernel(
     }
 
   const int q_start = causal ? blockIdx.x * 128 : 0;
-  const int swz_dst = (st_row * 128 + st_c16) ^ ((st_row & 15) << 4);
-  bf16x8v qstage = *reinterpret_cast<const bf16x8v*>(
-      qp + (long)min(q_start + st_row, S - 1) * sio.rs + st_c16 / 2);
-  bf16x8v dstage = *reinterpret_cast<const bf16x8v*>(
-      dop + (long)min(q_start + st_row, S - 1) * sdo.rs + st_c16 / 2);
-  *reinterpret_cast<bf16x8v*>(smem + swz_dst) = qstage;
-  *reinterpret_cast<bf16x8v*>(smem + 4096 + swz_dst) = dstage;
+  // glds staging: both tiles are XOR-swizzled images, so both use the
+  // source-side inverse permutation (see the dv kernel's comment)
+  const int wdst = ((int)threadIdx.x >> 6) * 1024;
+  const int p16 = (int)threadIdx.x * 16;
+  const int inv_r7 = ((p16 >> 7) & 1) ^ ((p16 >> 10) & 1);
+  const int p_src = (p16 & ~0xF0) | ((((p16 >> 4) & 1) ^ inv_r7) << 4) |
+                    ((((p16 >> 5) & 1) ^ ((p16 >> 8) & 1)) << 5) |
+                    ((((p16 >> 6) & 1) ^ ((p16 >> 9) & 1)) << 6) |
+                    (inv_r7 << 7);
+  const int s_row = p_src >> 7, s_col = p_src & 127;
+  auto stage2 = [&](int qt, char* sl) {
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) void*)(
+            reinterpret_cast<const char*>(qp) +
+            (long)min(qt + s_row, S - 1) * sio.rs * 2 + s_col),
+        (__attribute__((address_space(3))) void*)(sl + wdst), 16, 0, 0);
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) void*)(
+            reinterpret_cast<const char*>(dop) +
+            (long)min(qt + s_row, S - 1) * sdo.rs * 2 + s_col),
+        (__attribute__((address_space(3))) void*)(sl + 4096 + wdst),
+        16, 0, 0);
+  };
+  stage2(q_start, smem);
   for (int q0 = q_start; q0 < S; q0 += 32) {
     const int slot = ((q0 - q_start) >> 5) & 1;
     const char* q_lds = smem + slot * 8192;
     const char* do_lds = q_lds + 4096;
     __syncthreads();
     const bool have_next = q0 + 32 < S;
-    if (have_next) {
-      qstage = *reinterpret_cast<const bf16x8v*>(
-          qp + (long)min(q0 + 32 + st_row, S - 1) * sio.rs + st_c16 / 2);
-      dstage = *reinterpret_cast<const bf16x8v*>(
-          dop + (long)min(q0 + 32 + st_row, S - 1) * sdo.rs + st_c16 / 2);
-    }
+    if (have_next)  // T14: next tile's LDS-DMA under this compute
+      stage2(q0 + 32, smem + (slot ^ 1) * 8192);
     const bool compute = live_wave && (!causal || q0 + 31 >= k0);
     if (compute) {
       f32x16 s_acc, dp_acc;
@@ -596,11 +629,6 @@ __global__ __launch_bounds__(256) void attn_bwd_dvdk_kernel(
               dk_acc[hh], 0, 0, 0);
         }
       }
-    }
-    if (have_next) {
-      char* nb = smem + (slot ^ 1) * 8192;
-      *reinterpret_cast<bf16x8v*>(nb + swz_dst) = qstage;
-      *reinterpret_cast<bf16x8v*>(nb + 4096 + swz_dst) = dstage;
     }
   }
   if (!live_wave) return;
@@ -674,25 +702,38 @@ __global__ __launch_bounds__(256, 3) void attn_bwd_dq_kernel(
 
   // block-wide kv range (barriers are block-wide); waves guard compute
   const int kv_end = causal ? min(S, blockIdx.x * 128 + 128) : S;
-  const int swz_dst = (st_row * 128 + st_c16) ^ ((st_row & 15) << 4);
-  bf16x8v kstage = *reinterpret_cast<const bf16x8v*>(
-      kp + (long)min(st_row, S - 1) * sio.rs + st_c16 / 2);
-  bf16x8v vstage = *reinterpret_cast<const bf16x8v*>(
-      vp + (long)min(st_row, S - 1) * sio.rs + st_c16 / 2);
-  *reinterpret_cast<bf16x8v*>(smem + swz_dst) = kstage;
-  *reinterpret_cast<bf16x8v*>(smem + 4096 + swz_dst) = vstage;
+  // glds staging: K and V tiles, source-side inverse of the XOR
+  // swizzle (see the dv kernel's comment)
+  const int wdst = ((int)threadIdx.x >> 6) * 1024;
+  const int p16 = (int)threadIdx.x * 16;
+  const int inv_r7 = ((p16 >> 7) & 1) ^ ((p16 >> 10) & 1);
+  const int p_src = (p16 & ~0xF0) | ((((p16 >> 4) & 1) ^ inv_r7) << 4) |
+                    ((((p16 >> 5) & 1) ^ ((p16 >> 8) & 1)) << 5) |
+                    ((((p16 >> 6) & 1) ^ ((p16 >> 9) & 1)) << 6) |
+                    (inv_r7 << 7);
+  const int s_row = p_src >> 7, s_col = p_src & 127;
+  auto stage2 = [&](int kt, char* sl) {
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) void*)(
+            reinterpret_cast<const char*>(kp) +
+            (long)min(kt + s_row, S - 1) * sio.rs * 2 + s_col),
+        (__attribute__((address_space(3))) void*)(sl + wdst), 16, 0, 0);
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) void*)(
+            reinterpret_cast<const char*>(vp) +
+            (long)min(kt + s_row, S - 1) * sio.rs * 2 + s_col),
+        (__attribute__((address_space(3))) void*)(sl + 4096 + wdst),
+        16, 0, 0);
+  };
+  stage2(0, smem);
   for (int k0 = 0; k0 < kv_end; k0 += 32) {
     const int slot = (k0 >> 5) & 1;
     const char* k_ldsb = smem + slot * 8192;
     const char* v_ldsb = k_ldsb + 4096;
     __syncthreads();
     const bool have_next = k0 + 32 < kv_end;
-    if (have_next) {  // T14: next tile's loads under this compute
-      kstage = *reinterpret_cast<const bf16x8v*>(
-          kp + (long)min(k0 + 32 + st_row, S - 1) * sio.rs + st_c16 / 2);
-      vstage = *reinterpret_cast<const bf16x8v*>(
-          vp + (long)min(k0 + 32 + st_row, S - 1) * sio.rs + st_c16 / 2);
-    }
+    if (have_next)  // T14: next tile's LDS-DMA under this compute
+      stage2(k0 + 32, smem + (slot ^ 1) * 8192);
     const bool compute = live_wave && (!causal || k0 <= q0 + 31);
     if (compute) {
     f32x16 s_acc, dp_acc;
@@ -752,11 +793,6 @@ __global__ __launch_bounds__(256, 3) void attn_bwd_dq_kernel(
       }
     }
     }  // compute
-    if (have_next) {
-      char* nb = smem + (slot ^ 1) * 8192;
-      *reinterpret_cast<bf16x8v*>(nb + swz_dst) = kstage;
-      *reinterpret_cast<bf16x8v*>(nb + 4096 + swz_dst) = vstage;
-    }
   }
   if (!live_wave) return;
 

@@ -320,14 +320,29 @@ __global__ __launch_bounds__(256, 4) void attn_fwd_lds_kernel(
   // after compute (the slot's last readers passed the barrier at the top
   // of THIS iteration).
   const int kv_all = causal ? min(S, blockIdx.x * 128 + 128) : S;
-  bf16x8v kstage = *reinterpret_cast<const bf16x8v*>(
-      kp + (long)min(st_row, S - 1) * sk.rs + st_c16 / 2);
+  // K tile rides global_load_lds (lane-linear dest; the XOR swizzle
+  // becomes the closed-form source inverse — see attention_bwd.hip).
+  // V cannot: its TR16 image uses 144-byte padded rows, which a
+  // contiguous lane-linear DMA cannot produce, so V keeps the register
+  // round-trip.
+  const int wdst = ((int)threadIdx.x >> 6) * 1024;
+  const int p16 = (int)threadIdx.x * 16;
+  const int inv_r7 = ((p16 >> 7) & 1) ^ ((p16 >> 10) & 1);
+  const int p_src = (p16 & ~0xF0) | ((((p16 >> 4) & 1) ^ inv_r7) << 4) |
+                    ((((p16 >> 5) & 1) ^ ((p16 >> 8) & 1)) << 5) |
+                    ((((p16 >> 6) & 1) ^ ((p16 >> 9) & 1)) << 6) |
+                    (inv_r7 << 7);
+  auto stage_k = [&](int kt, char* sl) {
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) void*)(
+            reinterpret_cast<const char*>(kp) +
+            (long)min(kt + (p_src >> 7), S - 1) * sk.rs * 2 + (p_src & 127)),
+        (__attribute__((address_space(3))) void*)(sl + wdst), 16, 0, 0);
+  };
   bf16x8v vstage = *reinterpret_cast<const bf16x8v*>(
       vp + (long)min(st_row, S - 1) * sv.rs + st_c16 / 2);
-  {  // prologue: stage tile 0 into slot 0
-    *reinterpret_cast<bf16x8v*>(smem + k_dst) = kstage;
-    *reinterpret_cast<bf16x8v*>(smem + 4096 + v_dst) = vstage;
-  }
+  stage_k(0, smem);  // prologue: stage tile 0 into slot 0
+  *reinterpret_cast<bf16x8v*>(smem + 4096 + v_dst) = vstage;
   for (int k0 = 0; k0 < kv_all; k0 += 32) {
     const int slot = (k0 >> 5) & 1;
     char* k_lds = smem + slot * SLOT;
@@ -335,11 +350,9 @@ __global__ __launch_bounds__(256, 4) void attn_fwd_lds_kernel(
     __syncthreads();  // this slot staged; other slot's readers done
     const bool have_next = k0 + 32 < kv_all;
     if (have_next) {  // issue next tile's loads under this compute
-      const int nrow = min(k0 + 32 + st_row, S - 1);
-      kstage = *reinterpret_cast<const bf16x8v*>(
-          kp + (long)nrow * sk.rs + st_c16 / 2);
+      stage_k(k0 + 32, smem + (slot ^ 1) * SLOT);
       vstage = *reinterpret_cast<const bf16x8v*>(
-          vp + (long)nrow * sv.rs + st_c16 / 2);
+          vp + (long)min(k0 + 32 + st_row, S - 1) * sv.rs + st_c16 / 2);
     }
     const bool compute = live_wave && (!causal || k0 <= q0 + 31);
     if (compute) {
@@ -495,10 +508,10 @@ __global__ __launch_bounds__(256, 4) void attn_fwd_lds_kernel(
         }
       }
     }
-    if (have_next) {  // stage t+1 into the other slot (safe: its last
-                      // readers passed this iteration's barrier)
+    if (have_next) {  // stage V t+1 into the other slot (safe: its last
+                      // readers passed this iteration's barrier; K's
+                      // DMA was issued at the top of this iteration)
       char* nk = smem + (slot ^ 1) * SLOT;
-      *reinterpret_cast<bf16x8v*>(nk + k_dst) = kstage;
       *reinterpret_cast<bf16x8v*>(nk + 4096 + v_dst) = vstage;
     }
   }

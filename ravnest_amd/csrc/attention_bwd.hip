@@ -153,7 +153,7 @@ __global__ void attn_bwd_delta_kernel(const bf16_t* __restrict__ dout,
 // cycles parked on memory waits — PMC profile r01)
 // -------------------------------------------------------------------
 template <bool DROP = false>
-__global__ __launch_bounds__(256, 4) void attn_bwd_dv_kernel(
+__global__ __launch_bounds__(256, DROP ? 4 : 5) void attn_bwd_dv_kernel(
     const bf16_t* __restrict__ q, const bf16_t* __restrict__ k,
     const bf16_t* __restrict__ dout, const float* __restrict__ lse,
     const float* __restrict__ mask, bf16_t* __restrict__ dv, int S,
@@ -307,7 +307,7 @@ __global__ __launch_bounds__(256, 4) void attn_bwd_dv_kernel(
 // dk kernel: wave owns keys; S, dP -> dS -> dK (~150 VGPR, 3 waves/SIMD)
 // -------------------------------------------------------------------
 template <bool DROP = false>
-__global__ __launch_bounds__(256, 4) void attn_bwd_dk_kernel(
+__global__ __launch_bounds__(256, DROP ? 3 : 4) void attn_bwd_dk_kernel(
     const bf16_t* __restrict__ q, const bf16_t* __restrict__ k,
     const bf16_t* __restrict__ v, const bf16_t* __restrict__ dout,
     const float* __restrict__ lse, const float* __restrict__ delta,
@@ -651,7 +651,7 @@ __global__ __launch_bounds__(256) void attn_bwd_dvdk_kernel(
 // orientation: acc = [key-pattern][qrow=lane&31]).
 // -------------------------------------------------------------------
 template <bool DROP = false>
-__global__ __launch_bounds__(256, 3) void attn_bwd_dq_kernel(
+__global__ __launch_bounds__(256, DROP ? 3 : 4) void attn_bwd_dq_kernel(
     const bf16_t* __restrict__ q, const bf16_t* __restrict__ k,
     const bf16_t* __restrict__ v, const bf16_t* __restrict__ dout,
     const float* __restrict__ lse, const float* __restrict__ delta,

@@ -45,14 +45,14 @@ def test_merge_partials_algebra():
     assert torch.allclose(om, torch.softmax(s2, -1) @ v, atol=1e-5)
 
 
-def _worker(rank, port, out_dir, causal, masked):
+def _worker(rank, port, out_dir, causal, masked, world=2):
     os.environ["MASTER_ADDR"] = "127.0.0.1"
     os.environ["MASTER_PORT"] = str(port)
-    dist.init_process_group("gloo", rank=rank, world_size=2)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
     from ravnest_amd.parallel import ring_attention
     torch.manual_seed(11)
-    B, H, Sg, D = 2, 2, 64, 32
-    Sl = Sg // 2
+    B, H, Sg, D = 2, 2, 96, 32
+    Sl = Sg // world
     q = torch.randn(B, H, Sg, D)
     k = torch.randn(B, H, Sg, D)
     v = torch.randn(B, H, Sg, D)
@@ -74,13 +74,14 @@ def _worker(rank, port, out_dir, causal, masked):
     dist.destroy_process_group()
 
 
-def _run(tmp_path, causal, masked):
+def _run(tmp_path, causal, masked, world=2):
     port = 29820 + (os.getpid() % 30) + (7 if causal else 0) + \
-        (13 if masked else 0)
+        (13 if masked else 0) + 3 * world
     ctx = mp.get_context("spawn")
     procs = [ctx.Process(target=_worker,
-                         args=(r, port, str(tmp_path), causal, masked))
-             for r in range(2)]
+                         args=(r, port, str(tmp_path), causal, masked,
+                               world))
+             for r in range(world)]
     for p in procs:
         p.start()
     for p in procs:
@@ -90,8 +91,8 @@ def _run(tmp_path, causal, masked):
 
     # single-process reference on the SAME tensors
     torch.manual_seed(11)
-    B, H, Sg, D = 2, 2, 64, 32
-    Sl = Sg // 2
+    B, H, Sg, D = 2, 2, 96, 32
+    Sl = Sg // world
     q = torch.randn(B, H, Sg, D).requires_grad_(True)
     k = torch.randn(B, H, Sg, D).requires_grad_(True)
     v = torch.randn(B, H, Sg, D).requires_grad_(True)
@@ -105,7 +106,7 @@ def _run(tmp_path, causal, masked):
         s = s + ((1 - am) * -10000.0).view(B, 1, 1, Sg)
     ref = torch.softmax(s, -1) @ v
     ref.backward(do)
-    for r in range(2):
+    for r in range(world):
         got = torch.load(tmp_path / f"ring_{r}.pt")
         sl = slice(r * Sl, (r + 1) * Sl)
         for name, mine, full in [("o", got["o"], ref[:, :, sl]),
@@ -126,6 +127,11 @@ def test_ring_attention_causal(tmp_path):
 
 def test_ring_attention_masked(tmp_path):
     _run(tmp_path, causal=False, masked=True)
+
+
+def test_ring_attention_world3_causal(tmp_path):
+    """3 hops: multi-shift ordering + causal chunk skips."""
+    _run(tmp_path, causal=True, masked=False, world=3)
 
 
 def _gpu_worker(rank, port, out_dir):

@@ -5,10 +5,10 @@ Reference workload parity: minGPT causal self-attention
 SURVEY.md section 2.3. Forward is the hand-written HIP kernel
 (csrc/attention.hip): per Q-tile online-softmax over K/V tiles, bf16 MFMA
 (32x32x16), XOR-swizzled K LDS tiles, no S x S score matrix in HBM; it
-returns O and the log-sum-exp rows. Backward recomputes P from (Q,K,lse)
-with plain hipBLASLt GEMMs + elementwise torch (library GEMMs are the
-sanctioned path for non-fused matmuls; the fused bwd kernel is tracked
-for a later round).
+returns O and the log-sum-exp rows. Backward on the D=64 path runs the
+fused flash backward kernels (csrc/attention_bwd.hip: delta + dv + dk +
+dq, P recomputed from (Q,K,lse) on-chip); other head dims recompute P
+with plain hipBLASLt GEMMs + elementwise torch.
 
 Shapes: q,k,v (B, H, S, D); additive mask broadcastable to (B, 1, S, S)
 or None; causal flag for GPT-style models.

@@ -154,3 +154,41 @@ def test_model_fusion(tmp_path):
     assert set(fused.keys()) == set(orig.keys())
     for k in orig:
         assert torch.equal(fused[k], orig[k])
+
+
+def test_form_clusters_heterogeneous():
+    """GA placement on an uneven pool (reference genetic.py parity): a
+    model needing 10 GiB over {8, 8, 4, 4, 12, 12} GiB nodes — every
+    cluster must hold the model, small nodes must pair with big ones,
+    and the speed balance must keep clusters within the pool's spread."""
+    from ravnest_amd.planner.placement import NodeSpec, form_clusters
+    G = 2**30
+    pool = [NodeSpec("a", 8 * G, bandwidth=1.0),
+            NodeSpec("b", 8 * G, bandwidth=1.0),
+            NodeSpec("c", 4 * G, bandwidth=2.0),
+            NodeSpec("d", 4 * G, bandwidth=2.0),
+            NodeSpec("e", 12 * G, bandwidth=0.5),
+            NodeSpec("f", 12 * G, bandwidth=0.5)]
+    clusters = form_clusters(pool, model_bytes=10 * G, seed=3)
+    assert len(clusters) >= 2  # enough RAM for at least two replicas
+    for c in clusters:
+        assert c.total_ram >= 10 * G, \
+            f"cluster {c.cid} cannot hold the model: {c.total_ram/G} GiB"
+    # all nodes placed exactly once
+    placed = sorted(n.name for c in clusters for n in c.nodes)
+    assert placed == sorted(n.name for n in pool)
+    # proportional splits follow member RAM
+    big = max(clusters, key=lambda c: len(c.nodes))
+    props = big.split_proportions(10 * G)
+    assert abs(sum(props) - 1.0) < 1e-6
+    rams = [n.ram for n in big.nodes]
+    order_by_prop = sorted(range(len(props)), key=lambda i: props[i])
+    order_by_ram = sorted(range(len(rams)), key=lambda i: rams[i])
+    assert order_by_prop == order_by_ram
+
+
+def test_form_clusters_insufficient_ram():
+    from ravnest_amd.planner.placement import NodeSpec, form_clusters
+    import pytest as _pytest
+    with _pytest.raises(ValueError):
+        form_clusters([NodeSpec("a", 2**30)], model_bytes=10 * 2**30)

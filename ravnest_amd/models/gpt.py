@@ -60,7 +60,11 @@ class GPTEmbeddings(nn.Module):
                              persistent=False)
 
     def forward(self, idx):
-        if idx.is_cuda and self.tok_emb.weight.dtype == torch.bfloat16:
+        # sequence-parallel shards carry a position offset in pos_ids
+        # (parallel/sp.py); the fused stem indexes positions from 0, so
+        # SP routes through the eager gather
+        if idx.is_cuda and self.tok_emb.weight.dtype == torch.bfloat16 \
+                and not getattr(self, "_sp_offset", 0):
             x = embedding_add(idx, self.tok_emb.weight, self.pos_emb.weight)
             return self.drop(x)
         S = idx.size(1)

@@ -10,5 +10,8 @@ bwd with a caller-provided GLOBAL lse).
 """
 from .ring_attention import (RingAttention, ring_attention,
                              merge_partials)
+from .sp import (RingCoreQKV, allreduce_gradients,
+                 sequence_parallelize)
 
-__all__ = ["RingAttention", "ring_attention", "merge_partials"]
+__all__ = ["RingAttention", "ring_attention", "merge_partials",
+           "RingCoreQKV", "sequence_parallelize", "allreduce_gradients"]

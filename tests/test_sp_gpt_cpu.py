@@ -92,3 +92,56 @@ def test_sequence_parallel_gpt(tmp_path):
         for r in range(2):
             err = (res[r][name] - ref).abs().max()
             assert err < 1e-4, f"rank {r} {name} err {err}"
+
+
+def _gpu_worker(rank, port, out_dir):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=2)
+    from ravnest_amd.parallel import sequence_parallelize
+    model, cfg = _build()
+    model = model.to("cuda", torch.bfloat16)
+    sequence_parallelize(model)
+    torch.manual_seed(99)
+    B, Sg = 2, cfg.block_size
+    Sl = Sg // 2
+    idx = torch.randint(0, cfg.vocab_size, (B, Sg)).cuda()
+    logits = model(idx[:, rank * Sl:(rank + 1) * Sl])
+    torch.save({"logits": logits.detach().float().cpu()},
+               os.path.join(out_dir, f"spg_{rank}.pt"))
+    dist.barrier()
+    dist.destroy_process_group()
+
+
+import pytest  # noqa: E402
+
+
+@pytest.mark.gpu
+def test_sequence_parallel_gpt_gpu(tmp_path):
+    """Whole-model SP forward on GPU (bf16, ring kernel chunks, eager
+    embedding offsets) vs the single-process bf16 model."""
+    if not torch.cuda.is_available():
+        pytest.skip("no GPU")
+    port = 29750 + (os.getpid() % 30)
+    ctx = mp.get_context("spawn")
+    procs = [ctx.Process(target=_gpu_worker, args=(r, port, str(tmp_path)))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(timeout=300)
+    for p in procs:
+        assert p.exitcode == 0, p.exitcode
+    model, cfg = _build()
+    model = model.to("cuda", torch.bfloat16)
+    torch.manual_seed(99)
+    B, Sg = 2, cfg.block_size
+    Sl = Sg // 2
+    idx = torch.randint(0, cfg.vocab_size, (B, Sg)).cuda()
+    ref = model(idx).float().cpu()
+    scale = ref.abs().max().item()
+    for r in range(2):
+        got = torch.load(tmp_path / f"spg_{r}.pt")["logits"]
+        sl = slice(r * Sl, (r + 1) * Sl)
+        err = (got - ref[:, sl]).abs().max().item()
+        assert err < 0.05 * scale + 0.05, f"rank {r} err {err} scale {scale}"

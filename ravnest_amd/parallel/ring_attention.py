@@ -201,9 +201,11 @@ class _RingAttnFn(torch.autograd.Function):
                 dk_c += dk_i
                 dv_c += dv_i
             # one shift per step: after `world` shifts every chunk's
-            # accumulated (dk, dv) is back at its owner
-            k_c, v_c, dk_c, dv_c = _ring_shift(
-                [k_c, v_c, dk_c, dv_c], group, nxt, prv)
+            # accumulated (dk, dv) is back at its owner (world 1: the
+            # chunk is already home; a send-to-self would deadlock)
+            if world > 1:
+                k_c, v_c, dk_c, dv_c = _ring_shift(
+                    [k_c, v_c, dk_c, dv_c], group, nxt, prv)
         return dq, dk_c, dv_c, None, None, None, None
 
 

@@ -204,3 +204,44 @@ def test_ring_attention_gpu_kernel_path(tmp_path):
                                  ("dv", got["dv"], v.grad[:, :, sl])]:
             err = (mine.float() - full.detach()).abs().max().item()
             assert err < 5e-2, f"rank {r} {name} err {err}"
+
+
+def _w1_worker(rank, port, out_dir):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=0, world_size=1)
+    from ravnest_amd.parallel import ring_attention
+    torch.manual_seed(2)
+    q = torch.randn(1, 2, 32, 16, requires_grad=True)
+    k = torch.randn(1, 2, 32, 16, requires_grad=True)
+    v = torch.randn(1, 2, 32, 16, requires_grad=True)
+    do = torch.randn(1, 2, 32, 16)
+    o = ring_attention(q, k, v, causal=True)
+    o.backward(do)  # world-1 backward must not ring-shift to itself
+    torch.save({"o": o.detach(), "dq": q.grad, "dk": k.grad,
+                "dv": v.grad}, os.path.join(out_dir, "w1.pt"))
+    dist.destroy_process_group()
+
+
+def test_ring_attention_world1(tmp_path):
+    """Degenerate single-rank group == plain attention (fwd + bwd)."""
+    ctx = mp.get_context("spawn")
+    p = ctx.Process(target=_w1_worker, args=(0, 29815 + os.getpid() % 30,
+                                             str(tmp_path)))
+    p.start()
+    p.join(timeout=120)
+    assert p.exitcode == 0, p.exitcode
+    torch.manual_seed(2)
+    q = torch.randn(1, 2, 32, 16, requires_grad=True)
+    k = torch.randn(1, 2, 32, 16, requires_grad=True)
+    v = torch.randn(1, 2, 32, 16, requires_grad=True)
+    do = torch.randn(1, 2, 32, 16)
+    s = (q @ k.transpose(-2, -1)) * (1.0 / math.sqrt(16))
+    s = s + torch.triu(torch.full((32, 32), float("-inf")), 1)
+    ref = torch.softmax(s, -1) @ v
+    ref.backward(do)
+    got = torch.load(tmp_path / "w1.pt")
+    for name, mine, full in [("o", got["o"], ref), ("dq", got["dq"], q.grad),
+                             ("dk", got["dk"], k.grad),
+                             ("dv", got["dv"], v.grad)]:
+        assert (mine - full.detach()).abs().max() < 2e-4, name

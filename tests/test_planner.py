@@ -192,3 +192,25 @@ def test_form_clusters_insufficient_ram():
     import pytest as _pytest
     with _pytest.raises(ValueError):
         form_clusters([NodeSpec("a", 2**30)], model_bytes=10 * 2**30)
+
+
+def test_load_node_pool_reference_format(tmp_path):
+    """Reference-compatible node_configs.json round trip (parity:
+    spawn_node_pool load_from_configs, operations/utils.py:24-50)."""
+    import json
+    from ravnest_amd.planner.placement import load_node_pool, mi355x_pool
+    cfg = {"0": {"IP": "0.0.0.0:8080", "benchmarks": {"ram": 8,
+                                                      "bandwidth": 10}},
+           "1": {"IP": "0.0.0.0:8081", "benchmarks": {"ram": 16,
+                                                      "bandwidth": 5}},
+           "10": {"IP": "0.0.0.0:8090"}}  # defaults + numeric sort
+    p = tmp_path / "node_configs.json"
+    p.write_text(json.dumps(cfg))
+    pool = load_node_pool(p)
+    assert [n.name for n in pool] == ["node_0", "node_1", "node_10"]
+    assert pool[0].ram == 8 * 2**30 and pool[0].bandwidth == 10
+    assert pool[1].address == "0.0.0.0:8081"
+    assert pool[2].ram == 8 * 2**30  # default
+    gpus = mi355x_pool(8)
+    assert len(gpus) == 8 and gpus[3].device_index == 3
+    assert gpus[0].ram > 250 * 2**30  # 288 GB minus reserve

@@ -93,3 +93,59 @@ def test_channel_fp8_wire(tmp_path):
     g_err = (rx["grads"] - grads).abs().max()
     assert g_err <= grads.abs().max() * 0.04, g_err
     assert (rx["ids"] == torch.arange(6)).all()
+
+
+def _gpu_worker(rank, port, out_dir):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    os.environ["RAVNEST_WIRE_FP8"] = "1"
+    dist.init_process_group("gloo", rank=rank, world_size=2)
+    got = {}
+
+    def deliver(ch, msg):
+        got["msg"] = msg
+
+    dev = torch.device("cuda", 0)
+    ch = Channel(src=0, dst=1, kind="fwd", group=dist.group.WORLD,
+                 device=dev, my_rank=rank, deliver=deliver)
+    ch.start()
+    torch.manual_seed(9)
+    acts = (torch.randn(4, 16, 64) * 30).to(dev, torch.bfloat16)
+    if rank == 0:
+        ch.send(Message(action=ActionTypes.FORWARD, fpid=1,
+                        tensors=[(7, acts)]))
+        ch.close()
+        ch.join(timeout=60)
+    else:
+        ch.join(timeout=60)
+        (gid, rx) = got["msg"].tensors[0]
+        assert gid == 7 and rx.is_cuda and rx.dtype == torch.bfloat16
+        err = (rx.float() - acts.float()).abs().max()
+        amax = acts.float().abs().max()
+        assert err <= amax * 0.04, f"{err} vs {amax}"
+        torch.save({"ok": True}, os.path.join(out_dir, "f8g.pt"))
+    dist.barrier()
+    dist.destroy_process_group()
+
+
+import pytest  # noqa: E402
+
+
+@pytest.mark.gpu
+def test_channel_fp8_wire_gpu(tmp_path):
+    """fp8 wire with CUDA payloads: quantize on device, host-staged
+    gloo transport (wire_cpu), reconstruct on the receiver's GPU."""
+    if not torch.cuda.is_available():
+        pytest.skip("no GPU")
+    import torch.multiprocessing as mp2
+    port = 29690 + (os.getpid() % 30)
+    ctx = mp2.get_context("spawn")
+    procs = [ctx.Process(target=_gpu_worker, args=(r, port, str(tmp_path)))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(timeout=240)
+    for p in procs:
+        assert p.exitcode == 0, p.exitcode
+    assert (tmp_path / "f8g.pt").exists()

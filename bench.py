@@ -49,6 +49,8 @@ def build_model(name: str, seq: int):
         return BertForMLM(BertConfig.tiny(max_seq=seq)), "bert"
     if name == "gpt2-small":
         return GPT(GPTConfig.gpt2_small(block_size=seq)), "gpt"
+    if name == "gpt-nano":
+        return GPT(GPTConfig.nano64(block_size=seq)), "gpt"
     if name == "gpt2-small-fp8":
         # BASELINE.json config 5: block projections on the MX fp8 MFMA
         # path (ops.MXLinear -> csrc/mx_gemm2.hip)
@@ -172,6 +174,88 @@ def run_pp(args, rank, world, device, on_gpu, family, model, batches,
     os._exit(0)
 
 
+def run_sp(args, rank, world, device, on_gpu, family, model, batches,
+           labels, share_gpu=False):
+    """Sequence-parallel bench: ONE GPT with the sequence sharded across
+    the ranks — ring attention in every block, token-local everything
+    else, parameter grads all-reduced over the SP group
+    (ravnest_amd/parallel/sp.py). Strong scaling: the global
+    (batch, seq) problem is fixed while ranks grow."""
+    import time as _time
+    import torch.distributed as dist
+    if family != "gpt":
+        raise SystemExit("--parallelism sp supports the GPT family")
+    if args.seq % world:
+        raise SystemExit("--seq must divide by the SP world size")
+    backend = "gloo" if (share_gpu or not on_gpu) else "nccl"
+    dist.init_process_group(backend, rank=rank, world_size=world)
+    from ravnest_amd.parallel import (allreduce_gradients,
+                                      sequence_parallelize)
+    sequence_parallelize(model)
+    Sl = args.seq // world
+    sl = slice(rank * Sl, (rank + 1) * Sl)
+    shards = [b["idx"][:, sl].contiguous() for b in batches]
+    labs = [l[:, sl].contiguous() for l in labels]
+    from ravnest_amd.ops import FusedAdam
+    opt = (FusedAdam(model.parameters(), lr=1e-4) if on_gpu
+           else torch.optim.Adam(model.parameters(), lr=1e-4))
+    crit = torch.nn.CrossEntropyLoss(ignore_index=-100)
+    V = model.cfg.vocab_size
+
+    def step(i):
+        opt.zero_grad(set_to_none=True)
+        logits = model(shards[i % len(shards)])
+        loss = crit(logits.reshape(-1, V).float(),
+                    labs[i % len(labs)].reshape(-1))
+        loss.backward()
+        allreduce_gradients(model)
+        opt.step()
+
+    def sync():
+        if on_gpu:
+            torch.cuda.synchronize(device)
+
+    for i in range(args.warmup):
+        step(i)
+    sync()
+    dist.barrier()
+    sync()
+    t0 = _time.perf_counter()
+    for i in range(args.steps):
+        step(i)
+    sync()
+    elapsed = _time.perf_counter() - t0
+    dist.barrier()
+    t = torch.tensor([elapsed], dtype=torch.float64,
+                     device=device if backend == "nccl" else "cpu")
+    dist.all_reduce(t, op=dist.ReduceOp.MAX)
+    elapsed = float(t.item())
+    if rank == 0:
+        samples_per_sec = args.micro_batch * args.steps / elapsed
+        print(json.dumps({
+            "metric": f"samples/sec (whole node), {args.model} "
+                      "sequence-parallel",
+            "value": round(samples_per_sec, 2),
+            "unit": "samples/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(elapsed / args.steps * 1000.0, 2),
+            "higher_is_better": True,
+            "scaling": "strong",
+            "vs_baseline": None,
+            "dtype": args.dtype if on_gpu else "fp32",
+            "data": "synthetic",
+            "config": {"model": args.model,
+                       "global_batch": args.micro_batch,
+                       "seq_len": args.seq,
+                       "parallelism": f"sp{world}",
+                       "optimizer": "fused_adam" if on_gpu else "adam"},
+        }), flush=True)
+    dist.barrier()
+    dist.destroy_process_group()
+
+
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
@@ -182,10 +266,13 @@ def main():
     ap.add_argument("--seq", type=int, default=512)
     ap.add_argument("--reduce-factor", type=int, default=4,
                     help="average DP replicas every N steps")
-    ap.add_argument("--parallelism", default="dp", choices=["dp", "pp"],
+    ap.add_argument("--parallelism", default="dp",
+                    choices=["dp", "pp", "sp"],
                     help="dp: one fused replica per GPU (driver default); "
                          "pp: ONE async pipeline of N stages across the "
-                         "N ranks (the reference's headline topology)")
+                         "N ranks (the reference's headline topology); "
+                         "sp: ONE model with the sequence sharded across "
+                         "the N ranks (ring attention)")
     ap.add_argument("--dtype", default=None, choices=["bf16", "fp32"])
     ap.add_argument("--cpu", action="store_true",
                     help="CPU sanity mode (tiny model)")
@@ -272,6 +359,10 @@ def main():
 
     if args.parallelism == "pp" and world > 1:
         run_pp(args, rank, world, device, on_gpu, family, model,
+               batches, labels, share_gpu=share_gpu)
+        return
+    if args.parallelism == "sp" and world > 1:
+        run_sp(args, rank, world, device, on_gpu, family, model,
                batches, labels, share_gpu=share_gpu)
         return
 
